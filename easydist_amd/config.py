@@ -1,0 +1,121 @@
+"""Global configuration for easydist_amd.
+
+Environment-variable driven flags, mirroring the capability of the reference's
+``easydist/config.py`` (reference: easydist/config.py:28-126) but with MI355X
+defaults: the communication cost model is expressed in xGMI terms (7 p2p links
+x ~153 GB/s per GPU), memory budget in terms of 288 GB HBM3E, and the compute
+roofline in CDNA4 MFMA terms.
+"""
+import os
+import sys
+
+
+def _env_flag(name: str, default: bool = False) -> bool:
+    v = os.environ.get(name)
+    if v is None:
+        return default
+    return v.lower() in ("1", "true", "yes", "on")
+
+
+def _env_int(name: str, default: int) -> int:
+    return int(os.environ.get(name, default))
+
+
+def _env_float(name: str, default: float) -> float:
+    return float(os.environ.get(name, default))
+
+
+# ---------------------------------------------------------------- logging ----
+log_level = os.environ.get("EASYDIST_LOGLEVEL", "INFO")
+
+# ------------------------------------------------------------------ dumps ----
+dump_dir = os.environ.get("EASYDIST_DUMP_DIR", "./md_dump")
+dump_fx_graph = _env_flag("EASYDIST_DUMP_FX_GRAPH")
+dump_strategy = _env_flag("EASYDIST_DUMP_STRATEGY")
+dump_metair = _env_flag("EASYDIST_DUMP_METAIR")
+
+# ---------------------------------------------------------------- devices ----
+easydist_device = os.environ.get("EASYDIST_DEVICE", "cuda")
+forced_compile = _env_flag("EASYDIST_FORCED_COMPILE")
+
+# ------------------------------------------------------------- compile cache --
+enable_compile_cache = _env_flag("EASYDIST_COMPILE_CACHE")
+compile_cache_dir = os.environ.get("EASYDIST_COMPILE_CACHE_DIR", "./md_compiled")
+
+# --------------------------------------------------------- rule discovery ----
+# During ShardCombine discovery huge dims are shrunk to this size so that the
+# op executions used for rule search stay cheap and never OOM
+# (reference behavior: easydist/torch/sharding_interpreter.py:256-281).
+use_hint = False
+discovery_max_dim = _env_int("EASYDIST_DISCOVERY_MAX_DIM", 1024)
+# number of shards used while searching for rules (not the mesh size)
+discovery_num_shards = _env_int("EASYDIST_DISCOVERY_NUM_SHARDS", 2)
+discovery_rtol = _env_float("EASYDIST_DISCOVERY_RTOL", 1e-3)
+discovery_atol = _env_float("EASYDIST_DISCOVERY_ATOL", 1e-4)
+max_halo = _env_int("EASYDIST_MAX_HALO", 3)
+
+# ----------------------------------------------------------------- solver ----
+# Time limit for the per-mesh-dim MILP (scipy HiGHS).
+solver_time_limit = _env_float("EASYDIST_SOLVER_TIME_LIMIT", 120.0)
+max_seconds_same_incumbent = float("inf")
+enable_graph_coarsen = _env_flag("EASYDIST_ENABLE_GRAPH_COARSEN", True)
+coarsen_level = _env_int("EASYDIST_COARSEN_LEVEL", 1)
+solver_mode = os.environ.get("EASYDIST_SOLVER_MODE", "ilp")  # ilp | beam
+beam_width = _env_int("EASYDIST_BEAM_WIDTH", 1024)
+all_to_all_punish_factor = _env_float("EASYDIST_A2A_PUNISH", 1.5)
+liveness_only_input = False
+
+# ----------------------------------------------------- MI355X cost model -----
+# xGMI: each MI355X has 7 point-to-point links of ~153 GB/s against its peers.
+# Ring collectives are bound by ONE link's bandwidth per direction.
+XGMI_LINK_BW = _env_float("EASYDIST_XGMI_LINK_BW", 153e9)       # bytes/sec
+XGMI_NUM_LINKS = _env_int("EASYDIST_XGMI_NUM_LINKS", 7)
+HBM_BYTES = _env_float("EASYDIST_HBM_BYTES", 288e9)             # per GPU
+HBM_BW = _env_float("EASYDIST_HBM_BW", 6.3e12)                  # measured achievable
+MFMA_BF16_FLOPS = _env_float("EASYDIST_MFMA_BF16_FLOPS", 2.5e15)
+MFMA_FP32_FLOPS = _env_float("EASYDIST_MFMA_FP32_FLOPS", 157.3e12)
+COLLECTIVE_LATENCY = _env_float("EASYDIST_COLL_LATENCY", 10e-6)  # seconds per call
+# fraction of HBM the memory-aware solver may plan into
+mem_ratio = _env_float("EASYDIST_MEM_RATIO", 0.9)
+
+# ------------------------------------------------------------------ comms ----
+# Use the hand-written xGMI all-to-all (pairwise p2p) instead of RCCL's
+# generic path for EP dispatch.
+use_xgmi_all_to_all = _env_flag("EASYDIST_XGMI_A2A", True)
+comm_optimization = _env_flag("EASYDIST_COMM_OPT")
+rcpsp_method = os.environ.get("EASYDIST_RCPSP_METHOD", "odd_even")
+enable_tile_comm = _env_flag("EASYDIST_TILE_COMM")
+override_dtensor_rule = _env_flag("EASYDIST_OVERRIDE_DTENSOR_RULE")
+
+# ----------------------------------------------------------------- kernels ---
+# Hand-written HIP/CDNA4 kernels for the hot ops. On a GPU box the extension
+# must load; on CPU-only hosts the aten fallback is used.
+use_hip_kernels = _env_flag("EASYDIST_USE_HIP_KERNELS", True)
+# Per-op overrides (profiling-driven): "gemm" may fall back to hipBLASLt
+# (torch.mm) per shape; see easydist_amd/ops/gemm.py.
+hip_gemm = _env_flag("EASYDIST_HIP_GEMM", False)
+
+# ------------------------------------------------------------------ runtime --
+enable_hip_graph = _env_flag("EASYDIST_HIP_GRAPH", True)
+use_contiguous_buffer = _env_flag("EASYDIST_CONTIGUOUS_BUFFER")
+enable_memory_opt = _env_flag("EASYDIST_MEM_OPT")
+mem_opt_by_module = _env_flag("EASYDIST_MEM_OPT_BY_MODULE")
+enable_runtime_trace = _env_flag("EASYDIST_RUNTIME_TRACE")
+ignore_memory_plan = _env_flag("EASYDIST_IGNORE_MEMORY_PLAN")
+
+# ------------------------------------------------------------------ prof -----
+prof_warmup = _env_int("EASYDIST_PROF_WARMUP", 2)
+prof_trials = _env_int("EASYDIST_PROF_TRIALS", 5)
+profile_db_path = os.environ.get("EASYDIST_PERFDB",
+                                 os.path.expanduser("~/.easydist_amd/perf.db"))
+enable_perfdb = _env_flag("EASYDIST_PERFDB_ENABLE", True)
+
+# ---------------------------------------------------------------- pipeline ---
+pp_local_stage_cnt = _env_int("EASYDIST_PP_LOCAL_STAGES", 0)
+
+
+def get_solver_time_limit() -> float:
+    limit = solver_time_limit
+    if max_seconds_same_incumbent != float("inf"):
+        limit = min(limit, max_seconds_same_incumbent)
+    return limit
