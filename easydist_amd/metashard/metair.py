@@ -91,7 +91,8 @@ class NodeSPMDStrategy:
 class MetaNode:
     def __init__(self, name: str, op_name: str, invars: List[MetaVar],
                  outvars: List[Optional[MetaVar]], sharding_ann=None,
-                 combination_ann=None, is_placeholder: bool = False):
+                 combination_ann=None, is_placeholder: bool = False,
+                 flops: float = 0.0):
         self.name = name
         self.op_name = op_name
         self.invars = invars
@@ -99,6 +100,9 @@ class MetaNode:
         self.sharding_ann = sharding_ann
         self.combination_ann = combination_ann or {}
         self.is_placeholder = is_placeholder
+        # FLOPs of the op at global shapes (matmul-family filled by the
+        # bridge); 0 means "memory-bound: use the byte-count proxy"
+        self.flops = flops
         self._pool_cache: Dict[Tuple, List[NodeSPMDStrategy]] = {}
 
     # ------------------------------------------------------- strategy pool ---
@@ -269,13 +273,22 @@ class MetaGraph:
                     cnt += 1
             consumers_count[node.name] = cnt
 
+        def is_shrink(node: MetaNode) -> bool:
+            """Cone roots sit at shrinking nodes (reference metair.py:852-917):
+            a node whose outputs are smaller than its inputs ends the
+            sync-free region, so resharding may happen after it."""
+            out_b = sum(v.nbytes for v in node.outvars if v is not None)
+            in_b = sum(v.nbytes for v in node.invars if v is not None)
+            return out_b < in_b
+
         cluster_of: Dict[str, MetaNodeCluster] = {}
         clusters: List[MetaNodeCluster] = []
         MAX_CLUSTER = 12
         # reverse topo: consumers first
         for node in reversed(self.nodes):
             target_cluster = None
-            if level >= 1 and consumers_count[node.name] == 1:
+            is_root = consumers_count[node.name] != 1 or is_shrink(node)
+            if level >= 1 and not is_root:
                 # the single consumer node
                 cons = None
                 for v in node.outvars:
@@ -314,6 +327,7 @@ class ClusterStrategy:
     in_placements: Dict[str, SPMD]    # external invar name -> required placement
     out_placements: Dict[str, SPMD]   # externally visible outvar -> placement
     mem_cost: float = 0.0
+    comp_cost: float = 0.0            # seconds on the CDNA4 roofline
 
 
 class MetaNodeCluster:
@@ -361,6 +375,7 @@ class MetaNodeCluster:
             self.strategies.append(st)
         for st in self.strategies:
             st.mem_cost = self._mem_cost(st, mesh_size)
+            st.comp_cost = self._comp_cost(st, mesh_size)
         return self.strategies
 
     def _derive(self, root, root_strat, interior_names, produced_here,
@@ -439,6 +454,25 @@ class MetaNodeCluster:
                 pl = strat.out_placements[k]
                 local = v.nbytes / (mesh_size if pl.is_shard() else 1)
                 total += local
+        return total
+
+    def _comp_cost(self, st: ClusterStrategy, mesh_size: int) -> float:
+        """CDNA4 roofline estimate: max(MFMA time, HBM time) per node,
+        divided by mesh_size when the node's work is actually partitioned
+        (any output SHARD/PARTIAL)."""
+        from .. import config as mdconfig
+        total = 0.0
+        for n in self.nodes:
+            strat = st.node_strategies.get(n.name)
+            if strat is None or n.is_placeholder:
+                continue
+            nbytes = sum(v.nbytes for v in n.outvars if v is not None)
+            nbytes += sum(v.nbytes for v in n.invars if v is not None)
+            t_mem = nbytes / mdconfig.HBM_BW
+            t_mfma = n.flops / mdconfig.MFMA_BF16_FLOPS
+            t = max(t_mem, t_mfma)
+            partitioned = any(not p.is_replicate() for p in strat.out_placements)
+            total += t / mesh_size if partitioned else t
         return total
 
     def __repr__(self):
