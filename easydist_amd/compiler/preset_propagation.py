@@ -1,0 +1,165 @@
+"""Preset sharding rules for ops where execution-based discovery is wrong
+or wasteful (randomness, views, factories).
+
+Capability parity with reference ``easydist/torch/preset_propagation.py``
+(registry + rules for placeholder/view/factory/dropout ops, lines 28-130).
+"""
+from __future__ import annotations
+
+import functools
+from typing import Callable, Dict, List, Optional, Tuple
+
+import torch
+
+from ..metashard.annotation import NoShardDim, ShardAnnotation, ShardDim
+from ..metashard.combination import CombinationFunc
+from ..metashard.view_propagation import view_propagation
+
+aten = torch.ops.aten
+
+# op overload -> fn(input_shapes, args, kwargs) -> (ShardAnnotation, {id: comb})
+_PRESET_REGISTRY: Dict[object, Callable] = {}
+
+
+def register_preset(*ops):
+    def deco(fn):
+        for op in ops:
+            _PRESET_REGISTRY[op] = fn
+        return fn
+    return deco
+
+
+def preset_meta_spmd(op, input_shapes, args, kwargs):
+    fn = _PRESET_REGISTRY.get(op)
+    if fn is None:
+        return None
+    return fn(input_shapes, args, kwargs)
+
+
+def _gather(d):
+    return functools.partial(CombinationFunc.gather, dim=d)
+
+
+# ------------------------------------------------------------------ views ----
+@register_preset(aten.view.default, aten._unsafe_view.default,
+                 aten.reshape.default)
+def _view_rule(input_shapes, args, kwargs):
+    in_shape = list(input_shapes[0])
+    out_shape = list(args[1])
+    mapping = view_propagation(in_shape, out_shape)
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for in_dim, out_dim in mapping.items():
+        if in_shape[in_dim] <= 1:
+            continue
+        ann[0][in_dim] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(out_dim)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.expand.default)
+def _expand_rule(input_shapes, args, kwargs):
+    in_shape = list(input_shapes[0])
+    out_shape = list(args[1])
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    # trailing-aligned dims; an input dim only shards when it is not being
+    # broadcast (size preserved and > 1)
+    off = len(out_shape) - len(in_shape)
+    for i, s in enumerate(in_shape):
+        o = out_shape[i + off]
+        if s > 1 and (o == s or o == -1):
+            ann[0][i] = ShardDim.get_shard_dim(sid)
+            combs[sid] = _gather(i + off)
+            sid += 1
+    return ann, combs
+
+
+# ---------------------------------------------------------------- dropout ----
+@register_preset(aten.native_dropout.default)
+def _dropout_rule(input_shapes, args, kwargs):
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(input_shapes[0])):
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = [_gather(d), _gather(d)]   # output + mask
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.native_dropout_backward.default)
+def _dropout_bwd_rule(input_shapes, args, kwargs):
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(input_shapes[0])):
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        if len(input_shapes) > 1 and d < len(input_shapes[1]):
+            ann[1][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs
+
+
+# -------------------------------------------------------------- factories ----
+@register_preset(aten.ones_like.default, aten.zeros_like.default,
+                 aten.empty_like.default, aten.full_like.default,
+                 aten.rand_like.default, aten.randn_like.default,
+                 aten.clone.default, aten.detach.default,
+                 aten._to_copy.default, aten.alias.default,
+                 aten.contiguous.default)
+def _like_rule(input_shapes, args, kwargs):
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(input_shapes[0])):
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs
+
+
+# -------------------------------------------------------------- embedding ----
+@register_preset(aten.embedding.default)
+def _embedding_rule(input_shapes, args, kwargs):
+    # inputs: (weight [V, D], indices [...])
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    # weight embedding-dim shard -> output last dim shard
+    if len(input_shapes[0]) == 2:
+        ann[0][1] = ShardDim.get_shard_dim(sid)
+        out_rank = len(input_shapes[1]) + 1
+        combs[sid] = _gather(out_rank - 1)
+        sid += 1
+    # indices dims shard -> output same dim shard
+    for d in range(len(input_shapes[1])):
+        ann[1][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.embedding_dense_backward.default)
+def _embedding_bwd_rule(input_shapes, args, kwargs):
+    # inputs: (grad_out [..., D], indices [...]); output [V, D]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    # shard grad_out feature dim -> output dim 1 shard
+    g_rank = len(input_shapes[0])
+    ann[0][g_rank - 1] = ShardDim.get_shard_dim(sid)
+    combs[sid] = _gather(1)
+    sid += 1
+    # shard batch dims of grad+indices together -> PARTIAL(sum) grad weight
+    for d in range(len(input_shapes[1])):
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        ann[1][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = functools.partial(CombinationFunc.reduce,
+                                       ops=__import__("operator").add)
+        sid += 1
+    return ann, combs

@@ -1,0 +1,93 @@
+"""Whole-graph tracing: fwd + bwd + optimizer step as ONE fx graph.
+
+Capability parity with reference ``easydist/torch/compile.py``
+(stateless_func / ed_compile_func, lines 25-120), re-designed for
+torch-2.10/ROCm: fused-optimizer decomposition happens at trace time and the
+resulting graph is made pure by the functionalize pass.
+"""
+from __future__ import annotations
+
+import logging
+from contextlib import nullcontext
+from functools import partial
+from typing import Callable, Optional
+
+import torch
+import torch.utils._pytree as pytree
+from torch._subclasses.fake_tensor import FakeTensor
+from torch.fx.experimental.proxy_tensor import make_fx
+from torch.nn.utils import stateless
+
+from ..utils import _enable_compile, _rematerialize_optimizer
+from .decomp import EASYDIST_DECOMP_TABLE
+
+logger = logging.getLogger(__name__)
+
+
+def stateless_func(func, module, opt, params, buffers, named_states, args,
+                   kwargs):
+    """Make the user's train_step pure in (params, buffers, states)."""
+    ctx1 = (stateless._reparametrize_module(module, {**params, **buffers},
+                                            tie_weights=True)
+            if module is not None else nullcontext())
+    ctx2 = (_rematerialize_optimizer(opt, named_states, params)
+            if opt is not None else nullcontext())
+    with ctx1, ctx2:
+        ret = func(*args, **kwargs)
+    grads = {k: v.grad for k, v in params.items()}
+    return params, buffers, named_states, grads, ret
+
+
+def warmup_optimizer(module: torch.nn.Module, opt) -> dict:
+    """Create optimizer states before tracing by running one zero-grad step.
+
+    (reference behavior: compile.py:50-66 — the warm-up step materializes
+    exp_avg etc.; the step counter is rewound by one so the traced graph's
+    increment reproduces the true first step when executed on live state.)
+    """
+    named_states = {}
+    if opt is None:
+        return named_states
+    params = dict(module.named_parameters())
+    with torch.no_grad():
+        snapshot = {n: p.detach().clone() for n, p in params.items()}
+        for p in params.values():
+            if p.grad is None:
+                p.grad = torch.zeros_like(p)
+    opt.step()
+    opt.zero_grad(True)
+    with torch.no_grad():
+        # the warm-up step is fake: undo any param drift (weight decay etc.)
+        for n, p in params.items():
+            p.copy_(snapshot[n])
+    for n, p in params.items():
+        if p in opt.state:
+            named_states[n] = dict(opt.state[p])
+            if "step" in named_states[n]:
+                named_states[n]["step"] = named_states[n]["step"] - 1
+    return named_states
+
+
+def ed_compile_func(func: Callable, tracing_mode: str, args, kwargs,
+                    module: Optional[torch.nn.Module], opt,
+                    split_patcher_ctx=None):
+    """Trace `func(module, opt, *args)` into one whole-step fx graph.
+
+    Returns (params, buffers, named_states, traced_graph).
+    """
+    params, buffers = {}, {}
+    if module is not None:
+        params = dict(module.named_parameters())
+        buffers = dict(module.named_buffers())
+    named_states = warmup_optimizer(module, opt) if opt is not None else {}
+
+    ctx = split_patcher_ctx if split_patcher_ctx is not None else nullcontext()
+    with _enable_compile(), ctx:
+        traced_graph = make_fx(partial(stateless_func, func, module, opt),
+                               tracing_mode=tracing_mode,
+                               decomposition_table=EASYDIST_DECOMP_TABLE,
+                               _allow_non_fake_inputs=False)(
+                                   params, buffers, named_states, args, kwargs)
+    traced_graph.graph.eliminate_dead_code()
+    traced_graph.recompile()
+    return params, buffers, named_states, traced_graph

@@ -46,6 +46,11 @@ class CombinationFunc:
         return functools.reduce(ops, shards)
 
     @staticmethod
+    def reduce_mean(shards: List[torch.Tensor]) -> torch.Tensor:
+        """Mean across equal-size shards (RCCL ReduceOp.AVG on the wire)."""
+        return functools.reduce(operator.add, shards) / len(shards)
+
+    @staticmethod
     def gather(shards: List[torch.Tensor], dim: int, halowidth: int = 0,
                chunk: int = 1) -> torch.Tensor:
         if halowidth == 0 and chunk == 1:
@@ -109,6 +114,13 @@ def _try_combination_single(shards: List[torch.Tensor],
                         return functools.partial(CombinationFunc.reduce, ops=op)
                 except Exception:
                     continue
+            try:
+                if (len({tuple(s.shape) for s in shards}) == 1
+                        and _allclose(CombinationFunc.reduce_mean(shards),
+                                      global_out)):
+                    return functools.partial(CombinationFunc.reduce_mean)
+            except Exception:
+                pass
 
     # gather candidates: dims where per-shard sizes sum to the global size
     if all(s.dim() == global_out.dim() for s in shards):

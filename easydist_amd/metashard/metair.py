@@ -184,6 +184,8 @@ class MetaNode:
                 return None
             if c.func is CombinationFunc.identity:
                 out_pl.append(R)
+            elif c.func is CombinationFunc.reduce_mean:
+                out_pl.append(P("avg"))
             elif c.func is CombinationFunc.reduce:
                 op = c.keywords.get("ops", operator.add)
                 name = {operator.add: "sum"}.get(op, None)

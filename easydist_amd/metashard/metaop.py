@@ -58,23 +58,28 @@ class MetaOp:
 
     def __init__(self, func: Callable, input_args: Tuple, kwargs: Optional[dict] = None,
                  name: str = ""):
+        import torch.utils._pytree as pytree
         self.func = func
-        self.input_args = input_args
         self.kwargs = kwargs or {}
         self.name = name or getattr(func, "__name__", str(func))
         # stencil ops (conv/pool family): on a plain failure retry the group
         # with halo-widened input shards (boundary overlap)
         self.try_halo = any(k in self.name for k in ("conv", "pool"))
-        # flat list of tensor inputs (what the annotation indexes)
+        # flat list of tensor inputs (what the annotation indexes) — pytree
+        # flattened so list-typed args (cat, stack, foreach) participate
+        self._flat_args, self._args_spec = pytree.tree_flatten(
+            (input_args, self.kwargs))
         self.flat_tensors: List[torch.Tensor] = [
-            a for a in input_args if isinstance(a, torch.Tensor)
+            a for a in self._flat_args if isinstance(a, torch.Tensor)
         ]
 
     def _call_with_tensors(self, tensors: List[torch.Tensor]):
+        import torch.utils._pytree as pytree
         it = iter(tensors)
-        args = tuple(next(it) if isinstance(a, torch.Tensor) else a
-                     for a in self.input_args)
-        return self.func(*args, **self.kwargs)
+        flat = [next(it) if isinstance(a, torch.Tensor) else a
+                for a in self._flat_args]
+        args, kwargs = pytree.tree_unflatten(flat, self._args_spec)
+        return self.func(*args, **kwargs)
 
     def exec_global(self):
         return self._call_with_tensors(self.flat_tensors)

@@ -1,0 +1,192 @@
+"""EDCompiledFunc: execute the sharded graph with persistent local state.
+
+Capability parity with reference EDCompiledFunc (compile_auto.py:720-822):
+holds the local (pre-sharded) params/buffers/optimizer states, runs the
+transformed fx graph each step, writes updated state back into its
+persistent buffers, and optionally replays the whole step as ONE hipGraph
+(torch.cuda.CUDAGraph on ROCm) after warmup.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.utils._pytree as pytree
+
+from .. import config as mdconfig
+from ..metashard.metair import SPMD
+from ..parallel import comm
+from ..parallel.device_mesh import get_device_mesh
+
+logger = logging.getLogger(__name__)
+
+
+def shard_tensor_local(t: torch.Tensor, placements: List[SPMD],
+                       mesh) -> torch.Tensor:
+    """Apply a placement vector: outer mesh dims chunk first."""
+    out = t
+    for d, p in enumerate(placements):
+        if p.is_shard():
+            out = comm.local_chunk(out, p.dim, mesh.get_group(d))
+        elif p.is_partial():
+            out = comm.partial_localize(out, mesh.get_group(d))
+    return out
+
+
+def unshard_tensor(t: torch.Tensor, placements: List[SPMD], mesh) -> torch.Tensor:
+    """Gather a local shard back to the global tensor (for state_dict)."""
+    out = t
+    for d in reversed(range(len(placements))):
+        p = placements[d]
+        if p.is_shard():
+            out = comm.all_gather(out, p.dim, mesh.get_group(d))
+        elif p.is_partial():
+            out = comm.all_reduce(out, p.reduce_op or "sum", mesh.get_group(d))
+    return out
+
+
+class EDCompiledFunc:
+    def __init__(self, gm, in_spec, out_spec, input_placements,
+                 output_placements, state_input_positions,
+                 io_pos_map: Dict[int, int], param_count: int,
+                 flat_param_names: List[str], device: str):
+        self.gm = gm
+        self.in_spec = in_spec
+        self.out_spec = out_spec
+        # per flat-input position: placement vector (or None for non-tensors)
+        self.input_placements = input_placements
+        self.output_placements = output_placements
+        # positions in the flat input list that are params/buffers/states
+        self.state_input_positions = state_input_positions
+        # flat input pos -> flat output pos (state round trip)
+        self.io_pos_map = io_pos_map
+        self.param_count = param_count
+        self.flat_param_names = flat_param_names
+        self.device = device
+        # persistent local state, keyed by flat input position
+        self.state: Dict[int, torch.Tensor] = {}
+        self._graph: Optional[torch.cuda.CUDAGraph] = None
+        self._graph_args_buf: List[torch.Tensor] = []
+        self._graph_out = None
+
+    # ------------------------------------------------------------- state -----
+    def init_state(self, flat_inputs: List):
+        mesh = get_device_mesh()
+        for pos in self.state_input_positions:
+            t = flat_inputs[pos]
+            if not isinstance(t, torch.Tensor):
+                continue
+            pl = self.input_placements[pos]
+            local = shard_tensor_local(t.detach().to(self.device), pl, mesh)
+            self.state[pos] = local.clone()
+
+    def _prepare_inputs(self, flat_inputs: List) -> List:
+        mesh = get_device_mesh()
+        prepared = list(flat_inputs)
+        for pos, val in enumerate(prepared):
+            if pos in self.state:
+                prepared[pos] = self.state[pos]
+            elif isinstance(val, torch.Tensor):
+                pl = self.input_placements[pos]
+                t = val.to(self.device) if val.device.type != self.device.split(":")[0] else val
+                if pl is not None:
+                    t = shard_tensor_local(t, pl, mesh)
+                prepared[pos] = t
+        return prepared
+
+    def _writeback(self, flat_outs: List):
+        for in_pos, out_pos in self.io_pos_map.items():
+            new = flat_outs[out_pos]
+            if isinstance(new, torch.Tensor) and in_pos in self.state:
+                buf = self.state[in_pos]
+                if buf.shape == new.shape and buf.dtype == new.dtype:
+                    buf.copy_(new)
+                else:
+                    self.state[in_pos] = new
+
+    # -------------------------------------------------------------- call -----
+    def __call__(self, *args, **kwargs):
+        flat_inputs, spec = pytree.tree_flatten(args[0]) if len(args) == 1 and isinstance(args[0], list) else (None, None)
+        raise RuntimeError("use run(flat_inputs)")
+
+    def run(self, flat_inputs: List):
+        if not self.state:
+            self.init_state(flat_inputs)
+        prepared = self._prepare_inputs(flat_inputs)
+        flat_outs = self.gm(*prepared)
+        self._writeback(flat_outs)
+        return flat_outs
+
+    # ------------------------------------------------- state access APIs ----
+    def named_parameters(self) -> Dict[str, torch.Tensor]:
+        """Gather the (sharded) params back to global tensors."""
+        mesh = get_device_mesh()
+        out = {}
+        for i, name in enumerate(self.flat_param_names):
+            if i in self.state:
+                pl = self.input_placements[i] or []
+                out[name] = unshard_tensor(self.state[i], pl, mesh)
+        return out
+
+    def get_state(self):
+        """All persistent state (params/buffers/opt states), gathered."""
+        mesh = get_device_mesh()
+        out = {}
+        for pos, t in self.state.items():
+            pl = self.input_placements[pos] or []
+            out[pos] = unshard_tensor(t, pl, mesh)
+        return out
+
+    def local_state(self):
+        return dict(self.state)
+
+    def run_graph(self, flat_inputs: List):
+        """hipGraph capture + replay (torch.cuda.CUDAGraph is hipGraph on
+        ROCm). Non-state inputs are copied into static buffers each step."""
+        if not torch.cuda.is_available():
+            return self.run(flat_inputs)
+        if self._graph is None:
+            if not self.state:
+                self.init_state(flat_inputs)
+            prepared = self._prepare_inputs(flat_inputs)
+            # static buffers for the dynamic (non-state) tensor inputs
+            self._dyn_pos = [i for i, v in enumerate(prepared)
+                             if isinstance(v, torch.Tensor)
+                             and i not in self.state]
+            self._static = {i: prepared[i].clone() for i in self._dyn_pos}
+            for i in self._dyn_pos:
+                prepared[i] = self._static[i]
+            # warmup on a side stream (RCCL + hipGraph requirement)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    outs = self.gm(*prepared)
+                    self._writeback(outs)
+            torch.cuda.current_stream().wait_stream(s)
+            self._graph = torch.cuda.CUDAGraph()
+            self._graph_prepared = prepared
+            with torch.cuda.graph(self._graph):
+                outs = self.gm(*prepared)
+                # writeback inside the graph: copy into persistent state
+                for in_pos, out_pos in self.io_pos_map.items():
+                    new = outs[out_pos]
+                    if isinstance(new, torch.Tensor) and in_pos in self.state:
+                        buf = self.state[in_pos]
+                        if buf.shape == new.shape and buf.dtype == new.dtype:
+                            buf.copy_(new)
+                self._graph_out = outs
+            return self._graph_out
+        # replay path
+        for i in self._dyn_pos:
+            v = flat_inputs[i]
+            if isinstance(v, torch.Tensor):
+                mesh = get_device_mesh()
+                pl = self.input_placements[i]
+                t = v.to(self.device)
+                if pl is not None:
+                    t = shard_tensor_local(t, pl, mesh)
+                self._static[i].copy_(t)
+        self._graph.replay()
+        return self._graph_out
