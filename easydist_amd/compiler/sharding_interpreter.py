@@ -26,14 +26,18 @@ logger = logging.getLogger(__name__)
 _DISCOVERY_CACHE: Dict[Tuple, Tuple] = {}
 
 
-def _to_real(meta: torch.Tensor, device=None) -> torch.Tensor:
-    """Materialize a random tensor matching a FakeTensor's meta."""
+def _to_real(meta: torch.Tensor, device=None, promote_fp64=True) -> torch.Tensor:
+    """Materialize a random tensor matching a FakeTensor's meta.
+
+    Floats materialize as fp64 by default: discovery compares shard
+    recombinations against the global output at ~1e-6 relative, which only
+    separates true rules from coincidences at fp64 precision.
+    """
     device = device or meta.device
     if meta.dtype.is_floating_point:
-        # [0.5, 1.5): well-conditioned for discovery (no near-zero divides,
-        # no catastrophic cancellation in sums)
-        t = torch.rand(meta.shape, dtype=torch.float32, device=device) + 0.5
-        return t.to(meta.dtype)
+        dt = torch.float64 if promote_fp64 else meta.dtype
+        t = torch.rand(meta.shape, dtype=torch.float64, device=device) + 0.5
+        return t.to(dt)
     if meta.dtype == torch.bool:
         return torch.rand(meta.shape, device=device) > 0.5
     # integer tensors: small non-negative values (safe for index ops)
@@ -105,16 +109,26 @@ class EDTorchShardingAnn:
         return info
 
     def _discover(self, node: fx.Node):
-        def realize(a):
-            if isinstance(a, fx.Node):
-                v = a.meta.get("val")
-                if isinstance(v, torch.Tensor):
-                    return _to_real(v, self.device)
-                return v
-            return a
-        args = pytree.tree_map(
-            lambda a: realize(a) if isinstance(a, fx.Node) else a, node.args)
-        kwargs = pytree.tree_map(
-            lambda a: realize(a) if isinstance(a, fx.Node) else a, node.kwargs)
-        op = MetaOp(node.target, args, kwargs, name=str(node.target))
-        return op.sharding_discovery()
+        def make(promote):
+            def realize(a):
+                if isinstance(a, fx.Node):
+                    v = a.meta.get("val")
+                    if isinstance(v, torch.Tensor):
+                        return _to_real(v, self.device, promote)
+                    return v
+                return a
+            args = pytree.tree_map(
+                lambda a: realize(a) if isinstance(a, fx.Node) else a,
+                node.args)
+            kwargs = pytree.tree_map(
+                lambda a: realize(a) if isinstance(a, fx.Node) else a,
+                node.kwargs)
+            return MetaOp(node.target, args, kwargs, name=str(node.target))
+
+        op = make(True)
+        try:
+            op.exec_global()
+            return op.sharding_discovery()
+        except Exception:
+            # op rejects fp64 (fused kernels etc.): original dtype + loose tol
+            return make(False).sharding_discovery()

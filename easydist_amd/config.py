@@ -50,8 +50,15 @@ use_hint = False
 discovery_max_dim = _env_int("EASYDIST_DISCOVERY_MAX_DIM", 1024)
 # number of shards used while searching for rules (not the mesh size)
 discovery_num_shards = _env_int("EASYDIST_DISCOVERY_NUM_SHARDS", 2)
-discovery_rtol = _env_float("EASYDIST_DISCOVERY_RTOL", 1e-3)
-discovery_atol = _env_float("EASYDIST_DISCOVERY_ATOL", 1e-4)
+# Discovery runs in fp64: a TRUE rule reproduces the global output to
+# ~1e-15 relative (just reduction reordering), while a numeric coincidence
+# (e.g. shard-mean ~= global-mean on iid data) differs by ~1e-3. The tight
+# tolerance is what separates them.
+discovery_rtol = _env_float("EASYDIST_DISCOVERY_RTOL", 1e-6)
+discovery_atol = _env_float("EASYDIST_DISCOVERY_ATOL", 1e-9)
+# fallback tolerances when an op cannot execute in fp64
+discovery_rtol_lowprec = _env_float("EASYDIST_DISCOVERY_RTOL_LP", 1e-2)
+discovery_atol_lowprec = _env_float("EASYDIST_DISCOVERY_ATOL_LP", 1e-2)
 max_halo = _env_int("EASYDIST_MAX_HALO", 3)
 
 # ----------------------------------------------------------------- solver ----

@@ -28,10 +28,13 @@ def _allclose(a: torch.Tensor, b: torch.Tensor) -> bool:
         return False
     if not a.is_floating_point():
         return bool(torch.equal(a, b))
-    a32 = a.float()
-    b32 = b.float()
-    return bool(torch.allclose(a32, b32, rtol=mdconfig.discovery_rtol,
-                               atol=mdconfig.discovery_atol, equal_nan=True))
+    if a.dtype == torch.float64:
+        rtol, atol = mdconfig.discovery_rtol, mdconfig.discovery_atol
+    else:
+        rtol = mdconfig.discovery_rtol_lowprec
+        atol = mdconfig.discovery_atol_lowprec
+    return bool(torch.allclose(a.double(), b.double(), rtol=rtol, atol=atol,
+                               equal_nan=True))
 
 
 class CombinationFunc:
