@@ -155,6 +155,13 @@ class MetaNode:
                     if not dim_ok(self.invars[i], d):
                         ok = False
                         break
+                    # halo-carrying groups need boundary exchange the plain
+                    # SHARD placement doesn't express; they are kept in the
+                    # annotation (conv spatial sharding) but not offered to
+                    # the solver until halo reshard lands
+                    if self.sharding_ann[i][d].halo != 0:
+                        ok = False
+                        break
                     in_pl[i] = S(d)
                 if not ok:
                     continue

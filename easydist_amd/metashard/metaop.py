@@ -123,8 +123,18 @@ class MetaOp:
             return comb
         if not allow_halo or not (halo_hinted or self.try_halo):
             return None
+        # halo only makes sense when it is smaller than a shard: otherwise
+        # every "shard" degenerates to (nearly) the whole tensor and the
+        # identity check passes vacuously
+        min_base = min(
+            (self.flat_tensors[i].shape[d] // num_shards
+             for i, dims in enumerate(annotation.annotation)
+             for d, sd in enumerate(dims) if sd.shard_dim_id == shard_dim_id),
+            default=0)
         # retry with halo-widened input shards
         for width in range(1, mdconfig.max_halo + 1):
+            if width >= min_base:
+                break
             ann_h = annotation.inject_haloinfo(width, shard_dim_id)
             try:
                 sharded = self.exec_sharded(ann_h, shard_dim_id, num_shards)
@@ -183,6 +193,24 @@ class MetaOp:
                     if comb is not None:
                         found_comb, group_ann = comb, trial2
                         break
+                if found_comb is None:
+                    # 3-way+ groups (e.g. batched attention: q,k,v batch dims
+                    # must shard together): tag the SAME dim in every input
+                    # that has it with matching size
+                    trial3 = ann.copy()
+                    trial3[i][d] = ShardDim.get_shard_dim(next_id)
+                    tags = 1
+                    for j in range(len(shapes)):
+                        if j == i or d >= len(shapes[j]):
+                            continue
+                        if shapes[j][d] == shapes[i][d] and shardable(j, d):
+                            trial3[j][d] = ShardDim.get_shard_dim(next_id)
+                            tags += 1
+                    if tags >= 3:
+                        comb = self._group_works(trial3, next_id, num_shards,
+                                                 global_out)
+                        if comb is not None:
+                            found_comb, group_ann = comb, trial3
             if found_comb is None:
                 continue
             # 3) greedy extension with remaining dims (one per input)

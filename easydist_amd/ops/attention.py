@@ -1,0 +1,123 @@
+"""Flash attention as a custom op: HIP kernel forward on gfx950.
+
+Forward: hand-written CDNA4 flash kernel (csrc/attn_fwd.hip) — online
+softmax, MFMA 16x16x32 bf16 tiles, K/V staged through LDS with XOR swizzle.
+Backward: recompute-based composite (P from saved LSE + batched GEMMs on
+hipBLASLt) — the classic flash backward; the fused HIP backward is a later
+optimization.
+
+CPU fallback implements the same math in aten for tests.
+"""
+from __future__ import annotations
+
+import math
+from typing import Tuple
+
+import torch
+
+from . import load_extension
+
+lib = torch.library.Library("easydist_amd", "DEF")
+lib.define("flash_attention(Tensor q, Tensor k, Tensor v, bool causal) "
+           "-> (Tensor, Tensor)")
+lib.define("flash_attention_bwd(Tensor grad, Tensor q, Tensor k, Tensor v, "
+           "Tensor out, Tensor lse, bool causal) -> (Tensor, Tensor, Tensor)")
+
+
+def _math_fwd(q, k, v, causal):
+    # q,k,v: [B, H, S, D]
+    scale = 1.0 / math.sqrt(q.shape[-1])
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        S, T = s.shape[-2], s.shape[-1]
+        mask = torch.ones(S, T, dtype=torch.bool, device=s.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    lse = torch.logsumexp(s, dim=-1)
+    p = torch.exp(s - lse.unsqueeze(-1))
+    out = torch.matmul(p.to(v.dtype), v)
+    return out, lse
+
+
+def _math_bwd(grad, q, k, v, out, lse, causal):
+    scale = 1.0 / math.sqrt(q.shape[-1])
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        S, T = s.shape[-2], s.shape[-1]
+        mask = torch.ones(S, T, dtype=torch.bool, device=s.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    p = torch.exp(s - lse.unsqueeze(-1).float())
+    gradf = grad.float()
+    dv = torch.matmul(p.transpose(-1, -2), gradf)
+    dp = torch.matmul(gradf, v.float().transpose(-1, -2))
+    d = (gradf * out.float()).sum(-1, keepdim=True)
+    ds = p * (dp - d) * scale
+    dq = torch.matmul(ds, k.float())
+    dk = torch.matmul(ds.transpose(-1, -2), q.float())
+    return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype)
+
+
+def _fwd_cpu(q, k, v, causal):
+    return _math_fwd(q, k, v, causal)
+
+
+def _fwd_cuda(q, k, v, causal):
+    ext = load_extension()
+    if ext is not None and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        q, k, v = (t.contiguous() for t in (q, k, v))
+        return ext.flash_attn_fwd(q, k, v, causal)
+    from . import require_hip_ops
+    if q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        require_hip_ops()   # raises when the extension should exist
+    return _math_fwd(q, k, v, causal)
+
+
+def _bwd_cpu(grad, q, k, v, out, lse, causal):
+    return _math_bwd(grad, q, k, v, out, lse, causal)
+
+
+def _bwd_cuda(grad, q, k, v, out, lse, causal):
+    return _math_bwd(grad, q, k, v, out, lse, causal)
+
+
+lib.impl("flash_attention", _fwd_cpu, "CPU")
+lib.impl("flash_attention", _fwd_cuda, "CUDA")
+lib.impl("flash_attention_bwd", _bwd_cpu, "CPU")
+lib.impl("flash_attention_bwd", _bwd_cuda, "CUDA")
+
+
+@torch.library.register_fake("easydist_amd::flash_attention")
+def _fa_fake(q, k, v, causal):
+    lse = q.new_empty(q.shape[:-1], dtype=torch.float32)
+    # the kernel returns a CONTIGUOUS [B,H,S,D] tensor regardless of input
+    # strides — the fake must match or view/reshape traces diverge
+    return q.new_empty(tuple(q.shape)), lse
+
+
+@torch.library.register_fake("easydist_amd::flash_attention_bwd")
+def _fab_fake(grad, q, k, v, out, lse, causal):
+    # contiguous outputs (matmul results), independent of input strides
+    return (q.new_empty(tuple(q.shape)), k.new_empty(tuple(k.shape)),
+            v.new_empty(tuple(v.shape)))
+
+
+def _fa_backward(ctx, grad_out, grad_lse):
+    q, k, v, out, lse = ctx.saved_tensors
+    dq, dk, dv = torch.ops.easydist_amd.flash_attention_bwd(
+        grad_out, q, k, v, out, lse, ctx.causal)
+    return dq, dk, dv, None
+
+
+def _fa_setup_ctx(ctx, inputs, output):
+    q, k, v, causal = inputs
+    out, lse = output
+    ctx.save_for_backward(q, k, v, out, lse)
+    ctx.causal = causal
+
+
+torch.library.register_autograd("easydist_amd::flash_attention", _fa_backward,
+                                setup_context=_fa_setup_ctx)
+
+
+def scaled_dot_product_attention(q, k, v, causal: bool = True):
+    out, _ = torch.ops.easydist_amd.flash_attention(q, k, v, causal)
+    return out
