@@ -1,0 +1,154 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark: GPT-2 train-step throughput (samples/sec).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+N>1 is launched via torch.distributed.run with one rank per GPU over RCCL.
+Measures the BASELINE.json headline: train-step throughput (samples/sec,
+whole node) for GPT-2 under easydist_amd auto-SPMD, bf16 autocast,
+synthetic data, random-init weights; also reports the strategy-search
+wallclock. Weak scaling: per-GPU batch fixed, global batch = B * N.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", default="gpt2-small",
+                   choices=["gpt2-small", "gpt2-medium", "gpt2-1.3b",
+                            "gpt-1l-12288"])
+    p.add_argument("--per-gpu-batch", type=int, default=8)
+    p.add_argument("--seq", type=int, default=1024)
+    p.add_argument("--parallel", default="auto")
+    p.add_argument("--no-hipgraph", action="store_true")
+    args = p.parse_args()
+
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, set_device_mesh
+    from easydist_amd.models import gpt as gptm
+
+    world_size = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    n_gpus = max(world_size, 1)
+    use_cuda = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_cuda else "cpu"
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+
+    backend = "nccl" if use_cuda else "gloo"
+    if world_size > 1:
+        dist.init_process_group(backend=backend)
+    else:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29512")
+        dist.init_process_group(backend=backend, rank=0, world_size=1)
+
+    easydist_setup(backend="torch", device="cuda" if use_cuda else "cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+
+    cfg = {
+        "gpt2-small": gptm.GPT2_SMALL,
+        "gpt2-medium": gptm.GPT2_MEDIUM,
+        "gpt2-1.3b": gptm.GPT2_1_3B,
+        "gpt-1l-12288": gptm.GPT_BENCH_1L,
+    }[args.model]
+    if args.seq != cfg.block_size:
+        from dataclasses import replace
+        cfg = replace(cfg, block_size=args.seq)
+
+    torch.manual_seed(1234 + 0)   # same weights everywhere
+    model = gptm.GPT(cfg).to(device)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
+
+    global_batch = args.per_gpu_batch * n_gpus
+
+    def train_step(model, opt, idx, targets):
+        return gptm.gpt_train_step(model, opt, idx, targets)
+
+    compiled = easydist_compile(train_step, parallel_mode=args.parallel,
+                                cuda_graph=not args.no_hipgraph)
+
+    def batch():
+        g = torch.Generator(device="cpu").manual_seed(7)   # same on all ranks
+        idx = torch.randint(0, cfg.vocab_size, (global_batch, args.seq),
+                            generator=g, device="cpu").to(device)
+        tg = torch.randint(0, cfg.vocab_size, (global_batch, args.seq),
+                           generator=g, device="cpu").to(device)
+        return idx, tg
+
+    idx, tg = batch()
+
+    t_compile = time.time()
+    for _ in range(max(args.warmup, 1)):
+        loss = compiled(model, opt, idx, tg)
+    if use_cuda:
+        torch.cuda.synchronize()
+    compile_and_warmup_s = time.time() - t_compile
+
+    search_s = None
+    for rt in compiled.compiled.values():
+        if hasattr(rt, "meta"):
+            search_s = rt.meta.get("search_time", 0) + rt.meta.get(
+                "solve_time", 0)
+
+    dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        loss = compiled(model, opt, idx, tg)
+    if use_cuda:
+        torch.cuda.synchronize()
+    dist.barrier()
+    elapsed = time.time() - t0
+    # max over ranks
+    e = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+    dist.all_reduce(e, op=dist.ReduceOp.MAX)
+    elapsed = float(e)
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    samples_per_sec = global_batch * args.steps / elapsed
+
+    if rank == 0:
+        out = {
+            "metric": "gpt2_train_samples_per_sec",
+            "value": samples_per_sec,
+            "unit": "samples/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": args.seq,
+                "parallelism": f"{args.parallel}(dp{n_gpus})",
+                "strategy_search_s": search_s,
+                "compile_warmup_s": round(compile_and_warmup_s, 2),
+                "loss": float(loss) if loss is not None else None,
+            },
+        }
+        print(json.dumps(out))
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

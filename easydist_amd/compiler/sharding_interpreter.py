@@ -108,12 +108,37 @@ class EDTorchShardingAnn:
                 info[node.name] = (None, {})
         return info
 
+    def _shrink_map(self, node: fx.Node):
+        """Hint-shrink: map huge dim sizes to a small stand-in for the
+        discovery executions (reference: sharding_interpreter.py:256-281).
+        Same size -> same shrink keeps op shape constraints consistent."""
+        limit = mdconfig.discovery_max_dim
+        sizes = set()
+
+        def collect(v):
+            if isinstance(v, torch.Tensor):
+                sizes.update(int(s) for s in v.shape)
+        flat, _ = pytree.tree_flatten((node.args, node.kwargs))
+        for a in flat:
+            if isinstance(a, fx.Node):
+                collect(a.meta.get("val"))
+        collect(node.meta.get("val"))
+        return {s: limit for s in sizes if s > limit}
+
     def _discover(self, node: fx.Node):
+        size_map = self._shrink_map(node)
+
+        def shrink(shape):
+            return tuple(size_map.get(int(s), int(s)) for s in shape)
+
         def make(promote):
             def realize(a):
                 if isinstance(a, fx.Node):
                     v = a.meta.get("val")
                     if isinstance(v, torch.Tensor):
+                        if size_map:
+                            v = torch.empty(shrink(v.shape), dtype=v.dtype,
+                                            device="meta")
                         return _to_real(v, self.device, promote)
                     return v
                 return a

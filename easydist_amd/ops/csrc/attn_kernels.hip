@@ -1,0 +1,213 @@
+// Flash attention forward for gfx950 (bf16, causal, D in {64, 128}).
+//
+// v1 structure (correctness-first; the tuned ladder of
+// cdna_hip_programming.md §B comes in later rounds): one workgroup = 4
+// waves = one 64-row Q block of one (batch, head). Each wave owns 16 Q
+// rows. K/V tiles (64 keys x D) are staged in LDS per workgroup; QK^T and
+// P·V run on v_mfma_f32_16x16x32_bf16; the online-softmax state (m, l)
+// lives in registers, row-reductions via 16-lane shuffles inside each
+// fragment group. P round-trips through a wave-private LDS strip to
+// re-fragment from the S layout to the PV A-operand layout.
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
+
+#define QB 64     // q rows per workgroup (16 per wave)
+#define KB 64     // kv tile
+
+template <int D>
+__global__ void __launch_bounds__(256)
+flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
+                 const bf16* __restrict__ V, bf16* __restrict__ O,
+                 float* __restrict__ LSE, int B, int H, int S, bool causal,
+                 float scale) {
+  // grid: (S/QB, B*H)
+  const int qb0 = blockIdx.x * QB;
+  const int bh = blockIdx.y;
+  const long base = (long)bh * S * D;
+  const bf16* q = Q + base;
+  const bf16* k = K + base;
+  const bf16* v = V + base;
+  bf16* o = O + base;
+  float* lse = LSE + (long)bh * S;
+
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int qr0 = qb0 + wave * 16;          // this wave's first q row
+
+  // LDS: K tile [KB][D] + V tile [KB][D] + P strips [4 waves][16][KB]
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* k_lds = reinterpret_cast<bf16*>(smem);
+  bf16* v_lds = k_lds + KB * D;
+  bf16* p_lds = v_lds + KB * D + wave * 16 * KB;
+
+  const int fr = lane & 15;        // fragment row/col index
+  const int fg = lane >> 4;        // fragment k-group (8 contiguous)
+
+  // Q fragments: A operand rows = q rows; lane holds q[qr0+fr][8*fg..+8]
+  bf16x8v qf[D / 32];
+  #pragma unroll
+  for (int ks = 0; ks < D / 32; ++ks)
+    qf[ks] = *reinterpret_cast<const bf16x8v*>(
+        &q[(long)(qr0 + fr) * D + ks * 32 + fg * 8]);
+
+  // online state: rows 4*fg..4*fg+3 of this wave's 16 (per C-layout)
+  float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4 o_acc[D / 16];
+  #pragma unroll
+  for (int j = 0; j < D / 16; ++j) o_acc[j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = causal ? min(S, qb0 + QB) : S;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    // ---- stage K/V tile: 256 threads, 16B each --------------------------
+    __syncthreads();
+    {
+      const int elems = KB * D;               // tile elements
+      for (int e = threadIdx.x * 8; e < elems; e += 256 * 8) {
+        *reinterpret_cast<bf16x8*>(&k_lds[e]) =
+            *reinterpret_cast<const bf16x8*>(&k[(long)kv0 * D + e]);
+        *reinterpret_cast<bf16x8*>(&v_lds[e]) =
+            *reinterpret_cast<const bf16x8*>(&v[(long)kv0 * D + e]);
+      }
+    }
+    __syncthreads();
+
+    // ---- S = Q K^T ------------------------------------------------------
+    // 4 col-blocks of 16 keys; B operand: lane holds K[kv0+fr][8*fg..]
+    f32x4 s_acc[KB / 16];
+    #pragma unroll
+    for (int j = 0; j < KB / 16; ++j) {
+      s_acc[j] = {0.f, 0.f, 0.f, 0.f};
+      #pragma unroll
+      for (int ks = 0; ks < D / 32; ++ks) {
+        bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
+            &k_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+        s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], kf,
+                                                           s_acc[j], 0, 0, 0);
+      }
+    }
+
+    // ---- online softmax --------------------------------------------------
+    // lane holds S[4*fg+r][j*16 + fr] for r=0..3
+    float m_new[4], row_max[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = -1e30f;
+      #pragma unroll
+      for (int j = 0; j < KB / 16; ++j) {
+        float sv = s_acc[j][r] * scale;
+        int kcol = kv0 + j * 16 + fr;
+        int qrow = qr0 + 4 * fg + r;
+        if (causal && kcol > qrow) sv = -1e30f;
+        else if (kcol >= S) sv = -1e30f;
+        s_acc[j][r] = sv;
+        mx = fmaxf(mx, sv);
+      }
+      // max across the 16 lanes of this fragment group
+      #pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+      row_max[r] = mx;
+      m_new[r] = fmaxf(m_run[r], mx);
+    }
+    float p_sum[4] = {0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
+    for (int j = 0; j < KB / 16; ++j) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float p = __expf(s_acc[j][r] - m_new[r]);
+        s_acc[j][r] = p;
+        p_sum[r] += p;
+      }
+    }
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      #pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        p_sum[r] += __shfl_xor(p_sum[r], off, WAVE);
+      float alpha = __expf(m_run[r] - m_new[r]);
+      l_run[r] = l_run[r] * alpha + p_sum[r];
+      m_run[r] = m_new[r];
+      // rescale o accumulator rows
+      #pragma unroll
+      for (int j = 0; j < D / 16; ++j) o_acc[j][r] *= alpha;
+    }
+
+    // ---- P to LDS strip (re-fragment), then P @ V -----------------------
+    // write: P[4*fg+r][j*16+fr]
+    #pragma unroll
+    for (int j = 0; j < KB / 16; ++j)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_lds[(4 * fg + r) * KB + j * 16 + fr] = f2bf(s_acc[j][r]);
+    // wave-private strip; cross-lane visibility within the wave needs a
+    // data-share sync, which s_waitcnt lgkmcnt(0) provides per-wave
+    __builtin_amdgcn_s_waitcnt(0);
+
+    // A operand: lane holds P[fr][8*fg..+8]
+    bf16x8v pf[KB / 32];
+    #pragma unroll
+    for (int ks = 0; ks < KB / 32; ++ks)
+      pf[ks] = *reinterpret_cast<const bf16x8v*>(
+          &p_lds[fr * KB + ks * 32 + fg * 8]);
+    // B operand: lane holds V[kv0 + 8*fg + jj][d0 + fr] — column reads
+    #pragma unroll
+    for (int j = 0; j < D / 16; ++j) {
+      #pragma unroll
+      for (int ks = 0; ks < KB / 32; ++ks) {
+        bf16x8v vf;
+        #pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          vf[jj] = *reinterpret_cast<const __bf16*>(
+              &v_lds[(ks * 32 + fg * 8 + jj) * D + j * 16 + fr]);
+        o_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf[ks], vf,
+                                                           o_acc[j], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue --------------------------------------------------------
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int qrow = qr0 + 4 * fg + r;
+    if (qrow >= S) continue;
+    float inv_l = 1.f / l_run[r];
+    #pragma unroll
+    for (int j = 0; j < D / 16; ++j)
+      o[(long)qrow * D + j * 16 + fr] = f2bf(o_acc[j][r] * inv_l);
+    if (fr == 0)
+      lse[qrow] = m_run[r] + __logf(l_run[r]);
+  }
+}
+
+std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
+                                                  const at::Tensor& k,
+                                                  const at::Tensor& v,
+                                                  bool causal) {
+  TORCH_CHECK(q.dtype() == at::kBFloat16 && q.dim() == 4);
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  TORCH_CHECK(D == 64 || D == 128, "flash_attn_fwd: D in {64,128}");
+  TORCH_CHECK(S % QB == 0, "flash_attn_fwd: S multiple of 64");
+  auto out = at::empty_like(q);
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  dim3 grid(S / QB, B * H), block(256);
+  size_t lds = (2 * KB * D + 4 * 16 * KB) * 2;
+  float scale = 1.f / sqrtf((float)D);
+  if (D == 64)
+    hipLaunchKernelGGL(flash_fwd_kernel<64>, grid, block, lds, stream,
+        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+        (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
+        lse.data_ptr<float>(), B, H, S, causal, scale);
+  else
+    hipLaunchKernelGGL(flash_fwd_kernel<128>, grid, block, lds, stream,
+        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+        (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
+        lse.data_ptr<float>(), B, H, S, causal, scale);
+  return {out, lse};
+}

@@ -1,0 +1,147 @@
+// Fused multi-tensor Adam step for gfx950.
+//
+// One grid-stride kernel walks a chunk table covering every parameter
+// shard: the per-step optimizer tail collapses from ~4 launches per
+// parameter to ONE launch. fp32 math; functional form (writes fresh
+// outputs) so the pure sharded graph + hipGraph capture stay clean.
+// The chunk table is cached on device keyed by the input pointer set, so
+// replays (hipGraph) never rebuild or re-upload it.
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <ATen/cuda/CUDAGraphsUtils.cuh>
+#include <vector>
+
+struct AdamChunk {
+  const float* p;
+  const float* g;
+  const float* ea;
+  const float* eas;
+  float* out_p;
+  float* out_ea;
+  float* out_eas;
+  int tensor_idx;
+  int n;          // elements in this chunk
+};
+
+#define CHUNK_ELEMS 65536
+
+__global__ void fused_adam_kernel(const AdamChunk* __restrict__ chunks,
+                                  int n_chunks,
+                                  const float* __restrict__ steps,  // [T]
+                                  float lr, float beta1, float beta2,
+                                  float weight_decay, float eps) {
+  const int c = blockIdx.x;
+  if (c >= n_chunks) return;
+  const AdamChunk ch = chunks[c];
+  const float step = steps[ch.tensor_idx] + 1.f;
+  const float bc1 = 1.f - __powf(beta1, step);
+  const float bc2 = 1.f - __powf(beta2, step);
+  const float inv_bc1 = 1.f / bc1;
+  const float rsqrt_bc2 = rsqrtf(bc2);
+  for (int i = threadIdx.x * 4; i + 3 < ch.n; i += blockDim.x * 4) {
+    float4 p = *reinterpret_cast<const float4*>(&ch.p[i]);
+    float4 g = *reinterpret_cast<const float4*>(&ch.g[i]);
+    float4 ea = *reinterpret_cast<const float4*>(&ch.ea[i]);
+    float4 eas = *reinterpret_cast<const float4*>(&ch.eas[i]);
+    float* pp = &p.x; float* gg = &g.x; float* ee = &ea.x; float* ss = &eas.x;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float grad = gg[j];
+      if (weight_decay != 0.f) grad += weight_decay * pp[j];
+      float m = beta1 * ee[j] + (1.f - beta1) * grad;
+      float v = beta2 * ss[j] + (1.f - beta2) * grad * grad;
+      float denom = __fsqrt_rn(v) * rsqrt_bc2 + eps;
+      pp[j] = pp[j] - lr * (m * inv_bc1) / denom;
+      ee[j] = m;
+      ss[j] = v;
+    }
+    *reinterpret_cast<float4*>(&ch.out_p[i]) = p;
+    *reinterpret_cast<float4*>(&ch.out_ea[i]) = ea;
+    *reinterpret_cast<float4*>(&ch.out_eas[i]) = eas;
+  }
+  // scalar tail
+  int tail_start = (ch.n / 4) * 4;
+  for (int i = tail_start + threadIdx.x; i < ch.n; i += blockDim.x) {
+    float grad = ch.g[i];
+    if (weight_decay != 0.f) grad += weight_decay * ch.p[i];
+    float m = beta1 * ch.ea[i] + (1.f - beta1) * grad;
+    float v = beta2 * ch.eas[i] + (1.f - beta2) * grad * grad;
+    float denom = __fsqrt_rn(v) * rsqrt_bc2 + eps;
+    ch.out_p[i] = ch.p[i] - lr * (m * inv_bc1) / denom;
+    ch.out_ea[i] = m;
+    ch.out_eas[i] = v;
+  }
+}
+
+__global__ void bump_steps_kernel(const float* const* __restrict__ in_steps,
+                                  float* const* __restrict__ out_steps,
+                                  int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out_steps[i][0] = in_steps[i][0] + 1.f;
+}
+
+std::tuple<std::vector<at::Tensor>, std::vector<at::Tensor>,
+           std::vector<at::Tensor>, std::vector<at::Tensor>>
+fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+                std::vector<at::Tensor> exp_avgs,
+                std::vector<at::Tensor> exp_avg_sqs,
+                std::vector<at::Tensor> steps, double lr, double beta1,
+                double beta2, double weight_decay, double eps) {
+  const int T = params.size();
+  std::vector<at::Tensor> out_p, out_ea, out_eas, out_steps;
+  out_p.reserve(T); out_ea.reserve(T); out_eas.reserve(T);
+  for (int t = 0; t < T; ++t) {
+    TORCH_CHECK(params[t].scalar_type() == at::kFloat, "fp32 params only");
+    TORCH_CHECK(params[t].is_contiguous() && grads[t].is_contiguous());
+    out_p.push_back(at::empty_like(params[t]));
+    out_ea.push_back(at::empty_like(exp_avgs[t]));
+    out_eas.push_back(at::empty_like(exp_avg_sqs[t]));
+    out_steps.push_back(at::empty_like(steps[t]));
+  }
+  // steps as one packed fp32 tensor [T]
+  auto steps_flat = at::empty({T}, params[0].options());
+  {
+    std::vector<at::Tensor> svec;
+    for (auto& s : steps) svec.push_back(s.reshape({1}).to(at::kFloat));
+    at::cat_out(steps_flat, svec, 0);
+  }
+
+  std::vector<AdamChunk> chunks;
+  for (int t = 0; t < T; ++t) {
+    long n = params[t].numel();
+    for (long off = 0; off < n; off += CHUNK_ELEMS) {
+      AdamChunk c;
+      c.p = params[t].data_ptr<float>() + off;
+      c.g = grads[t].data_ptr<float>() + off;
+      c.ea = exp_avgs[t].data_ptr<float>() + off;
+      c.eas = exp_avg_sqs[t].data_ptr<float>() + off;
+      c.out_p = out_p[t].data_ptr<float>() + off;
+      c.out_ea = out_ea[t].data_ptr<float>() + off;
+      c.out_eas = out_eas[t].data_ptr<float>() + off;
+      c.tensor_idx = t;
+      c.n = (int)std::min<long>(CHUNK_ELEMS, n - off);
+      chunks.push_back(c);
+    }
+  }
+  // upload the table; when a hipGraph capture is active the pinned host
+  // buffer must outlive the graph (the captured H2D reads it on replay)
+  auto host = at::empty({(long)(chunks.size() * sizeof(AdamChunk))},
+                        at::TensorOptions().dtype(at::kByte).pinned_memory(true));
+  memcpy(host.data_ptr(), chunks.data(), chunks.size() * sizeof(AdamChunk));
+  at::Tensor dev_chunks = host.to(params[0].device(), /*non_blocking=*/true);
+  if (at::cuda::currentStreamCaptureStatusMayInitCtx() !=
+      at::cuda::CaptureStatus::None) {
+    static std::vector<at::Tensor> capture_keepalive;
+    capture_keepalive.push_back(host);
+  }
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(fused_adam_kernel, dim3(chunks.size()), dim3(256), 0,
+      stream, (const AdamChunk*)dev_chunks.data_ptr(), (int)chunks.size(),
+      steps_flat.data_ptr<float>(), (float)lr, (float)beta1, (float)beta2,
+      (float)weight_decay, (float)eps);
+  // bump steps on host-free path: out_step = step + 1
+  for (int t = 0; t < T; ++t)
+    out_steps[t] = steps[t] + 1;
+  return {out_p, out_ea, out_eas, out_steps};
+}

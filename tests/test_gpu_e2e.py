@@ -1,0 +1,78 @@
+"""GPU end-to-end: compiled GPT train step golden vs vanilla on cuda:0."""
+import copy
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs MI355X")
+
+
+@requires_gpu
+def test_gpt_golden_cuda():
+    from easydist_amd import easydist_compile, easydist_setup, set_device_mesh
+    from easydist_amd.models.gpt import GPT, GPTConfig
+    from easydist_amd.utils.testing import init_single_process
+
+    init_single_process()
+    easydist_setup(device="cuda")
+    set_device_mesh([0], ["spmd0"])
+    torch.manual_seed(5)
+    cfg = GPTConfig(vocab_size=512, n_layer=2, n_head=4, n_embd=256,
+                    block_size=128)
+    model = GPT(cfg).cuda()
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+    opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-3, fused=True)
+
+    def train_step(model, opt, idx, targets):
+        loss = model.loss(idx, targets)
+        loss.backward()
+        opt.step()
+        opt.zero_grad(True)
+        return loss
+
+    compiled = easydist_compile(train_step, cuda_graph=False)
+    torch.manual_seed(6)
+    for _ in range(3):
+        idx = torch.randint(0, 512, (4, 128), device="cuda")
+        tg = torch.randint(0, 512, (4, 128), device="cuda")
+        loss = compiled(model, opt, idx, tg)
+        ref = train_step(model_ref, opt_ref, idx, tg)
+        assert abs(float(loss) - float(ref)) < 5e-3, (float(loss), float(ref))
+
+
+@requires_gpu
+def test_gpt_hipgraph_cuda():
+    """Same model under hipGraph capture+replay."""
+    from easydist_amd import easydist_compile, easydist_setup, set_device_mesh
+    from easydist_amd.models.gpt import GPT, GPTConfig
+    from easydist_amd.utils.testing import init_single_process
+
+    init_single_process()
+    easydist_setup(device="cuda")
+    set_device_mesh([0], ["spmd0"])
+    torch.manual_seed(7)
+    cfg = GPTConfig(vocab_size=512, n_layer=2, n_head=4, n_embd=256,
+                    block_size=128)
+    model = GPT(cfg).cuda()
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+
+    def train_step(model, opt, idx, targets):
+        loss = model.loss(idx, targets)
+        loss.backward()
+        opt.step()
+        opt.zero_grad(True)
+        return loss
+
+    compiled = easydist_compile(train_step, cuda_graph=True)
+    losses = []
+    for _ in range(6):
+        idx = torch.randint(0, 512, (8, 128), device="cuda")
+        tg = torch.randint(0, 512, (8, 128), device="cuda")
+        loss = compiled(model, opt, idx, tg)
+        losses.append(float(loss))
+    # training under replay should make progress (loss drops from random)
+    assert losses[-1] < losses[0], losses
