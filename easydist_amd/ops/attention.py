@@ -60,13 +60,21 @@ def _fwd_cpu(q, k, v, causal):
     return _math_fwd(q, k, v, causal)
 
 
+def _kernel_ok(q, k, v):
+    """The HIP kernel's supported envelope. Discovery probes this op with
+    arbitrarily sharded shapes — anything outside goes to the math path."""
+    return (q.dtype == torch.bfloat16 and q.dim() == 4
+            and q.shape == k.shape and q.shape == v.shape
+            and q.shape[-1] in (64, 128) and q.shape[2] % 64 == 0)
+
+
 def _fwd_cuda(q, k, v, causal):
     ext = load_extension()
-    if ext is not None and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+    if ext is not None and _kernel_ok(q, k, v):
         q, k, v = (t.contiguous() for t in (q, k, v))
         return ext.flash_attn_fwd(q, k, v, causal)
     from . import require_hip_ops
-    if q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+    if _kernel_ok(q, k, v):
         require_hip_ops()   # raises when the extension should exist
     return _math_fwd(q, k, v, causal)
 

@@ -190,6 +190,8 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
                                                   bool causal) {
   TORCH_CHECK(q.dtype() == at::kBFloat16 && q.dim() == 4);
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.sizes() == k.sizes() && q.sizes() == v.sizes(),
+              "flash_attn_fwd: q/k/v shapes must match");
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(D == 64 || D == 128, "flash_attn_fwd: D in {64,128}");
   TORCH_CHECK(S % QB == 0, "flash_attn_fwd: S multiple of 64");

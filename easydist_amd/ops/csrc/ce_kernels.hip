@@ -69,6 +69,7 @@ std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
                                           const at::Tensor& targets) {
   TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
   TORCH_CHECK(targets.scalar_type() == at::kLong);
+  TORCH_CHECK(targets.numel() == logits.size(0), "targets/rows mismatch");
   const int N = logits.size(0), V = logits.size(1);
   auto nll = at::empty({N}, logits.options().dtype(at::kFloat));
   auto lse = at::empty({N}, logits.options().dtype(at::kFloat));
@@ -89,6 +90,9 @@ std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
 
 at::Tensor ce_bwd(const at::Tensor& grad, const at::Tensor& logits,
                   const at::Tensor& targets, const at::Tensor& lse) {
+  TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
+  TORCH_CHECK(targets.numel() == logits.size(0));
+  TORCH_CHECK(lse.numel() == logits.size(0));
   const int N = logits.size(0), V = logits.size(1);
   auto dlogits = at::empty_like(logits);
   auto gradf = grad.to(at::kFloat).contiguous();

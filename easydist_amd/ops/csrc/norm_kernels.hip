@@ -209,10 +209,18 @@ static void check_lastdim(const at::Tensor& x) {
   TORCH_CHECK(x.size(-1) % 8 == 0, "feature dim must be a multiple of 8");
 }
 
+static void check_affine(const at::Tensor& x,
+                         const std::optional<at::Tensor>& w,
+                         const std::optional<at::Tensor>& b) {
+  if (w) TORCH_CHECK(w->numel() == x.size(-1), "weight/feature mismatch");
+  if (b) TORCH_CHECK(b->numel() == x.size(-1), "bias/feature mismatch");
+}
+
 std::tuple<at::Tensor, at::Tensor, at::Tensor>
 layer_norm_fwd(const at::Tensor& x, const std::optional<at::Tensor>& w,
                const std::optional<at::Tensor>& b, double eps) {
   check_lastdim(x);
+  check_affine(x, w, b);
   const int D = x.size(-1);
   const long rows = x.numel() / D;
   auto out = at::empty_like(x);
@@ -243,6 +251,9 @@ layer_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
                const std::optional<at::Tensor>& w,
                std::vector<bool> mask) {
   check_lastdim(x);
+  check_affine(x, w, std::nullopt);
+  TORCH_CHECK(grad.sizes() == x.sizes() && grad.is_contiguous());
+  TORCH_CHECK(mean.numel() == x.numel() / x.size(-1));
   TORCH_CHECK(x.scalar_type() == at::kBFloat16,
               "layer_norm_bwd kernel: bf16 only (fp32 falls back to aten)");
   const int D = x.size(-1);
@@ -272,6 +283,7 @@ std::tuple<at::Tensor, at::Tensor>
 rms_norm_fwd(const at::Tensor& x, const std::optional<at::Tensor>& w,
              double eps) {
   check_lastdim(x);
+  check_affine(x, w, std::nullopt);
   const int D = x.size(-1);
   const long rows = x.numel() / D;
   auto out = at::empty_like(x);
@@ -299,6 +311,8 @@ std::tuple<at::Tensor, at::Tensor>
 rms_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
              const at::Tensor& rstd, const std::optional<at::Tensor>& w) {
   check_lastdim(x);
+  check_affine(x, w, std::nullopt);
+  TORCH_CHECK(grad.sizes() == x.sizes() && grad.is_contiguous());
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "rms_norm_bwd: bf16 only");
   const int D = x.size(-1);
   const long rows = x.numel() / D;
