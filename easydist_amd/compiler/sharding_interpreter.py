@@ -9,6 +9,7 @@ is targeting — on an MI355X box this runs the real HIP kernels.
 from __future__ import annotations
 
 import logging
+import os
 from typing import Dict, Optional, Tuple
 
 import torch
@@ -101,6 +102,9 @@ class EDTorchShardingAnn:
                 continue
             # 3) execution-based discovery with materialized tensors
             try:
+                if os.environ.get("EASYDIST_DEBUG_DISCOVERY"):
+                    print(f"[discover] {node.name} {node.target} "
+                          f"{input_shapes}", flush=True)
                 info[node.name] = self._discover(node)
                 _DISCOVERY_CACHE[sig] = info[node.name]
             except Exception as e:

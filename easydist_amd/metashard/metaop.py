@@ -101,6 +101,15 @@ class MetaOp:
             shards = shard_tensor(t, d, num_shards, halo=sd.halo, chunk=sd.chunk)
             for s in range(num_shards):
                 per_shard_inputs[s].append(shards[s])
+        # meta dry-run first: many aten CUDA kernels skip shape validation
+        # and OOB-fault on inconsistent shard combinations; the meta kernels
+        # DO validate, so a malformed combo dies here on the CPU instead of
+        # taking down the GPU.
+        try:
+            metas = [t.to("meta") for t in per_shard_inputs[0]]
+            self._call_with_tensors(metas)
+        except NotImplementedError:
+            pass   # no meta kernel: proceed (plain eager will raise cleanly)
         outs = []
         for s in range(num_shards):
             outs.append(self._call_with_tensors(per_shard_inputs[s]))
