@@ -70,10 +70,12 @@ def warmup_optimizer(module: torch.nn.Module, opt) -> dict:
 
 def ed_compile_func(func: Callable, tracing_mode: str, args, kwargs,
                     module: Optional[torch.nn.Module], opt,
-                    split_patcher_ctx=None):
+                    split_patcher_ctx=None, decomp_table=None):
     """Trace `func(module, opt, *args)` into one whole-step fx graph.
 
     Returns (params, buffers, named_states, traced_graph).
+    `decomp_table` defaults to EASYDIST_DECOMP_TABLE; the manual-DP path
+    passes {} so fused/foreach optimizer ops stay whole in the graph.
     """
     params, buffers = {}, {}
     if module is not None:
@@ -81,11 +83,13 @@ def ed_compile_func(func: Callable, tracing_mode: str, args, kwargs,
         buffers = dict(module.named_buffers())
     named_states = warmup_optimizer(module, opt) if opt is not None else {}
 
+    if decomp_table is None:
+        decomp_table = EASYDIST_DECOMP_TABLE
     ctx = split_patcher_ctx if split_patcher_ctx is not None else nullcontext()
     with _enable_compile(), ctx:
         traced_graph = make_fx(partial(stateless_func, func, module, opt),
                                tracing_mode=tracing_mode,
-                               decomposition_table=EASYDIST_DECOMP_TABLE,
+                               decomposition_table=decomp_table,
                                _allow_non_fake_inputs=False)(
                                    params, buffers, named_states, args, kwargs)
     traced_graph.graph.eliminate_dead_code()

@@ -20,14 +20,18 @@ logger = logging.getLogger(__name__)
 # --------------------------------------------------------------- placements --
 @dataclass(frozen=True)
 class SPMD:
-    """Single-mesh-dim placement: REPLICATE, SHARD(dim) or PARTIAL(op)."""
-    state: str            # 'R' | 'S' | 'P'
+    """Single-mesh-dim placement: REPLICATE, SHARD(dim), PARTIAL(op) or
+    FLAT (1-D flatten+pad shard, used by the ZeRO transforms where state
+    tensors are sharded regardless of their shape)."""
+    state: str            # 'R' | 'S' | 'P' | 'F'
     dim: int = -1         # tensor dim for 'S'
     reduce_op: str = ""   # 'sum' | 'max' | 'min' for 'P'
+    shape: Tuple[int, ...] = ()   # global shape for 'F' (unshard needs it)
 
     REPLICATE = "R"
     SHARD = "S"
     PARTIAL = "P"
+    FLAT = "F"
 
     def is_replicate(self):
         return self.state == "R"
@@ -38,11 +42,16 @@ class SPMD:
     def is_partial(self):
         return self.state == "P"
 
+    def is_flat_shard(self):
+        return self.state == "F"
+
     def __repr__(self):
         if self.state == "R":
             return "R"
         if self.state == "S":
             return f"S({self.dim})"
+        if self.state == "F":
+            return f"F{list(self.shape)}"
         return f"P({self.reduce_op})"
 
 
@@ -55,6 +64,10 @@ def S(dim: int) -> SPMD:
 
 def P(op: str = "sum") -> SPMD:
     return SPMD(SPMD.PARTIAL, reduce_op=op)
+
+
+def F(shape) -> SPMD:
+    return SPMD(SPMD.FLAT, shape=tuple(shape))
 
 
 # ------------------------------------------------------------------- IR ------

@@ -190,6 +190,39 @@ def partial_localize(t: torch.Tensor, group=None) -> torch.Tensor:
     return torch.zeros_like(t)
 
 
+# ---------------------------------------------------- flat (ZeRO) helpers ----
+def _flat_pad(t: torch.Tensor, n: int) -> torch.Tensor:
+    """Flatten to 1-D and zero-pad so numel % n == 0."""
+    flat = t.reshape(-1)
+    pad = (-flat.numel()) % n
+    if pad:
+        flat = torch.cat([flat, flat.new_zeros(pad)])
+    return flat.contiguous()
+
+
+def flat_shard_local(t: torch.Tensor, group=None) -> torch.Tensor:
+    """REPLICATE -> FLAT shard: my 1-D slice of the padded flat tensor."""
+    n = _world(group)
+    flat = _flat_pad(t, n)
+    return torch.chunk(flat, n, dim=0)[_rank(group)].contiguous()
+
+
+def reduce_scatter_flat(t: torch.Tensor, op: str = "avg",
+                        group=None) -> torch.Tensor:
+    """grad (full, per-rank partial) -> reduced 1-D local shard."""
+    n = _world(group)
+    return reduce_scatter(_flat_pad(t, n), 0, op, group)
+
+
+def all_gather_flat(shard: torch.Tensor, shape, group=None) -> torch.Tensor:
+    """FLAT shard -> full tensor of `shape` (drops the pad)."""
+    full = all_gather(shard.contiguous(), 0, group)
+    numel = 1
+    for s in shape:
+        numel *= s
+    return full[:numel].reshape(tuple(shape))
+
+
 # ---------------------------------------------------------------- PP p2p -----
 def batch_p2p(p2p_ops: List[dist.P2POp]):
     if not p2p_ops:

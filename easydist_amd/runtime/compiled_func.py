@@ -31,6 +31,8 @@ def shard_tensor_local(t: torch.Tensor, placements: List[SPMD],
             out = comm.local_chunk(out, p.dim, mesh.get_group(d))
         elif p.is_partial():
             out = comm.partial_localize(out, mesh.get_group(d))
+        elif p.is_flat_shard():
+            out = comm.flat_shard_local(out, mesh.get_group(d))
     return out
 
 
@@ -43,6 +45,8 @@ def unshard_tensor(t: torch.Tensor, placements: List[SPMD], mesh) -> torch.Tenso
             out = comm.all_gather(out, p.dim, mesh.get_group(d))
         elif p.is_partial():
             out = comm.all_reduce(out, p.reduce_op or "sum", mesh.get_group(d))
+        elif p.is_flat_shard():
+            out = comm.all_gather_flat(out, p.shape, mesh.get_group(d))
     return out
 
 
@@ -100,6 +104,8 @@ class EDCompiledFunc:
             new = flat_outs[out_pos]
             if isinstance(new, torch.Tensor) and in_pos in self.state:
                 buf = self.state[in_pos]
+                if new is buf:          # in-place updated state: already done
+                    continue
                 if buf.shape == new.shape and buf.dtype == new.dtype:
                     buf.copy_(new)
                 else:
@@ -172,11 +178,15 @@ class EDCompiledFunc:
                 # writeback inside the graph: copy into persistent state
                 for in_pos, out_pos in self.io_pos_map.items():
                     new = outs[out_pos]
-                    if isinstance(new, torch.Tensor) and in_pos in self.state:
+                    if isinstance(new, torch.Tensor) and in_pos in self.state \
+                            and new is not self.state[in_pos]:
                         buf = self.state[in_pos]
                         if buf.shape == new.shape and buf.dtype == new.dtype:
                             buf.copy_(new)
                 self._graph_out = outs
+            # capture only RECORDS the kernels — nothing executed yet; replay
+            # once so this call's outputs (and state update) are real
+            self._graph.replay()
             return self._graph_out
         # replay path
         for i in self._dyn_pos:
