@@ -163,6 +163,9 @@ class EDCompiledFunc:
             self._static = {i: prepared[i].clone() for i in self._dyn_pos}
             for i in self._dyn_pos:
                 prepared[i] = self._static[i]
+            # the warmup steps below advance the persistent state; snapshot
+            # it so the captured graph replays from the true current step
+            snapshot = {pos: t.clone() for pos, t in self.state.items()}
             # warmup on a side stream (RCCL + hipGraph requirement)
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
@@ -171,6 +174,10 @@ class EDCompiledFunc:
                     outs = self.gm(*prepared)
                     self._writeback(outs)
             torch.cuda.current_stream().wait_stream(s)
+            with torch.no_grad():
+                for pos, t in self.state.items():
+                    t.copy_(snapshot[pos])
+            del snapshot
             self._graph = torch.cuda.CUDAGraph()
             self._graph_prepared = prepared
             with torch.cuda.graph(self._graph):
