@@ -47,6 +47,10 @@ def easydist_setup(backend: str = "torch", device: str = "cuda",
     # connection multiplexing
     os.environ.setdefault("CUDA_DEVICE_MAX_CONNECTIONS", "1")
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    if mdconfig.enable_memory_opt and device == "cuda":
+        # must happen before the first device allocation
+        from .memory import init_meta_allocator
+        init_meta_allocator()
     return True
 
 

@@ -43,12 +43,16 @@ def _find_module_opt(args, kwargs):
 
 class CompiledFuncWrapper:
     def __init__(self, func, parallel_mode="auto", tracing_mode="fake",
-                 cuda_graph=True, compile_only=False, **compile_kwargs):
+                 cuda_graph=True, compile_only=False, memory_opt=False,
+                 **compile_kwargs):
         update_wrapper(self, func)
         self.original_func = func
         self.parallel_mode = parallel_mode
         self.tracing_mode = tracing_mode
         self.enable_cuda_graph = cuda_graph and mdconfig.enable_hip_graph
+        self.memory_opt = memory_opt or mdconfig.enable_memory_opt
+        if self.memory_opt:
+            self.enable_cuda_graph = False   # plan playback is eager-mode
         self.compile_only = compile_only
         self.compile_kwargs = compile_kwargs
         self.compiled: Dict[str, object] = {}   # input signature -> runtime
@@ -93,7 +97,10 @@ class CompiledFuncWrapper:
         # note: state positions come from the runtime's persistent buffers
         # after the first call; the flat list only seeds them once.
         flat_inputs = self._reflatten(runtime, params, buffers, args, kwargs)
-        if (self.enable_cuda_graph and torch.cuda.is_available()):
+        if self.memory_opt and torch.cuda.is_available() \
+                and hasattr(runtime, "run_planned"):
+            outs = runtime.run_planned(flat_inputs)
+        elif (self.enable_cuda_graph and torch.cuda.is_available()):
             outs = runtime.run_graph(flat_inputs)
         else:
             outs = runtime.run(flat_inputs)
@@ -130,11 +137,12 @@ class CompiledFuncWrapper:
 def easydist_compile(func=None, parallel_mode="auto", tracing_mode="fake",
                      cuda_graph=True, use_hint=False,
                      max_solver_time=float("inf"), compile_only=False,
-                     **compile_kwargs):
+                     memory_opt=False, **compile_kwargs):
     mdconfig.use_hint = use_hint
     mdconfig.max_seconds_same_incumbent = max_solver_time
 
     def deco(fn):
         return CompiledFuncWrapper(fn, parallel_mode, tracing_mode,
-                                   cuda_graph, compile_only, **compile_kwargs)
+                                   cuda_graph, compile_only, memory_opt,
+                                   **compile_kwargs)
     return deco(func) if func else deco

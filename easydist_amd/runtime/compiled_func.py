@@ -124,6 +124,39 @@ class EDCompiledFunc:
         self._writeback(flat_outs)
         return flat_outs
 
+    # ---------------------------------------------- static memory plan ------
+    def run_planned(self, flat_inputs: List):
+        """PROFILE the first step's allocations, pack them with the
+        min-skyline scheduler, then play the plan back from the C++
+        allocator's arena on every subsequent step (reference memory_opt
+        flow, compile_auto.py:353-453)."""
+        from ..memory import allocator_installed
+        from ..memory import meta_allocator as ma
+        if not allocator_installed():
+            return self.run(flat_inputs)
+        if not self.state:
+            self.init_state(flat_inputs)
+        prepared = self._prepare_inputs(flat_inputs)
+        if getattr(self, "_mem_plan", None) is None:
+            from ..memory.allocator_profiler import AllocatorProfiler
+            from ..schedule import plan_memory
+            prof = AllocatorProfiler(self.gm)
+            mem_info = prof.profile(prepared)
+            entries, arena, stats = plan_memory(self.gm, mem_info)
+            ma.ctl().load_plan(entries, arena)
+            ma.ctl().set_mode(ma.RUNTIME)
+            self._mem_plan = stats
+            # the profiling run already executed the step: writeback + done
+            # is NOT possible (interpreter discarded outputs order) — rerun
+        c = ma.ctl()
+        c.start_region()
+        try:
+            flat_outs = self.gm(*prepared)
+        finally:
+            c.stop_region()
+        self._writeback(flat_outs)
+        return flat_outs
+
     # ------------------------------------------------- state access APIs ----
     def named_parameters(self) -> Dict[str, torch.Tensor]:
         """Gather the (sharded) params back to global tensors."""

@@ -32,7 +32,15 @@ setup(
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
             },
-        )
+        ),
+        CUDAExtension(
+            name="easydist_amd.memory._mem_alloc",
+            sources=["easydist_amd/memory/csrc/profiling_allocator.cpp"],
+            extra_compile_args={
+                "cxx": ["-O3"],
+                "nvcc": ["-O3"],
+            },
+        ),
     ],
     cmdclass={"build_ext": BuildExtension},
 )

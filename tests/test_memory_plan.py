@@ -1,0 +1,53 @@
+"""CPU unit tests for the min-skyline memory packer (schedule/)."""
+import random
+
+from easydist_amd.schedule.efficient_memory_scheduler import (ALIGN,
+                                                              pack_buffers)
+from easydist_amd.schedule.lifetime import Buffer
+
+
+def _overlap_time(a: Buffer, b: Buffer) -> bool:
+    return not (a.end < b.start or b.end < a.start)
+
+
+def _overlap_addr(oa, a: Buffer, ob, b: Buffer) -> bool:
+    return not (oa + a.size <= ob or ob + b.size <= oa)
+
+
+def test_pack_disjoint_lifetimes_share_memory():
+    bufs = [Buffer("a", 0, 1024, 0, 1, False),
+            Buffer("b", 0, 1024, 2, 3, False)]
+    addr, peak = pack_buffers(bufs)
+    assert peak == 1024           # reuse: both at offset 0
+    assert addr[("a", 0)] == addr[("b", 0)] == 0
+
+
+def test_pack_overlapping_lifetimes_disjoint_addresses():
+    bufs = [Buffer("a", 0, 1000, 0, 5, False),
+            Buffer("b", 0, 2000, 3, 8, False),
+            Buffer("c", 0, 500, 4, 4, True)]
+    addr, peak = pack_buffers(bufs)
+    for i, x in enumerate(bufs):
+        for y in bufs[i + 1:]:
+            if _overlap_time(x, y):
+                assert not _overlap_addr(addr[(x.node_name, 0)], x,
+                                         addr[(y.node_name, 0)], y), (x, y)
+    assert peak <= sum(b.size for b in bufs) + 2 * ALIGN
+
+
+def test_pack_random_no_overlap_and_saves():
+    rng = random.Random(0)
+    bufs = []
+    for i in range(200):
+        start = rng.randrange(0, 100)
+        end = start + rng.randrange(0, 10)
+        bufs.append(Buffer(f"n{i}", 0, rng.choice([256, 1024, 4096, 100000]),
+                           start, end, False))
+    addr, peak = pack_buffers(bufs)
+    for i, x in enumerate(bufs):
+        for y in bufs[i + 1:]:
+            if _overlap_time(x, y):
+                assert not _overlap_addr(addr[(x.node_name, 0)], x,
+                                         addr[(y.node_name, 0)], y)
+    naive = sum(b.size for b in bufs)
+    assert peak < naive * 0.6     # interleaved lifetimes must reuse memory
