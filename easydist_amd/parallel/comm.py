@@ -190,6 +190,58 @@ def partial_localize(t: torch.Tensor, group=None) -> torch.Tensor:
     return torch.zeros_like(t)
 
 
+# -------------------------------------------------------- expert parallel ----
+def _all_to_all_ep_impl(t: torch.Tensor, group) -> torch.Tensor:
+    n = _world(group)
+    assert t.shape[0] % n == 0
+    t = t.contiguous()
+    if n == 1:
+        return t
+    if _is_gloo(group):
+        gout = torch.empty((n * t.shape[0],) + tuple(t.shape[1:]),
+                           dtype=t.dtype, device=t.device)
+        dist.all_gather_into_tensor(gout, t, group=group)
+        rank = _rank(group)
+        # block (src s, chunk w) at index s*n + w; take w == my rank
+        blocks = torch.chunk(gout, n * n, dim=0)
+        return torch.cat([blocks[s * n + rank] for s in range(n)],
+                         dim=0).contiguous()
+    out = torch.empty_like(t)
+    dist.all_to_all_single(out, t, group=group)
+    return out
+
+
+class _AllToAllEP(torch.autograd.Function):
+    """Autograd-aware equal-split all-to-all: the backward of an
+    all-to-all permutation is the same all-to-all on the grads."""
+
+    @staticmethod
+    def forward(ctx, t, group):
+        ctx.group = group
+        with torch.no_grad():
+            return _all_to_all_ep_impl(t.detach(), group)
+
+    @staticmethod
+    def backward(ctx, grad):
+        with torch.no_grad():
+            g = _all_to_all_ep_impl(grad.contiguous(), ctx.group)
+        return g, None
+
+
+def all_to_all_ep(t: torch.Tensor, group=None) -> torch.Tensor:
+    """Equal-split all-to-all along dim 0 (expert dispatch/combine).
+
+    Input [W*L, ...]: chunk w goes to rank w; output chunk w is what rank
+    w sent here. On RCCL this is ONE all_to_all_single = W-1 simultaneous
+    pairwise xGMI exchanges; gloo (CPU tests) emulates via all_gather.
+    Differentiable (EP training needs grads flowing back through
+    dispatch AND combine).
+    """
+    if torch.is_grad_enabled() and t.requires_grad:
+        return _AllToAllEP.apply(t, group)
+    return _all_to_all_ep_impl(t, group)
+
+
 # ---------------------------------------------------- flat (ZeRO) helpers ----
 def _flat_pad(t: torch.Tensor, n: int) -> torch.Tensor:
     """Flatten to 1-D and zero-pad so numel % n == 0."""

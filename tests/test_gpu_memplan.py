@@ -55,6 +55,8 @@ def test_memory_plan_end_to_end(tmp_path):
     p.write_text(SCRIPT)
     env = dict(os.environ)
     env["EASYDIST_MEM_OPT"] = "1"
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
     r = subprocess.run([sys.executable, str(p)], capture_output=True,
                        text=True, timeout=420, env=env)
     assert "MEMPLAN_OK" in r.stdout, (r.stdout[-3000:], r.stderr[-3000:])
