@@ -100,11 +100,18 @@ def _ep_grad_body(world_size):
     # sum-loss so per-token grads just add across the batch split
     ep(shard).sum().backward()
     ref(x).sum().backward()
-    g = ep.experts.w2.grad.clone()
-    dist.all_reduce(g)   # tokens were split across ranks: sum contributions
-    want = ref.experts.w2.grad[r * L:(r + 1) * L]
-    assert torch.allclose(g, want, rtol=1e-4, atol=1e-5), \
-        (g - want).abs().max()
+    # expert grads are COMPLETE locally: each expert saw every rank's
+    # tokens through the dispatch all-to-all
+    for wname in ("w1", "w3", "w2"):
+        g = getattr(ep.experts, wname).grad
+        want = getattr(ref.experts, wname).grad[r * L:(r + 1) * L]
+        assert torch.allclose(g, want, rtol=1e-4, atol=1e-5), \
+            (wname, (g - want).abs().max())
+    # router grads cover only local tokens: sum over the batch split
+    gr = ep.router.weight.grad.clone()
+    dist.all_reduce(gr)
+    assert torch.allclose(gr, ref.router.weight.grad, rtol=1e-4,
+                          atol=1e-5)
 
 
 @pytest.mark.world2
