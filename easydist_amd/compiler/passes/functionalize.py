@@ -66,10 +66,14 @@ def fix_inplace(gm: fx.GraphModule) -> fx.GraphModule:
             func_pkt = getattr(aten, func_name, None)
             if func_pkt is None:
                 continue
-            try:
-                new_target = func_pkt.default
-            except Exception:
-                continue
+            # keep the same overload: add_.Tensor -> add.Tensor etc.
+            ol_name = getattr(node.target, "_overloadname", "default")
+            new_target = getattr(func_pkt, ol_name, None)
+            if new_target is None:
+                try:
+                    new_target = func_pkt.default
+                except Exception:
+                    continue
             with gm.graph.inserting_after(node):
                 new_node = gm.graph.call_function(new_target, node.args,
                                                   node.kwargs)
