@@ -84,18 +84,15 @@ def _mark_tree(out, idx: int):
 
 
 # ---------------------------------------------------------- annotation API ---
-_ANNOTATED: Dict[int, bool] = {}   # id(module) -> True
-
-
 def annotate_split_points(module: torch.nn.Module, spots: Set[str]):
     """Insert a pipeline boundary AFTER each named submodule's forward.
 
     reference: compile_pipeline.py:51-78."""
     for qualname in sorted(spots):
         sub = module.get_submodule(qualname)
-        if id(sub) in _ANNOTATED:
+        if getattr(sub, "_ed_pp_annotated", False):
             continue
-        _ANNOTATED[id(sub)] = True
+        sub._ed_pp_annotated = True
 
         def hook(mod, args, out):
             if not _SplitState.active:
