@@ -112,6 +112,10 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     gm, out_pl_env = sharding_transform(gm, strategies_per_dim, mesh.shape)
     gm = _fix_output_reshard(gm, out_pl_env, io_map, ret_names, mesh)
 
+    # ---- 5b. comm overlap: widen every start/wait window -----------------
+    from .passes.comm_optimize import comm_optimize
+    comm_optimize(gm)
+
     # strip the pytree codegen: the runtime calls the graph with the flat
     # placeholder list and receives the flat output list
     import torch.fx as _fx

@@ -68,11 +68,10 @@ def test_gpt_hipgraph_cuda():
         return loss
 
     compiled = easydist_compile(train_step, cuda_graph=True)
-    losses = []
-    for _ in range(6):
-        idx = torch.randint(0, 512, (8, 128), device="cuda")
-        tg = torch.randint(0, 512, (8, 128), device="cuda")
-        loss = compiled(model, opt, idx, tg)
-        losses.append(float(loss))
+    # FIXED batch: memorization must drive the loss down monotonically-ish;
+    # fresh random batches would keep loss pinned near ln(vocab)
+    idx = torch.randint(0, 512, (8, 128), device="cuda")
+    tg = torch.randint(0, 512, (8, 128), device="cuda")
+    losses = [float(compiled(model, opt, idx, tg)) for _ in range(6)]
     # training under replay should make progress (loss drops from random)
-    assert losses[-1] < losses[0], losses
+    assert losses[-1] < losses[0] - 0.5, losses
