@@ -153,6 +153,8 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
         gm, in_spec, out_spec_graph, input_placements, output_placements,
         state_positions, io_pos_map, len(params), list(params.keys()), device)
     compiled.init_named_states = named_states
+    compiled.ret_out_positions = list(range(n_state_outs + n_grads,
+                                            len(flat_outs)))
     compiled.meta = {
         "search_time": search_time, "solve_time": solve_time,
         "n_nodes": len(gm.graph.nodes), "out_spec": gm._out_spec

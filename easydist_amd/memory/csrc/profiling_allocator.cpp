@@ -48,11 +48,18 @@ struct PlanEntry {
   size_t size;
 };
 
+struct Event {
+  int8_t is_alloc;   // 1 alloc, 0 free
+  uintptr_t ptr;
+  size_t size;
+};
+
 std::mutex g_mu;
 Mode g_mode = Mode::PASSTHROUGH;
 bool g_in_region = false;
 std::string g_cur_op = "<unknown>";
 std::vector<AllocRecord> g_records;
+std::vector<Event> g_events;
 
 // runtime plan state
 std::vector<PlanEntry> g_plan;
@@ -122,12 +129,17 @@ void* ed_malloc(size_t size, int device, hipStream_t stream) {
     std::lock_guard<std::mutex> lk2(g_mu);
     g_records.push_back({g_cur_op, reinterpret_cast<uintptr_t>(p), size,
                          reinterpret_cast<uintptr_t>(stream)});
+    g_events.push_back({1, reinterpret_cast<uintptr_t>(p), size});
   }
   return p;
 }
 
 void ed_free(void* ptr, size_t size, int device, hipStream_t stream) {
   if (ptr == nullptr) return;
+  if (g_mode == Mode::PROFILE) {
+    std::lock_guard<std::mutex> lk2(g_mu);
+    g_events.push_back({0, reinterpret_cast<uintptr_t>(ptr), 0});
+  }
   if (in_arena(ptr)) return;  // plan-owned: lifetime handled statically
   std::lock_guard<std::mutex> lk(g_mu);
   auto it = g_cached_sizes.find(reinterpret_cast<uintptr_t>(ptr));
@@ -152,6 +164,19 @@ PYBIND11_MODULE(_mem_alloc, m) {
     std::lock_guard<std::mutex> lk(g_mu);
     g_records.clear();
   });
+  m.def("clear_events", [] {
+    std::lock_guard<std::mutex> lk(g_mu);
+    g_events.clear();
+  });
+  m.def("get_events", [] {
+    std::lock_guard<std::mutex> lk(g_mu);
+    py::list out;
+    for (const auto& e : g_events)
+      out.append(py::make_tuple(static_cast<int>(e.is_alloc), e.ptr,
+                                e.size));
+    return out;
+  });
+  m.def("arena_size", [] { return g_arena_size; });
   m.def("get_records", [] {
     std::lock_guard<std::mutex> lk(g_mu);
     py::list out;

@@ -109,7 +109,9 @@ class EDCompiledFunc:
                 if buf.shape == new.shape and buf.dtype == new.dtype:
                     buf.copy_(new)
                 else:
-                    self.state[in_pos] = new
+                    # clone: `new` may live in the memory-plan arena whose
+                    # addresses are re-served next step
+                    self.state[in_pos] = new.clone()
 
     # -------------------------------------------------------------- call -----
     def __call__(self, *args, **kwargs):
@@ -126,10 +128,13 @@ class EDCompiledFunc:
 
     # ---------------------------------------------- static memory plan ------
     def run_planned(self, flat_inputs: List):
-        """PROFILE the first step's allocations, pack them with the
-        min-skyline scheduler, then play the plan back from the C++
-        allocator's arena on every subsequent step (reference memory_opt
-        flow, compile_auto.py:353-453)."""
+        """First step runs under PROFILE recording the real alloc/free
+        EVENT sequence; the min-skyline packer assigns arena offsets from
+        those intervals; later steps play the plan back from the C++
+        allocator's arena (reference memory_opt flow,
+        compile_auto.py:353-453, re-designed event-exact so torch's
+        pluggable-allocator metadata never sees an address re-served
+        while a live tensor still holds it)."""
         from ..memory import allocator_installed
         from ..memory import meta_allocator as ma
         if not allocator_installed():
@@ -137,25 +142,50 @@ class EDCompiledFunc:
         if not self.state:
             self.init_state(flat_inputs)
         prepared = self._prepare_inputs(flat_inputs)
-        if getattr(self, "_mem_plan", None) is None:
-            from ..memory.allocator_profiler import AllocatorProfiler
-            from ..schedule import plan_memory
-            prof = AllocatorProfiler(self.gm)
-            mem_info = prof.profile(prepared)
-            entries, arena, stats = plan_memory(self.gm, mem_info)
-            ma.ctl().load_plan(entries, arena)
-            ma.ctl().set_mode(ma.RUNTIME)
-            self._mem_plan = stats
-            # the profiling run already executed the step: writeback + done
-            # is NOT possible (interpreter discarded outputs order) — rerun
         c = ma.ctl()
+        if getattr(self, "_mem_plan", None) is None:
+            from ..schedule.efficient_memory_scheduler import \
+                plan_from_events
+            c.clear_events()
+            c.set_mode(ma.PROFILE)
+            try:
+                flat_outs = self.gm(*prepared)
+                self._writeback(flat_outs)
+                import torch as _t
+                _t.cuda.synchronize()
+            finally:
+                c.set_mode(ma.PASSTHROUGH)
+            entries, arena, stats = plan_from_events(c.get_events())
+            c.load_plan(entries, arena)
+            c.set_mode(ma.RUNTIME)
+            self._mem_plan = stats
+            return self._detach_arena(flat_outs, c)
         c.start_region()
         try:
             flat_outs = self.gm(*prepared)
         finally:
             c.stop_region()
         self._writeback(flat_outs)
-        return flat_outs
+        return self._detach_arena(flat_outs, c)
+
+    def _detach_arena(self, flat_outs, c):
+        """Clone user-visible returns out of the plan arena: the caller
+        may hold them across steps, and an arena address will be re-served
+        next step."""
+        base = c.arena_base()
+        size = c.arena_size()
+        if not base or not size:
+            return flat_outs
+        out = list(flat_outs)
+        ret_positions = getattr(self, "ret_out_positions", None)
+        positions = (range(len(out)) if ret_positions is None
+                     else ret_positions)
+        for i in positions:
+            t = out[i]
+            if isinstance(t, torch.Tensor) and t.is_cuda \
+                    and base <= t.data_ptr() < base + size:
+                out[i] = t.clone()
+        return out
 
     # ------------------------------------------------- state access APIs ----
     def named_parameters(self) -> Dict[str, torch.Tensor]:
