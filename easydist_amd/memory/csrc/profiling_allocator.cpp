@@ -129,14 +129,15 @@ void* ed_malloc(size_t size, int device, hipStream_t stream) {
     std::lock_guard<std::mutex> lk2(g_mu);
     g_records.push_back({g_cur_op, reinterpret_cast<uintptr_t>(p), size,
                          reinterpret_cast<uintptr_t>(stream)});
-    g_events.push_back({1, reinterpret_cast<uintptr_t>(p), size});
+    if (g_in_region)
+      g_events.push_back({1, reinterpret_cast<uintptr_t>(p), size});
   }
   return p;
 }
 
 void ed_free(void* ptr, size_t size, int device, hipStream_t stream) {
   if (ptr == nullptr) return;
-  if (g_mode == Mode::PROFILE) {
+  if (g_mode == Mode::PROFILE && g_in_region) {
     std::lock_guard<std::mutex> lk2(g_mu);
     g_events.push_back({0, reinterpret_cast<uintptr_t>(ptr), 0});
   }

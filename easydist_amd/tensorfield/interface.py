@@ -1,0 +1,114 @@
+"""TFieldClient: python client of the tfield memory-pool server.
+
+Capability parity with reference ``easydist/torch/tensorfield/
+interface.py`` (TFieldClient:18, param-group APIs 56-120). Maps served
+buffers with hipIpcOpenMemHandle via ctypes and moves data with
+hipMemcpy, so tests can verify true cross-process sharing without the
+C allocator in the loop (that client lives in csrc/
+allocator_interface.cpp).
+"""
+from __future__ import annotations
+
+import ctypes
+import socket
+from typing import Optional, Tuple
+
+HIP_IPC_HANDLE_SIZE = 64
+hipMemcpyHostToDevice = 1
+hipMemcpyDeviceToHost = 2
+hipMemcpyDeviceToDevice = 3
+
+
+class _HipClient:
+    def __init__(self):
+        self.lib = ctypes.CDLL("libamdhip64.so")
+        self.lib.hipIpcOpenMemHandle.argtypes = [
+            ctypes.POINTER(ctypes.c_void_p), ctypes.c_char_p,
+            ctypes.c_uint]
+        self.lib.hipIpcCloseMemHandle.argtypes = [ctypes.c_void_p]
+        self.lib.hipMemcpy.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.c_size_t, ctypes.c_int]
+
+    def open_handle(self, handle: bytes) -> int:
+        p = ctypes.c_void_p()
+        # 1 == hipIpcMemLazyEnablePeerAccess
+        rc = self.lib.hipIpcOpenMemHandle(ctypes.byref(p), handle, 1)
+        if rc != 0:
+            raise RuntimeError(f"hipIpcOpenMemHandle rc={rc}")
+        return p.value
+
+    def close_handle(self, ptr: int):
+        self.lib.hipIpcCloseMemHandle(ctypes.c_void_p(ptr))
+
+    def memcpy(self, dst: int, src: int, size: int, kind: int):
+        rc = self.lib.hipMemcpy(ctypes.c_void_p(dst), ctypes.c_void_p(src),
+                                size, kind)
+        if rc != 0:
+            raise RuntimeError(f"hipMemcpy rc={rc}")
+
+
+class TFieldClient:
+    def __init__(self, path: str = "/tmp/easydist_tfield.sock"):
+        self.sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        self.sock.connect(path)
+        self.f = self.sock.makefile("rw")
+        self.hip = _HipClient()
+        import os
+        self._cmd(f"hello {os.getpid()}")
+
+    def _cmd(self, line: str) -> str:
+        self.f.write(line + "\n")
+        self.f.flush()
+        return self.f.readline().strip()
+
+    def alloc(self, size: int) -> Tuple[bytes, int, int]:
+        rep = self._cmd(f"alloc {size}")
+        h, off, sz = rep.split()
+        return bytes.fromhex(h), int(off), int(sz)
+
+    def free(self, handle: bytes):
+        assert self._cmd(f"free {handle.hex()}") == "ok"
+
+    def register_param(self, group: str, name: str, handle: bytes,
+                       offset: int, size: int):
+        assert self._cmd(
+            f"reg {group} {name} {handle.hex()} {offset} {size}") == "ok"
+
+    def get_param(self, group: str, name: str
+                  ) -> Optional[Tuple[bytes, int, int]]:
+        rep = self._cmd(f"get {group} {name}")
+        if rep == "none":
+            return None
+        h, off, sz = rep.split()
+        return bytes.fromhex(h), int(off), int(sz)
+
+    def list_params(self, group: str):
+        rep = self._cmd(f"list {group}")
+        return [s for s in rep.split(",") if s]
+
+    def stat(self):
+        n, b = self._cmd("stat").split()
+        return int(n), int(b)
+
+    def shutdown_server(self):
+        self._cmd("quit")
+
+    # ----------------------------------------------------- data helpers ----
+    def write_bytes(self, handle: bytes, offset: int, data: bytes):
+        ptr = self.hip.open_handle(handle)
+        try:
+            buf = ctypes.create_string_buffer(data, len(data))
+            self.hip.memcpy(ptr + offset, ctypes.addressof(buf), len(data),
+                            hipMemcpyHostToDevice)
+        finally:
+            self.hip.close_handle(ptr)
+
+    def read_bytes(self, handle: bytes, offset: int, size: int) -> bytes:
+        ptr = self.hip.open_handle(handle)
+        try:
+            buf = ctypes.create_string_buffer(size)
+            self.hip.memcpy(ctypes.addressof(buf), ptr + offset, size,
+                            hipMemcpyDeviceToHost)
+            return buf.raw
+        finally:
+            self.hip.close_handle(ptr)
