@@ -116,6 +116,23 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     from .passes.comm_optimize import comm_optimize
     comm_optimize(gm)
 
+    # ---- 5c. re-fuse the decomposed Adam chains into ONE kernel ----------
+    if opt is not None and getattr(mdconfig, "fuse_optimizer", True):
+        from .passes.fuse_optimizer import fuse_optimizer
+        out_node_f = next(n for n in gm.graph.nodes if n.op == "output")
+        flat_outs_f, spec_f = pytree.tree_flatten(out_node_f.args[0])
+        ppos = _adam_positions(params, buffers, named_states)
+        if ppos:
+            nfused = fuse_optimizer(gm, flat_outs_f,
+                                    [n for n in gm.graph.nodes
+                                     if n.op == "placeholder"], ppos, opt)
+            if nfused:
+                out_node_f.args = (pytree.tree_unflatten(flat_outs_f,
+                                                         spec_f),)
+                gm.graph.eliminate_dead_code()
+                gm.graph.lint()
+                gm.recompile()
+
     # strip the pytree codegen: the runtime calls the graph with the flat
     # placeholder list and receives the flat output list
     import torch.fx as _fx
@@ -161,6 +178,41 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
         if hasattr(gm, "_out_spec") else None,
     }
     return compiled
+
+
+def _adam_positions(params, buffers, named_states):
+    """Map each Adam-stepped param to its flat input/output positions.
+
+    Flat layout (pytree insertion order): params..., buffers...,
+    named_states[pname][state_key]... — identical on the input and output
+    side (the traced step returns the same structure)."""
+    param_names = list(params.keys())
+    n_p, n_b = len(params), len(buffers)
+    offset = n_p + n_b
+    out = {}
+    for pname, st in named_states.items():
+        if pname not in param_names:
+            offset += len([v for v in st.values()
+                           if isinstance(v, torch.Tensor)])
+            continue
+        p_pos = param_names.index(pname)
+        entry = {"param": p_pos}
+        for key, val in st.items():
+            if not isinstance(val, torch.Tensor):
+                continue
+            if key == "step":
+                entry["step"] = offset
+                entry["step_in"] = offset
+            elif key == "exp_avg":
+                entry["exp_avg"] = offset
+                entry["exp_avg_in"] = offset
+            elif key == "exp_avg_sq":
+                entry["exp_avg_sq"] = offset
+                entry["exp_avg_sq_in"] = offset
+            offset += 1
+        if {"step", "exp_avg", "exp_avg_sq"} <= set(entry):
+            out[p_pos] = entry
+    return out
 
 
 def _fix_output_reshard(gm, out_pl_env, io_map, ret_names, mesh):
