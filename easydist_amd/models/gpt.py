@@ -123,12 +123,19 @@ class GPT(nn.Module):
 
 
 def gpt_train_step(model, opt, idx, targets, autocast_device=None):
-    """The benchmarked train step: fwd+loss+bwd+adam, bf16 autocast."""
+    """The benchmarked train step: fwd+loss+bwd+adam, bf16 autocast.
+
+    The loss goes through the fused HIP cross-entropy (ops/ce.py): one
+    kernel computes logsumexp+NLL straight from the bf16 logits instead
+    of materializing a [tokens, vocab] fp32 log-softmax (1.6 GB at GPT-2
+    shapes) — lse/softmax reductions still accumulate in fp32 inside the
+    kernel."""
+    from ..ops import ce
     dev = autocast_device or ("cuda" if idx.is_cuda else "cpu")
     with torch.autocast(device_type=dev, dtype=torch.bfloat16):
         logits = model(idx)
-    loss = F.cross_entropy(
-        logits.float().view(-1, logits.size(-1)), targets.reshape(-1))
+    loss = ce.cross_entropy(logits.view(-1, logits.size(-1)),
+                            targets.reshape(-1))
     loss.backward()
     opt.step()
     opt.zero_grad(True)

@@ -21,7 +21,9 @@ def _ce_fwd_aten(logits, targets):
     lf = logits.float()
     lse = torch.logsumexp(lf, dim=-1)
     nll = lse - lf.gather(1, targets.unsqueeze(1)).squeeze(1)
-    return nll.mean(), lse
+    # SUM (not mean): under row sharding the output is PARTIAL(sum), which
+    # ShardCombine discovers and the solver reshards with one all_reduce
+    return nll.sum(), lse
 
 
 def _ce_fwd_cuda(logits, targets):
@@ -37,7 +39,7 @@ def _ce_bwd_aten(grad, logits, targets, lse):
     p.scatter_add_(1, targets.unsqueeze(1),
                    torch.full_like(targets.unsqueeze(1), -1.0,
                                    dtype=p.dtype))
-    return (p * (grad / logits.shape[0])).to(logits.dtype)
+    return (p * grad).to(logits.dtype)
 
 
 def _ce_bwd_cuda(grad, logits, targets, lse):
@@ -81,5 +83,8 @@ torch.library.register_autograd("easydist_amd::ce_fwd", _ce_backward,
 
 
 def cross_entropy(logits, targets):
-    loss, _ = torch.ops.easydist_amd.ce_fwd(logits, targets)
-    return loss
+    """Mean cross-entropy via the SUM kernel: the division by the global
+    token count is a traced scalar, so a row-sharded graph keeps the sum
+    PARTIAL until the output reshard's single all_reduce."""
+    loss_sum, _ = torch.ops.easydist_amd.ce_fwd(logits, targets)
+    return loss_sum / logits.shape[0]

@@ -54,7 +54,7 @@ __global__ void ce_bwd_kernel(const float* __restrict__ grad_scalar,
   if (row >= N) return;
   const T* lrow = logits + (long)row * V;
   T* drow = dlogits + (long)row * V;
-  const float scale = grad_scalar[0] / N;
+  const float scale = grad_scalar[0];   // SUM semantics: caller scales
   const float l = lse[row];
   const long tg = targets[row];
   for (int i = lane; i < V; i += WAVE) {
@@ -85,7 +85,7 @@ std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
         logits.data_ptr<float>(), targets.data_ptr<long>(),
         nll.data_ptr<float>(), lse.data_ptr<float>(), N, V);
   }
-  return {nll.mean(), lse};
+  return {nll.sum(), lse};
 }
 
 at::Tensor ce_bwd(const at::Tensor& grad, const at::Tensor& logits,
