@@ -84,6 +84,11 @@ def _bwd_cpu(grad, q, k, v, out, lse, causal):
 
 
 def _bwd_cuda(grad, q, k, v, out, lse, causal):
+    ext = load_extension()
+    if ext is not None and _kernel_ok(q, k, v):
+        q, k, v, out = (t.contiguous() for t in (q, k, v, out))
+        return ext.flash_attn_bwd(grad.contiguous(), q, k, v, out,
+                                  lse.contiguous(), causal)
     return _math_bwd(grad, q, k, v, out, lse, causal)
 
 

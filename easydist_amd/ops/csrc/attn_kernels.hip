@@ -213,3 +213,355 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
         lse.data_ptr<float>(), B, H, S, causal, scale);
   return {out, lse};
 }
+
+// ---------------------------------------------------------------------------
+// Flash attention backward: two kernels, no atomics.
+//   dq kernel: one workgroup per 64-row Q block; recomputes P from lse,
+//     dP = dO V^T, dS = P (dP - delta) scale, dq += dS K.
+//   dkv kernel: one workgroup per 64-key KV block; recomputes P^T,
+//     dV += P^T dO, dP^T = V dO^T, dS^T = P^T (dP^T - delta) scale,
+//     dK += dS^T Q.
+// delta = rowsum(dO * O) comes precomputed from the host (one fused
+// elementwise+reduce).
+// mfma contract (as in fwd): D[m][n] += sum_k A[m][k] B[n][k]; A/B lanes
+// hold row fr, k-slice fg; C/D lanes hold row 4*fg+r, col fr.
+// ---------------------------------------------------------------------------
+
+template <int D>
+__global__ void __launch_bounds__(256)
+flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
+                    const bf16* __restrict__ K, const bf16* __restrict__ V,
+                    const float* __restrict__ LSE,
+                    const float* __restrict__ DELTA, bf16* __restrict__ DQ,
+                    int B, int H, int S, bool causal, float scale) {
+  const int qb0 = blockIdx.x * QB;
+  const int bh = blockIdx.y;
+  const long base = (long)bh * S * D;
+  const bf16* dO_ = dO + base;
+  const bf16* q = Q + base;
+  const bf16* k = K + base;
+  const bf16* v = V + base;
+  bf16* dq = DQ + base;
+  const float* lse = LSE + (long)bh * S;
+  const float* delta = DELTA + (long)bh * S;
+
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int qr0 = qb0 + wave * 16;
+  const int fr = lane & 15;
+  const int fg = lane >> 4;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* k_lds = reinterpret_cast<bf16*>(smem);
+  bf16* v_lds = k_lds + KB * D;
+  bf16* s_lds = v_lds + KB * D + wave * 16 * KB;   // wave-private strip
+
+  // A-operand fragments for this wave's 16 q rows
+  bf16x8v qf[D / 32], dof[D / 32];
+  #pragma unroll
+  for (int ks = 0; ks < D / 32; ++ks) {
+    qf[ks] = *reinterpret_cast<const bf16x8v*>(
+        &q[(long)(qr0 + fr) * D + ks * 32 + fg * 8]);
+    dof[ks] = *reinterpret_cast<const bf16x8v*>(
+        &dO_[(long)(qr0 + fr) * D + ks * 32 + fg * 8]);
+  }
+  // per C-row lse/delta (rows 4*fg+r of this wave)
+  float lse_r[4], dlt_r[4];
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int qrow = qr0 + 4 * fg + r;
+    lse_r[r] = lse[qrow];
+    dlt_r[r] = delta[qrow];
+  }
+
+  f32x4 dq_acc[D / 16];
+  #pragma unroll
+  for (int j = 0; j < D / 16; ++j) dq_acc[j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = causal ? min(S, qb0 + QB) : S;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    __syncthreads();
+    for (int e = threadIdx.x * 8; e < KB * D; e += 256 * 8) {
+      *reinterpret_cast<bf16x8*>(&k_lds[e]) =
+          *reinterpret_cast<const bf16x8*>(&k[(long)kv0 * D + e]);
+      *reinterpret_cast<bf16x8*>(&v_lds[e]) =
+          *reinterpret_cast<const bf16x8*>(&v[(long)kv0 * D + e]);
+    }
+    __syncthreads();
+
+    // S = Q K^T and dP = dO V^T (both mfma-native: B = rows)
+    f32x4 s_acc[KB / 16], dp_acc[KB / 16];
+    #pragma unroll
+    for (int j = 0; j < KB / 16; ++j) {
+      s_acc[j] = {0.f, 0.f, 0.f, 0.f};
+      dp_acc[j] = {0.f, 0.f, 0.f, 0.f};
+      #pragma unroll
+      for (int ks = 0; ks < D / 32; ++ks) {
+        bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
+            &k_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+        bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
+            &v_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+        s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], kf,
+                                                           s_acc[j], 0, 0, 0);
+        dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[ks], vf,
+                                                            dp_acc[j], 0, 0,
+                                                            0);
+      }
+    }
+    // dS = P * (dP - delta) * scale, P = exp(S*scale - lse)
+    #pragma unroll
+    for (int j = 0; j < KB / 16; ++j) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int kcol = kv0 + j * 16 + fr;
+        int qrow = qr0 + 4 * fg + r;
+        float p = 0.f;
+        if (!(causal && kcol > qrow) && kcol < S)
+          p = __expf(s_acc[j][r] * scale - lse_r[r]);
+        s_acc[j][r] = p * (dp_acc[j][r] - dlt_r[r]) * scale;
+      }
+    }
+    // re-fragment dS through the wave strip
+    #pragma unroll
+    for (int j = 0; j < KB / 16; ++j)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        s_lds[(4 * fg + r) * KB + j * 16 + fr] = f2bf(s_acc[j][r]);
+    __builtin_amdgcn_s_waitcnt(0);
+    bf16x8v dsf[KB / 32];
+    #pragma unroll
+    for (int ks = 0; ks < KB / 32; ++ks)
+      dsf[ks] = *reinterpret_cast<const bf16x8v*>(
+          &s_lds[fr * KB + ks * 32 + fg * 8]);
+    // dq += dS @ K: B[n=d][k=c] = K[c][d] (column reads)
+    #pragma unroll
+    for (int j = 0; j < D / 16; ++j) {
+      #pragma unroll
+      for (int ks = 0; ks < KB / 32; ++ks) {
+        bf16x8v kcol;
+        #pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          kcol[jj] = *reinterpret_cast<const __bf16*>(
+              &k_lds[(ks * 32 + fg * 8 + jj) * D + j * 16 + fr]);
+        dq_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf[ks], kcol,
+                                                            dq_acc[j], 0, 0,
+                                                            0);
+      }
+    }
+  }
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int qrow = qr0 + 4 * fg + r;
+    if (qrow >= S) continue;
+    #pragma unroll
+    for (int j = 0; j < D / 16; ++j)
+      dq[(long)qrow * D + j * 16 + fr] = f2bf(dq_acc[j][r]);
+  }
+}
+
+template <int D>
+__global__ void __launch_bounds__(256)
+flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
+                     const bf16* __restrict__ K, const bf16* __restrict__ V,
+                     const float* __restrict__ LSE,
+                     const float* __restrict__ DELTA, bf16* __restrict__ DK,
+                     bf16* __restrict__ DV, int B, int H, int S, bool causal,
+                     float scale) {
+  const int kb0 = blockIdx.x * KB;
+  const int bh = blockIdx.y;
+  const long base = (long)bh * S * D;
+  const bf16* dO_ = dO + base;
+  const bf16* q = Q + base;
+  const bf16* k = K + base;
+  const bf16* v = V + base;
+  bf16* dk = DK + base;
+  bf16* dv = DV + base;
+  const float* lse = LSE + (long)bh * S;
+  const float* delta = DELTA + (long)bh * S;
+
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int kr0 = kb0 + wave * 16;          // this wave's first key row
+  const int fr = lane & 15;
+  const int fg = lane >> 4;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* q_lds = reinterpret_cast<bf16*>(smem);
+  bf16* do_lds = q_lds + QB * D;
+  bf16* s_lds = do_lds + QB * D + wave * 16 * QB;
+
+  bf16x8v kf[D / 32], vf[D / 32];
+  #pragma unroll
+  for (int ks = 0; ks < D / 32; ++ks) {
+    kf[ks] = *reinterpret_cast<const bf16x8v*>(
+        &k[(long)(kr0 + fr) * D + ks * 32 + fg * 8]);
+    vf[ks] = *reinterpret_cast<const bf16x8v*>(
+        &v[(long)(kr0 + fr) * D + ks * 32 + fg * 8]);
+  }
+
+  f32x4 dk_acc[D / 16], dv_acc[D / 16];
+  #pragma unroll
+  for (int j = 0; j < D / 16; ++j) {
+    dk_acc[j] = {0.f, 0.f, 0.f, 0.f};
+    dv_acc[j] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int q_start = causal ? kb0 : 0;
+  for (int q0 = q_start; q0 < S; q0 += QB) {
+    __syncthreads();
+    for (int e = threadIdx.x * 8; e < QB * D; e += 256 * 8) {
+      *reinterpret_cast<bf16x8*>(&q_lds[e]) =
+          *reinterpret_cast<const bf16x8*>(&q[(long)q0 * D + e]);
+      *reinterpret_cast<bf16x8*>(&do_lds[e]) =
+          *reinterpret_cast<const bf16x8*>(&dO_[(long)q0 * D + e]);
+    }
+    __syncthreads();
+
+    // S^T = K Q^T and dP^T = V dO^T (B operands: row reads from LDS)
+    f32x4 st_acc[QB / 16], dpt_acc[QB / 16];
+    #pragma unroll
+    for (int j = 0; j < QB / 16; ++j) {
+      st_acc[j] = {0.f, 0.f, 0.f, 0.f};
+      dpt_acc[j] = {0.f, 0.f, 0.f, 0.f};
+      #pragma unroll
+      for (int ks = 0; ks < D / 32; ++ks) {
+        bf16x8v qfb = *reinterpret_cast<const bf16x8v*>(
+            &q_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+        bf16x8v dob = *reinterpret_cast<const bf16x8v*>(
+            &do_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+        st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[ks], qfb,
+                                                            st_acc[j], 0, 0,
+                                                            0);
+        dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[ks], dob,
+                                                             dpt_acc[j], 0,
+                                                             0, 0);
+      }
+    }
+    // P^T = exp(S^T*scale - lse[qcol]); dS^T = P^T (dP^T - delta[qcol]) scale
+    // first pass: P^T into the strip for the dV mfma
+    #pragma unroll
+    for (int j = 0; j < QB / 16; ++j) {
+      int qcol = q0 + j * 16 + fr;
+      float l = lse[qcol];
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int krow = kr0 + 4 * fg + r;
+        float p = 0.f;
+        if (!(causal && krow > qcol) && qcol < S)
+          p = __expf(st_acc[j][r] * scale - l);
+        st_acc[j][r] = p;
+        s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(p);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    bf16x8v ptf[QB / 32];
+    #pragma unroll
+    for (int ks = 0; ks < QB / 32; ++ks)
+      ptf[ks] = *reinterpret_cast<const bf16x8v*>(
+          &s_lds[fr * QB + ks * 32 + fg * 8]);
+    // dV += P^T @ dO: B[n=d][k=q] = dO[q][d] (column reads)
+    #pragma unroll
+    for (int j = 0; j < D / 16; ++j) {
+      #pragma unroll
+      for (int ks = 0; ks < QB / 32; ++ks) {
+        bf16x8v docol;
+        #pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          docol[jj] = *reinterpret_cast<const __bf16*>(
+              &do_lds[(ks * 32 + fg * 8 + jj) * D + j * 16 + fr]);
+        dv_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf[ks], docol,
+                                                            dv_acc[j], 0, 0,
+                                                            0);
+      }
+    }
+    // dS^T into the strip for the dK mfma
+    __builtin_amdgcn_s_waitcnt(0);
+    #pragma unroll
+    for (int j = 0; j < QB / 16; ++j) {
+      int qcol = q0 + j * 16 + fr;
+      float dlt = delta[qcol];
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float ds = st_acc[j][r] * (dpt_acc[j][r] - dlt) * scale;
+        s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(ds);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    bf16x8v dstf[QB / 32];
+    #pragma unroll
+    for (int ks = 0; ks < QB / 32; ++ks)
+      dstf[ks] = *reinterpret_cast<const bf16x8v*>(
+          &s_lds[fr * QB + ks * 32 + fg * 8]);
+    // dK += dS^T @ Q: B[n=d][k=q] = Q[q][d] (column reads)
+    #pragma unroll
+    for (int j = 0; j < D / 16; ++j) {
+      #pragma unroll
+      for (int ks = 0; ks < QB / 32; ++ks) {
+        bf16x8v qcolf;
+        #pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          qcolf[jj] = *reinterpret_cast<const __bf16*>(
+              &q_lds[(ks * 32 + fg * 8 + jj) * D + j * 16 + fr]);
+        dk_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf[ks], qcolf,
+                                                            dk_acc[j], 0, 0,
+                                                            0);
+      }
+    }
+  }
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int krow = kr0 + 4 * fg + r;
+    if (krow >= S) continue;
+    #pragma unroll
+    for (int j = 0; j < D / 16; ++j) {
+      dk[(long)krow * D + j * 16 + fr] = f2bf(dk_acc[j][r]);
+      dv[(long)krow * D + j * 16 + fr] = f2bf(dv_acc[j][r]);
+    }
+  }
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor>
+flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
+               const at::Tensor& k, const at::Tensor& v,
+               const at::Tensor& out, const at::Tensor& lse, bool causal) {
+  TORCH_CHECK(q.dtype() == at::kBFloat16 && q.dim() == 4);
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.sizes() == k.sizes() && q.sizes() == v.sizes());
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  TORCH_CHECK(D == 64 || D == 128, "flash_attn_bwd: D in {64,128}");
+  TORCH_CHECK(S % QB == 0, "flash_attn_bwd: S multiple of 64");
+  auto gradc = grad.contiguous();
+  // delta = rowsum(dO * O), fp32
+  auto delta = (gradc.to(at::kFloat) * out.to(at::kFloat)).sum(-1);
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  dim3 grid(S / QB, B * H), block(256);
+  size_t lds = (2 * KB * (size_t)D + 4 * 16 * KB) * 2;
+  float scale = 1.f / sqrtf((float)D);
+  if (D == 64) {
+    hipLaunchKernelGGL(flash_bwd_dq_kernel<64>, grid, block, lds, stream,
+        (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
+        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
+        lse.data_ptr<float>(), delta.data_ptr<float>(),
+        (bf16*)dq.data_ptr(), B, H, S, causal, scale);
+    hipLaunchKernelGGL(flash_bwd_dkv_kernel<64>, grid, block, lds, stream,
+        (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
+        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
+        lse.data_ptr<float>(), delta.data_ptr<float>(),
+        (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(), B, H, S, causal, scale);
+  } else {
+    hipLaunchKernelGGL(flash_bwd_dq_kernel<128>, grid, block, lds, stream,
+        (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
+        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
+        lse.data_ptr<float>(), delta.data_ptr<float>(),
+        (bf16*)dq.data_ptr(), B, H, S, causal, scale);
+    hipLaunchKernelGGL(flash_bwd_dkv_kernel<128>, grid, block, lds, stream,
+        (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
+        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
+        lse.data_ptr<float>(), delta.data_ptr<float>(),
+        (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(), B, H, S, causal, scale);
+  }
+  return {dq, dk, dv};
+}

@@ -31,6 +31,10 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
                                                   const at::Tensor& k,
                                                   const at::Tensor& v,
                                                   bool causal);
+std::tuple<at::Tensor, at::Tensor, at::Tensor>
+flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
+               const at::Tensor& k, const at::Tensor& v,
+               const at::Tensor& out, const at::Tensor& lse, bool causal);
 
 // wrappers adapting std::vector<bool> / TensorList signatures
 static std::tuple<at::Tensor, at::Tensor, at::Tensor>
@@ -70,6 +74,8 @@ TORCH_LIBRARY(easydist_amd_hip, m) {
   m.def("gemm_nt(Tensor a, Tensor bt, Tensor? bias) -> Tensor");
   m.def("flash_attn_fwd(Tensor q, Tensor k, Tensor v, bool causal) "
         "-> (Tensor, Tensor)");
+  m.def("flash_attn_bwd(Tensor grad, Tensor q, Tensor k, Tensor v, "
+        "Tensor out, Tensor lse, bool causal) -> (Tensor, Tensor, Tensor)");
 }
 
 TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
@@ -82,6 +88,7 @@ TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
   m.impl("fused_adam_step", adam_wrap);
   m.impl("gemm_nt", gemm_nt);
   m.impl("flash_attn_fwd", flash_attn_fwd);
+  m.impl("flash_attn_bwd", flash_attn_bwd);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {}

@@ -156,3 +156,25 @@ def test_flash_attn_fwd(ext):
         s = s.masked_fill(~mask, float("-inf"))
         lref = torch.logsumexp(s, -1)
         assert torch.allclose(lse, lref, atol=1e-2, rtol=1e-3)
+
+
+@requires_gpu
+def test_flash_attn_bwd(ext):
+    torch.manual_seed(5)
+    for (B, H, S, D) in [(2, 3, 128, 64), (1, 2, 256, 128)]:
+        q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        g = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        out, lse = ext.flash_attn_fwd(q, k, v, True)
+        dq, dk, dv = ext.flash_attn_bwd(g, q, k, v, out, lse, True)
+        qf = q.float().requires_grad_(True)
+        kf = k.float().requires_grad_(True)
+        vf = v.float().requires_grad_(True)
+        ref = torch.nn.functional.scaled_dot_product_attention(
+            qf, kf, vf, is_causal=True)
+        ref.backward(g.float())
+        for got, want, name in ((dq, qf.grad, "dq"), (dk, kf.grad, "dk"),
+                                (dv, vf.grad, "dv")):
+            err = (got.float() - want).abs().max()
+            assert err < 8e-2, (name, float(err), B, H, S, D)
