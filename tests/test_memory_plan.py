@@ -51,3 +51,20 @@ def test_pack_random_no_overlap_and_saves():
                                          addr[(y.node_name, 0)], y)
     naive = sum(b.size for b in bufs)
     assert peak < naive * 0.6     # interleaved lifetimes must reuse memory
+
+
+def test_ilp_pack_at_least_as_good_as_skyline():
+    from easydist_amd.schedule.ilp_memory_scheduler import ilp_pack_buffers
+    bufs = [Buffer("a", 0, 1024, 0, 3, False),
+            Buffer("b", 0, 1024, 1, 2, False),
+            Buffer("c", 0, 2048, 2, 5, False),
+            Buffer("d", 0, 512, 4, 6, False),
+            Buffer("e", 0, 1024, 6, 8, False)]
+    addr_h, peak_h = pack_buffers(bufs)
+    addr_i, peak_i = ilp_pack_buffers(bufs)
+    assert peak_i <= peak_h
+    for i, x in enumerate(bufs):
+        for y in bufs[i + 1:]:
+            if _overlap_time(x, y):
+                assert not _overlap_addr(addr_i[(x.node_name, 0)], x,
+                                         addr_i[(y.node_name, 0)], y)
