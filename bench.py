@@ -31,7 +31,11 @@ def main():
     p.add_argument("--per-gpu-batch", type=int, default=8)
     p.add_argument("--seq", type=int, default=1024)
     p.add_argument("--parallel", default="auto")
-    p.add_argument("--no-hipgraph", action="store_true")
+    # eager is currently faster than hipGraph capture+replay: the fused
+    # multi-tensor Adam kernel is capture-unsafe (chunk-table upload) and
+    # capture falls back to the decomposed per-param math
+    p.add_argument("--hipgraph", action="store_true")
+    p.add_argument("--no-hipgraph", action="store_true")   # back-compat
     args = p.parse_args()
 
     import torch.distributed as dist
@@ -79,7 +83,7 @@ def main():
         return gptm.gpt_train_step(model, opt, idx, targets)
 
     compiled = easydist_compile(train_step, parallel_mode=args.parallel,
-                                cuda_graph=not args.no_hipgraph)
+                                cuda_graph=args.hipgraph)
 
     def batch():
         g = torch.Generator(device="cpu").manual_seed(7)   # same on all ranks
