@@ -20,7 +20,7 @@ hipMemcpyDeviceToDevice = 3
 
 
 class _HipClient:
-    def __init__(self):
+    def __init__(self, device: int = 0):
         self.lib = ctypes.CDLL("libamdhip64.so")
         self.lib.hipIpcOpenMemHandle.argtypes = [
             ctypes.POINTER(ctypes.c_void_p), ctypes.c_char_p,
@@ -28,6 +28,11 @@ class _HipClient:
         self.lib.hipIpcCloseMemHandle.argtypes = [ctypes.c_void_p]
         self.lib.hipMemcpy.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                        ctypes.c_size_t, ctypes.c_int]
+        self.lib.hipSetDevice.argtypes = [ctypes.c_int]
+        # opening an IPC handle needs an active device context
+        rc = self.lib.hipSetDevice(device)
+        if rc != 0:
+            raise RuntimeError(f"hipSetDevice({device}) rc={rc}")
 
     def open_handle(self, handle: bytes) -> int:
         p = ctypes.c_void_p()
