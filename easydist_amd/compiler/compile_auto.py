@@ -68,45 +68,59 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
         if hasattr(o, "name"):
             ret_names.add(o.name)
 
-    # ---- 3. rule discovery ----------------------------------------------
-    t0 = time.time()
-    sharding_info = EDTorchShardingAnn(gm, device=device).run()
-    search_time = time.time() - t0
-    logger.info("sharding discovery: %d annotated ops (%.2fs)",
-                len(sharding_info), search_time)
+    # ---- 3+4. discovery + solve (on-disk cache, reference
+    # compile_auto.py:97-106) -------------------------------------------
+    cache_path = _strategy_cache_path(gm, mesh)
+    cached = _load_strategy_cache(cache_path)
+    if cached is not None:
+        strategies_per_dim, search_time, solve_time = cached
+        logger.info("strategy cache hit: %s", cache_path)
+    else:
+        t0 = time.time()
+        sharding_info = EDTorchShardingAnn(gm, device=device).run()
+        search_time = time.time() - t0
+        logger.info("sharding discovery: %d annotated ops (%.2fs)",
+                    len(sharding_info), search_time)
 
-    # ---- 4. meta graph + per-mesh-dim solve ------------------------------
-    meta_graph, output_constraints, var_of = fx2meta_graph(
-        gm, sharding_info, io_map, ret_names)
-    spmd_dims = mesh.spmd_dims()
-    strategies_per_dim: List[Dict] = []
-    already_sharded: Dict[str, Dict[int, int]] = {}
-    t0 = time.time()
-    for mesh_dim in range(mesh.ndim):
-        size = mesh.size(mesh_dim)
-        if mesh_dim not in spmd_dims or size == 1:
-            strategies_per_dim.append({})
-            continue
-        clusters = meta_graph.coarsen(
-            mdconfig.coarsen_level if mdconfig.enable_graph_coarsen else 0)
-        solver = AutoFlowSolver1D(meta_graph, size, already_sharded,
-                                  output_constraints)
-        solver.add_coarsen_graph(clusters)
-        choice = (solver.beam_search() if mdconfig.solver_mode == "beam"
-                  else solver.ilp_solve())
-        node_strats: Dict = {}
-        for st in choice.values():
-            node_strats.update(st.node_strategies)
-        strategies_per_dim.append(node_strats)
-        # update already_sharded for the next dim
-        for st in choice.values():
-            for v, pl in st.out_placements.items():
-                if pl.is_shard():
-                    already_sharded.setdefault(v, {})
-                    already_sharded[v][pl.dim] = \
-                        already_sharded[v].get(pl.dim, 1) * size
-    solve_time = time.time() - t0
-    logger.info("strategy solve: %.2fs (search %.2fs)", solve_time, search_time)
+        meta_graph, output_constraints, var_of = fx2meta_graph(
+            gm, sharding_info, io_map, ret_names)
+        spmd_dims = mesh.spmd_dims()
+        strategies_per_dim = []
+        already_sharded: Dict[str, Dict[int, int]] = {}
+        t0 = time.time()
+        for mesh_dim in range(mesh.ndim):
+            size = mesh.size(mesh_dim)
+            if mesh_dim not in spmd_dims or size == 1:
+                strategies_per_dim.append({})
+                continue
+            clusters = meta_graph.coarsen(
+                mdconfig.coarsen_level if mdconfig.enable_graph_coarsen
+                else 0)
+            solver = AutoFlowSolver1D(meta_graph, size, already_sharded,
+                                      output_constraints)
+            solver.add_coarsen_graph(clusters)
+            choice = (solver.beam_search() if mdconfig.solver_mode == "beam"
+                      else solver.ilp_solve())
+            node_strats: Dict = {}
+            for st in choice.values():
+                node_strats.update(st.node_strategies)
+            strategies_per_dim.append(node_strats)
+            # update already_sharded for the next dim
+            for st in choice.values():
+                for v, pl in st.out_placements.items():
+                    if pl.is_shard():
+                        already_sharded.setdefault(v, {})
+                        already_sharded[v][pl.dim] = \
+                            already_sharded[v].get(pl.dim, 1) * size
+        solve_time = time.time() - t0
+        logger.info("strategy solve: %.2fs (search %.2fs)", solve_time,
+                    search_time)
+        _save_strategy_cache(cache_path,
+                             (strategies_per_dim, search_time, solve_time))
+
+    from ..utils.dumps import dump_graph, dump_strategies
+    dump_strategies(strategies_per_dim, "auto")
+    dump_graph(gm, "auto_pre_shard")
 
     # ---- 5. sharding transform ------------------------------------------
     gm, out_pl_env = sharding_transform(gm, strategies_per_dim, mesh.shape)
@@ -115,6 +129,10 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     # ---- 5b. comm overlap: widen every start/wait window -----------------
     from .passes.comm_optimize import comm_optimize
     comm_optimize(gm)
+
+    # ---- 5b2. lower hot aten ops to the gfx950 kernels -------------------
+    from .passes.lower_hip import lower_layer_norm
+    lower_layer_norm(gm)
 
     # ---- 5c. re-fuse the decomposed Adam chains into ONE kernel ----------
     if opt is not None and getattr(mdconfig, "fuse_optimizer", True):
@@ -132,6 +150,8 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
                 gm.graph.eliminate_dead_code()
                 gm.graph.lint()
                 gm.recompile()
+
+    dump_graph(gm, "auto_sharded")
 
     # strip the pytree codegen: the runtime calls the graph with the flat
     # placeholder list and receives the flat output list
@@ -178,6 +198,45 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
         if hasattr(gm, "_out_spec") else None,
     }
     return compiled
+
+
+def _strategy_cache_path(gm, mesh) -> Optional[str]:
+    """Strategy cache keyed by (graph text, mesh shape). reference:
+    compile_auto.py:97-106 (pickle keyed by input signature)."""
+    import hashlib
+    import os
+    if not mdconfig.enable_compile_cache:
+        return None
+    h = hashlib.sha256()
+    h.update(str(gm.graph).encode())
+    h.update(repr(mesh.shape).encode())
+    d = os.path.join(os.path.expanduser("~"), ".easydist_amd",
+                     "compile_cache")
+    os.makedirs(d, exist_ok=True)
+    return os.path.join(d, h.hexdigest()[:24] + ".pkl")
+
+
+def _load_strategy_cache(path):
+    import os
+    import pickle
+    if path is None or not os.path.exists(path):
+        return None
+    try:
+        with open(path, "rb") as f:
+            return pickle.load(f)
+    except Exception:
+        return None
+
+
+def _save_strategy_cache(path, payload):
+    import pickle
+    if path is None:
+        return
+    try:
+        with open(path, "wb") as f:
+            pickle.dump(payload, f)
+    except Exception as e:   # noqa: BLE001
+        logger.warning("strategy cache write failed: %s", e)
 
 
 def _adam_positions(params, buffers, named_states):

@@ -1,0 +1,139 @@
+"""Lower hot aten ops in the sharded graph to the gfx950 HIP kernels.
+
+The autocast'd trace runs LayerNorm in fp32: every norm is wrapped in
+bf16->fp32->bf16 casts (rocprof: ~290 cast kernels + aten layer_norm +
+layer_norm_grad_input per GPT-2 step). This pass rewrites
+``aten.native_layer_norm(_backward)`` to ``easydist_amd::layer_norm_fwd/
+bwd`` (one-pass Welford / two-pass bwd HIP kernels that take bf16
+directly with fp32 statistics inside), FOLDING the surrounding casts:
+
+* an input that is ``_to_copy(x_bf16, fp32)`` feeds the kernel as x_bf16;
+* a user that is ``_to_copy(out, bf16)`` reads the kernel's bf16 output
+  directly;
+* any remaining consumer dtype mismatch gets an explicit cast so graph
+  semantics never change.
+"""
+from __future__ import annotations
+
+import logging
+import operator
+from typing import Optional
+
+import torch
+import torch.fx as fx
+
+from ...ops import norms as _norms  # noqa: F401  (registers the ops)
+
+logger = logging.getLogger(__name__)
+
+aten = torch.ops.aten
+CAST_OPS = (aten._to_copy.default, aten.to.dtype)
+
+
+def _val(n):
+    return n.meta.get("val") if hasattr(n, "meta") else None
+
+
+def _uncast(n: fx.Node, want: torch.dtype) -> Optional[fx.Node]:
+    """If n is a cast of a `want`-dtype value, return the source."""
+    if isinstance(n, fx.Node) and n.op == "call_function" \
+            and n.target in CAST_OPS:
+        src = n.args[0]
+        v = _val(src)
+        if isinstance(v, torch.Tensor) and v.dtype == want:
+            return src
+    return None
+
+
+def _dtype_of(n) -> Optional[torch.dtype]:
+    v = _val(n)
+    return v.dtype if isinstance(v, torch.Tensor) else None
+
+
+def lower_layer_norm(gm: fx.GraphModule) -> int:
+    graph = gm.graph
+    n_lowered = 0
+    for n in list(graph.nodes):
+        if n.op != "call_function":
+            continue
+        if n.target is aten.native_layer_norm.default:
+            x, shape, w, b, eps = n.args
+            if not isinstance(shape, (list, tuple)) or len(shape) != 1 \
+                    or shape[0] % 8 != 0:
+                continue
+            x_bf = _uncast(x, torch.bfloat16)
+            src = x_bf if x_bf is not None else x
+            with graph.inserting_before(n):
+                new = graph.call_function(
+                    torch.ops.easydist_amd.layer_norm_fwd.default,
+                    (src, w, b, float(eps)))
+            _rewire_tuple(graph, n, new,
+                          kernel_dtypes=[_dtype_of(src), torch.float32,
+                                         torch.float32])
+            n_lowered += 1
+        elif n.target is aten.native_layer_norm_backward.default:
+            grad, x, shape, mean, rstd, w, b, mask = n.args
+            if not isinstance(shape, (list, tuple)) or len(shape) != 1 \
+                    or shape[0] % 8 != 0:
+                continue
+            g_bf = _uncast(grad, torch.bfloat16)
+            x_bf = _uncast(x, torch.bfloat16)
+            if _dtype_of(x) == torch.bfloat16:
+                x_bf = x
+            if _dtype_of(grad) == torch.bfloat16:
+                g_bf = grad
+            if g_bf is None or x_bf is None:
+                continue      # kernel is bf16-x only
+            with graph.inserting_before(n):
+                new = graph.call_function(
+                    torch.ops.easydist_amd.layer_norm_bwd.default,
+                    (g_bf, x_bf, mean, rstd, w, list(mask)))
+            _rewire_tuple(graph, n, new,
+                          kernel_dtypes=[torch.bfloat16, torch.float32,
+                                         torch.float32])
+            n_lowered += 1
+    if n_lowered:
+        graph.lint()
+        gm.recompile()
+        logger.info("lower_hip: lowered %d layer_norm nodes", n_lowered)
+    return n_lowered
+
+
+def _dtype_of_item(tuple_node: fx.Node, k: int) -> Optional[torch.dtype]:
+    for u in tuple_node.users:
+        if u.op == "call_function" and u.target is operator.getitem \
+                and u.args[1] == k:
+            return _dtype_of(u)
+    return None
+
+
+def _rewire_tuple(graph: fx.Graph, old: fx.Node, new: fx.Node,
+                  kernel_dtypes):
+    """Point every getitem(old, k) at getitem(new, k). A consumer that is
+    a cast TO the kernel's output dtype folds away; consumers expecting
+    the old aten dtype get one explicit cast so semantics never change."""
+    for u in list(old.users):
+        if not (u.op == "call_function" and u.target is operator.getitem):
+            continue
+        k = u.args[1]
+        with graph.inserting_before(u):
+            item = graph.call_function(operator.getitem, (new, k))
+        want = kernel_dtypes[k] if k < len(kernel_dtypes) else None
+        old_dtype = _dtype_of(u)
+        for uu in list(u.users):
+            if uu.op == "call_function" and uu.target in CAST_OPS and \
+                    _dtype_of(uu) == want:
+                uu.replace_all_uses_with(item)
+                graph.erase_node(uu)
+        if list(u.users):
+            if old_dtype is not None and want is not None \
+                    and old_dtype != want:
+                with graph.inserting_before(u):
+                    cast = graph.call_function(
+                        aten._to_copy.default, (item,),
+                        {"dtype": old_dtype})
+                u.replace_all_uses_with(cast)
+            else:
+                u.replace_all_uses_with(item)
+        graph.erase_node(u)
+    graph.erase_node(old)

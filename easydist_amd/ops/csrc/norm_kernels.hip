@@ -275,7 +275,7 @@ layer_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
       stream, (const bf16*)grad.data_ptr(), (const bf16*)x.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(), dwf.data_ptr<float>(),
       dbf.data_ptr<float>(), rows, D, rows_per_blk);
-  auto dtype = w ? w->scalar_type() : x.scalar_type();
+  auto dtype = at::kFloat;   // grads feed fp32 Adam: never truncate
   return {dx, dwf.to(dtype), dbf.to(dtype)};
 }
 
@@ -333,6 +333,6 @@ rms_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
       stream, (const bf16*)grad.data_ptr(), (const bf16*)x.data_ptr(),
       nullptr, rstd.data_ptr<float>(), dwf.data_ptr<float>(),
       dbf.data_ptr<float>(), rows, D, rows_per_blk);
-  auto dtype = w ? w->scalar_type() : x.scalar_type();
+  auto dtype = at::kFloat;   // grads feed fp32 Adam: never truncate
   return {dx, dwf.to(dtype)};
 }

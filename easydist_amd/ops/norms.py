@@ -32,10 +32,10 @@ def _ln_fwd_cuda(x, w, b, eps):
     ext = load_extension()
     if ext is not None and x.dtype in (torch.bfloat16, torch.float32) \
             and x.shape[-1] % 8 == 0:
-        return ext.layer_norm_fwd(x.contiguous(),
-                                  w.contiguous() if w is not None else None,
-                                  b.contiguous() if b is not None else None,
-                                  eps)
+        # the kernel reads w/b as x's dtype
+        w = w.to(x.dtype).contiguous() if w is not None else None
+        b = b.to(x.dtype).contiguous() if b is not None else None
+        return ext.layer_norm_fwd(x.contiguous(), w, b, eps)
     return _ln_fwd_aten(x, w, b, eps)
 
 
@@ -47,12 +47,12 @@ def _ln_bwd_aten(grad, x, mean, rstd, w, mask):
 
 def _ln_bwd_cuda(grad, x, mean, rstd, w, mask):
     ext = load_extension()
-    if ext is not None and x.dtype in (torch.bfloat16, torch.float32) \
+    if ext is not None and x.dtype == torch.bfloat16 \
             and x.shape[-1] % 8 == 0:
-        return ext.layer_norm_bwd(grad.contiguous(), x.contiguous(), mean,
-                                  rstd,
-                                  w.contiguous() if w is not None else None,
-                                  list(mask))
+        w2 = w.to(x.dtype).contiguous() if w is not None else None
+        return ext.layer_norm_bwd(grad.to(x.dtype).contiguous(),
+                                  x.contiguous(), mean.contiguous(),
+                                  rstd.contiguous(), w2, list(mask))
     return _ln_bwd_aten(grad, x, mean, rstd, w, mask)
 
 
