@@ -68,8 +68,91 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
         if hasattr(o, "name"):
             ret_names.add(o.name)
 
-    # ---- 3+4. discovery + solve (on-disk cache, reference
-    # compile_auto.py:97-106) -------------------------------------------
+    gm, out_pl_env, search_time, solve_time = shard_graph(
+        gm, mesh, io_map, ret_names, device)
+
+    # ---- 5b. comm overlap: widen every start/wait window -----------------
+    from .passes.comm_optimize import comm_optimize
+    comm_optimize(gm)
+
+    # ---- 5b2. lower hot aten ops to the gfx950 kernels -------------------
+    from .passes.lower_hip import lower_layer_norm
+    lower_layer_norm(gm)
+
+    # ---- 5c. re-fuse the decomposed Adam chains into ONE kernel ----------
+    if opt is not None and getattr(mdconfig, "fuse_optimizer", True):
+        from .passes.fuse_optimizer import fuse_optimizer
+        out_node_f = next(n for n in gm.graph.nodes if n.op == "output")
+        flat_outs_f, spec_f = pytree.tree_flatten(out_node_f.args[0])
+        ppos = _adam_positions(params, buffers, named_states)
+        if ppos:
+            nfused = fuse_optimizer(gm, flat_outs_f,
+                                    [n for n in gm.graph.nodes
+                                     if n.op == "placeholder"], ppos, opt)
+            if nfused:
+                out_node_f.args = (pytree.tree_unflatten(flat_outs_f,
+                                                         spec_f),)
+                gm.graph.eliminate_dead_code()
+                gm.graph.lint()
+                gm.recompile()
+
+    from ..utils.dumps import dump_graph
+    dump_graph(gm, "auto_sharded")
+
+    # strip the pytree codegen: the runtime calls the graph with the flat
+    # placeholder list and receives the flat output list
+    import torch.fx as _fx
+    gm.graph._codegen = _fx.graph.CodeGen()
+    gm.recompile()
+
+    # ---- 6. runtime ------------------------------------------------------
+    input_placements = []
+    for i, ph in enumerate(placeholders):
+        pls = out_pl_env.get(ph.name)
+        input_placements.append(pls[0] if pls else None)
+    # pad for non-placeholder flat inputs (shouldn't happen, but be safe)
+    while len(input_placements) < len(flat_inputs):
+        input_placements.append(None)
+
+    # flat-position io map: placeholder idx -> output idx
+    name_to_out_pos = {}
+    for k, o in enumerate(flat_outs):
+        if hasattr(o, "name"):
+            name_to_out_pos.setdefault(o.name, k)
+    io_pos_map = {}
+    ph_by_name = {ph.name: i for i, ph in enumerate(placeholders)}
+    for ph_name, src_name in io_map.items():
+        if ph_name in ph_by_name and src_name in name_to_out_pos:
+            io_pos_map[ph_by_name[ph_name]] = name_to_out_pos[src_name]
+
+    output_placements = []
+    for o in flat_outs:
+        if hasattr(o, "name") and o.name in out_pl_env:
+            output_placements.append(out_pl_env[o.name][0])
+        else:
+            output_placements.append(None)
+
+    compiled = EDCompiledFunc(
+        gm, in_spec, out_spec_graph, input_placements, output_placements,
+        state_positions, io_pos_map, len(params), list(params.keys()), device)
+    compiled.init_named_states = named_states
+    compiled.ret_out_positions = list(range(n_state_outs + n_grads,
+                                            len(flat_outs)))
+    compiled.meta = {
+        "search_time": search_time, "solve_time": solve_time,
+        "n_nodes": len(gm.graph.nodes), "out_spec": gm._out_spec
+        if hasattr(gm, "_out_spec") else None,
+    }
+    return compiled
+
+
+def shard_graph(gm, mesh, io_map, ret_names, device):
+    """Discovery + per-mesh-dim MILP solve + sharding transform.
+
+    Shards along the mesh's spmd dims only (a 'pp' dim is skipped), so
+    the hybrid pipeline path can run this on a marker-carrying graph
+    before splitting stages. Returns (gm, out_pl_env, search_s, solve_s).
+    """
     cache_path = _strategy_cache_path(gm, mesh)
     cached = _load_strategy_cache(cache_path)
     if cached is not None:
@@ -122,82 +205,9 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     dump_strategies(strategies_per_dim, "auto")
     dump_graph(gm, "auto_pre_shard")
 
-    # ---- 5. sharding transform ------------------------------------------
     gm, out_pl_env = sharding_transform(gm, strategies_per_dim, mesh.shape)
     gm = _fix_output_reshard(gm, out_pl_env, io_map, ret_names, mesh)
-
-    # ---- 5b. comm overlap: widen every start/wait window -----------------
-    from .passes.comm_optimize import comm_optimize
-    comm_optimize(gm)
-
-    # ---- 5b2. lower hot aten ops to the gfx950 kernels -------------------
-    from .passes.lower_hip import lower_layer_norm
-    lower_layer_norm(gm)
-
-    # ---- 5c. re-fuse the decomposed Adam chains into ONE kernel ----------
-    if opt is not None and getattr(mdconfig, "fuse_optimizer", True):
-        from .passes.fuse_optimizer import fuse_optimizer
-        out_node_f = next(n for n in gm.graph.nodes if n.op == "output")
-        flat_outs_f, spec_f = pytree.tree_flatten(out_node_f.args[0])
-        ppos = _adam_positions(params, buffers, named_states)
-        if ppos:
-            nfused = fuse_optimizer(gm, flat_outs_f,
-                                    [n for n in gm.graph.nodes
-                                     if n.op == "placeholder"], ppos, opt)
-            if nfused:
-                out_node_f.args = (pytree.tree_unflatten(flat_outs_f,
-                                                         spec_f),)
-                gm.graph.eliminate_dead_code()
-                gm.graph.lint()
-                gm.recompile()
-
-    dump_graph(gm, "auto_sharded")
-
-    # strip the pytree codegen: the runtime calls the graph with the flat
-    # placeholder list and receives the flat output list
-    import torch.fx as _fx
-    gm.graph._codegen = _fx.graph.CodeGen()
-    gm.recompile()
-
-    # ---- 6. runtime ------------------------------------------------------
-    input_placements = []
-    for i, ph in enumerate(placeholders):
-        pls = out_pl_env.get(ph.name)
-        input_placements.append(pls[0] if pls else None)
-    # pad for non-placeholder flat inputs (shouldn't happen, but be safe)
-    while len(input_placements) < len(flat_inputs):
-        input_placements.append(None)
-
-    # flat-position io map: placeholder idx -> output idx
-    name_to_out_pos = {}
-    for k, o in enumerate(flat_outs):
-        if hasattr(o, "name"):
-            name_to_out_pos.setdefault(o.name, k)
-    io_pos_map = {}
-    ph_by_name = {ph.name: i for i, ph in enumerate(placeholders)}
-    for ph_name, src_name in io_map.items():
-        if ph_name in ph_by_name and src_name in name_to_out_pos:
-            io_pos_map[ph_by_name[ph_name]] = name_to_out_pos[src_name]
-
-    output_placements = []
-    for o in flat_outs:
-        if hasattr(o, "name") and o.name in out_pl_env:
-            output_placements.append(out_pl_env[o.name][0])
-        else:
-            output_placements.append(None)
-
-    compiled = EDCompiledFunc(
-        gm, in_spec, out_spec_graph, input_placements, output_placements,
-        state_positions, io_pos_map, len(params), list(params.keys()), device)
-    compiled.init_named_states = named_states
-    compiled.ret_out_positions = list(range(n_state_outs + n_grads,
-                                            len(flat_outs)))
-    compiled.meta = {
-        "search_time": search_time, "solve_time": solve_time,
-        "n_nodes": len(gm.graph.nodes), "out_spec": gm._out_spec
-        if hasattr(gm, "_out_spec") else None,
-    }
-    return compiled
+    return gm, out_pl_env, search_time, solve_time
 
 
 def _strategy_cache_path(gm, mesh) -> Optional[str]:
