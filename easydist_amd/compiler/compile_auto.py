@@ -124,6 +124,15 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     for ph_name, src_name in io_map.items():
         if ph_name in ph_by_name and src_name in name_to_out_pos:
             io_pos_map[ph_by_name[ph_name]] = name_to_out_pos[src_name]
+    # positional fallback: non-fused optimizers trace as in-place ops
+    # (empty io_map after fix_inplace); output i of the state block IS the
+    # new value of placeholder i
+    for i in state_positions:
+        if i not in io_pos_map and i < len(flat_outs):
+            o = flat_outs[i]
+            if hasattr(o, "name") and (i >= len(placeholders)
+                                       or o.name != placeholders[i].name):
+                io_pos_map[i] = i
 
     output_placements = []
     for o in flat_outs:
@@ -146,7 +155,7 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     return compiled
 
 
-def shard_graph(gm, mesh, io_map, ret_names, device):
+def shard_graph(gm, mesh, io_map, ret_names, device, fix_rets=True):
     """Discovery + per-mesh-dim MILP solve + sharding transform.
 
     Shards along the mesh's spmd dims only (a 'pp' dim is skipped), so
@@ -206,7 +215,8 @@ def shard_graph(gm, mesh, io_map, ret_names, device):
     dump_graph(gm, "auto_pre_shard")
 
     gm, out_pl_env = sharding_transform(gm, strategies_per_dim, mesh.shape)
-    gm = _fix_output_reshard(gm, out_pl_env, io_map, ret_names, mesh)
+    gm = _fix_output_reshard(gm, out_pl_env, io_map,
+                             ret_names if fix_rets else set(), mesh)
     return gm, out_pl_env, search_time, solve_time
 
 

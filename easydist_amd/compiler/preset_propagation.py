@@ -163,3 +163,31 @@ def _embedding_bwd_rule(input_shapes, args, kwargs):
                                        ops=__import__("operator").add)
         sid += 1
     return ann, combs
+
+
+# ------------------------------------------------------- pipeline markers ----
+def _register_pp_markers():
+    """pp_split/step_split are identities: shard through any dim.
+
+    Registered lazily because the marker ops live in parallel.pp.split
+    (imported on first pipeline compile)."""
+    try:
+        import easydist_amd.parallel.pp.split  # noqa: F401
+        ops = [torch.ops.easydist_amd.pp_split.default,
+               torch.ops.easydist_amd.step_split.default]
+    except (ImportError, AttributeError):
+        return
+
+    @register_preset(*ops)
+    def _marker_rule(input_shapes, args, kwargs):
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        combs = {}
+        sid = 1
+        for d in range(len(input_shapes[0])):
+            ann[0][d] = ShardDim.get_shard_dim(sid)
+            combs[sid] = _gather(d)
+            sid += 1
+        return ann, combs
+
+
+_register_pp_markers()

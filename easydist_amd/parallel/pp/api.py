@@ -60,15 +60,55 @@ def _compile_pp(func, tracing_mode, args, kwargs, module, opt,
     n_state = (len(params) + len(buffers)
                + len(pytree.tree_flatten(named_states)[0]))
 
+    # ---- hybrid pp x spmd: auto-SPMD-shard the marker-carrying graph
+    # along the spmd mesh dims BEFORE stage splitting
+    # (reference: compile_auto.py:683-715 + test_hybrid.py meshes)
+    from ...parallel.device_mesh import get_device_mesh
+    mesh = get_device_mesh()
+    hybrid = (mesh is not None and mesh.has_dim("pp")
+              and mesh.size() > mesh.size(mesh.dim_index("pp")))
+    io_map = None
+    out_pl_env = {}
+    if hybrid:
+        from ...compiler.compile_auto import shard_graph
+        from ...compiler.passes.functionalize import canonicalize
+        gm, io_map = canonicalize(gm)
+        gm, out_pl_env, _st, _sv = shard_graph(gm, mesh, io_map, set(),
+                                               device, fix_rets=False)
+
     info = compile_pipeline(gm, flat_inputs, n_params, n_state,
-                            list(params.keys()))
+                            list(params.keys()), io_map=io_map)
     assert info.nstages == nstages, \
         f"trace produced {info.nstages} stages, expected {nstages}"
+
+    if hybrid:
+        info.pp_mesh_dim = mesh.dim_index("pp")
+        for name in info.ph_names:
+            pls = out_pl_env.get(name)
+            if pls:
+                info.ph_placements[name] = pls[0]
+        # ret + boundary placements for the runtime's spmd fixes
+        for sg in info.stages:
+            for name in sg.ret_names:
+                pls = out_pl_env.get(name)
+                if pls:
+                    info.ret_placements[name] = pls[0]
+            for name in sg.fw_recv + sg.bw_recv:
+                pls = out_pl_env.get(name)
+                if pls:
+                    info.boundary_placements[name] = pls[0]
 
     ph_values = dict(zip(info.ph_names, flat_inputs))
 
     world = dist.get_world_size() if dist.is_initialized() else 1
-    if world > 1:
+    if hybrid and world > 1:
+        assert world == mesh.size(), \
+            f"world {world} != mesh size {mesh.size()}"
+        assert mesh.size(info.pp_mesh_dim) == nstages
+        stage_idx = mesh.my_coords()[info.pp_mesh_dim]
+        rt = PipelineStage(info, stage_idx, device, nchunks,
+                           schedule=schedule, scale_grads=scale_grads)
+    elif world > 1:
         assert world == nstages, \
             f"world size {world} must equal nstages {nstages}"
         rt = PipelineStage(info, dist.get_rank(), device, nchunks,

@@ -49,6 +49,9 @@ class StageGraphs:
     bw_outputs: List[str]
     step_gm: Optional[fx.GraphModule]
     step_inputs: List[str]
+    # functional (hybrid) mode: step outputs + state writeback mapping
+    step_outputs: List[str] = field(default_factory=list)
+    writeback: Dict[str, str] = field(default_factory=dict)  # ph -> src
     # boundary activations: recv from prev stage / send to next
     fw_recv: List[str] = field(default_factory=list)
     fw_send: List[str] = field(default_factory=list)
@@ -80,6 +83,11 @@ class PipelineInfo:
     n_params: int
     param_names: List[str]       # torch param qualnames, traced order
     meta_vals: Dict[str, object]  # node name -> fake val (shape/dtype)
+    # hybrid pp x spmd: per-placeholder / per-ret spmd placement vectors
+    ph_placements: Dict[str, list] = field(default_factory=dict)
+    ret_placements: Dict[str, list] = field(default_factory=dict)
+    boundary_placements: Dict[str, list] = field(default_factory=dict)
+    pp_mesh_dim: Optional[int] = None
 
 
 def _val(node):
@@ -233,7 +241,13 @@ def _extract(gm: fx.GraphModule, seg: List[fx.Node], alias: Dict,
 
 
 def compile_pipeline(gm: fx.GraphModule, flat_inputs, n_params: int,
-                     n_state: int, param_names: List[str]) -> PipelineInfo:
+                     n_state: int, param_names: List[str],
+                     io_map: Optional[Dict[str, str]] = None
+                     ) -> PipelineInfo:
+    """io_map (functional/hybrid mode): placeholder name -> producing node
+    name for state round-trips — the canonicalized graph carries no
+    copy_ mutations, so each stage writes its updated state back from
+    its step outputs."""
     (nstages, fw_segs, bw_segs, step_seg, fw_marks, bw_marks, loss_mark,
      loss_twins, step_marks, placeholders, out_node) = _segment_graph(gm)
 
@@ -400,13 +414,35 @@ def compile_pipeline(gm: fx.GraphModule, flat_inputs, n_params: int,
         # stage-s nodes; ∅-stage nodes are dead (no path to any parameter)
         my_step = [n for n in step_seg if s in step_node_stages[n]]
         step_gm, step_in = None, []
+        step_out_names: List[str] = []
+        writeback: Dict[str, str] = {}
         if my_step:
-            step_gm, step_in, _ = _extract(gm, my_step, alias, [])
+            wanted_step: List[fx.Node] = []
+            if io_map is not None:
+                # functional mode: the state round trip is positional —
+                # output i of (params, buffers, states, ...) is the new
+                # value of placeholder i (io_map only covers copy_-style
+                # traces; in-place-op traces have an empty io_map)
+                my_names = {n.name for n in my_step}
+                for i in range(min(n_state, len(flat_outs),
+                                   len(placeholders))):
+                    o = flat_outs[i]
+                    if isinstance(o, fx.Node) and o.op != "placeholder" \
+                            and o.name in my_names:
+                        writeback[placeholders[i].name] = o.name
+                for ph_name, src_name in io_map.items():
+                    if src_name in my_names:
+                        writeback[ph_name] = src_name
+                wanted_step = [n for n in my_step
+                               if n.name in set(writeback.values())]
+            step_gm, step_in, step_out_names = _extract(gm, my_step, alias,
+                                                        wanted_step)
 
         sg = StageGraphs(
             stage_idx=s, fw_gm=fw_gm, fw_inputs=fw_in,
             fw_outputs=fw_out_names, bw_gm=bw_gm, bw_inputs=bw_in,
-            bw_outputs=bw_out_names, step_gm=step_gm, step_inputs=step_in)
+            bw_outputs=bw_out_names, step_gm=step_gm, step_inputs=step_in,
+            step_outputs=step_out_names, writeback=writeback)
 
         # classify interfaces by name
         ph_by_name = {p.name: i for i, p in enumerate(placeholders)}

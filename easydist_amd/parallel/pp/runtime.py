@@ -61,11 +61,17 @@ class StageExecutor:
         self.rets: List[Dict[str, torch.Tensor]] = []
 
     def init_state(self, ph_values: Dict[str, torch.Tensor]):
+        from ...parallel.device_mesh import get_device_mesh
+        from ...runtime.compiled_func import shard_tensor_local
+        mesh = get_device_mesh()
         for name in self.sg.state_names:
-            t = ph_values[name]
+            t = ph_values[name].detach().to(self.device)
+            pl = self.info.ph_placements.get(name)
+            if pl is not None and mesh is not None:
+                t = shard_tensor_local(t, pl, mesh)
             # clone: the runtime owns its state (mutated in place by
             # step_gm); never alias the user's module tensors
-            self.state_env[name] = t.detach().clone().to(self.device)
+            self.state_env[name] = t.clone()
 
     def reset_step(self, nchunks: int):
         self.stash = [dict() for _ in range(nchunks)]
@@ -111,7 +117,12 @@ class StageExecutor:
         for name, g in self.grad_acc.items():
             env[name] = g.div_(nchunks) if scale_grads else g
         args = [env[n] for n in self.sg.step_inputs]
-        self.sg.step_gm(*args)   # mutates state in place (copy_ nodes)
+        outs = self.sg.step_gm(*args)  # in-place (copy_) OR functional
+        if self.sg.writeback:
+            by_name = dict(zip(self.sg.step_outputs, outs))
+            for ph, src in self.sg.writeback.items():
+                if src in by_name and ph in self.state_env:
+                    self.state_env[ph] = by_name[src].detach()
 
     def fw_boundary_out(self, m: int):
         return [self.stash[m][n] for n in self.sg.fw_send]
@@ -142,12 +153,20 @@ class LocalPipelineRuntime:
             ex.init_state(ph_values)
 
     def _data_chunks(self, args, kwargs):
+        from ...parallel.device_mesh import get_device_mesh
+        from ...runtime.compiled_func import shard_tensor_local
+        mesh = get_device_mesh()
         da_flat, _ = pytree.tree_flatten((args, kwargs))
         chunks: Dict[str, List] = {}
         for j, v in enumerate(da_flat):
             name = self.info.ph_names[self.info.n_state + j]
             if isinstance(v, torch.Tensor) and v.ndim >= 1:
-                chunks[name] = _chunk(v.to(self.device), self.nchunks)
+                pieces = _chunk(v.to(self.device), self.nchunks)
+                pl = self.info.ph_placements.get(name)
+                if pl is not None and mesh is not None:
+                    pieces = [shard_tensor_local(c, pl, mesh)
+                              for c in pieces]
+                chunks[name] = pieces
             else:
                 chunks[name] = [v] * self.nchunks
         return chunks
@@ -207,7 +226,8 @@ class LocalPipelineRuntime:
 
 # --------------------------------------------------------- distributed mode --
 class PipelineStage:
-    """One rank = one stage. reference: runtime.py:113-567."""
+    """One rank = one stage (hybrid: one (pp, spmd...) mesh coordinate).
+    reference: runtime.py:113-567."""
 
     def __init__(self, info: PipelineInfo, stage_idx: int, device: str,
                  nchunks: int, schedule: str = "gpipe",
@@ -219,6 +239,9 @@ class PipelineStage:
         self.schedule = schedule
         self.scale_grads = scale_grads
         self.group = group
+        if info.pp_mesh_dim is not None and group is None:
+            from ...parallel.device_mesh import get_device_mesh
+            self.group = get_device_mesh().get_group(info.pp_mesh_dim)
         self.ex = StageExecutor(info.stages[stage_idx], info, device)
         self.meta: Dict = {"search_time": 0, "solve_time": 0}
         self._send_reqs: List = []
@@ -228,13 +251,31 @@ class PipelineStage:
 
     # ---------------------------------------------------------- p2p ---------
     def _peer(self, delta: int) -> int:
-        return dist.get_rank() + delta
+        if self.info.pp_mesh_dim is None:
+            return dist.get_rank() + delta
+        from ...parallel.device_mesh import get_device_mesh
+        mesh = get_device_mesh()
+        coords = mesh.my_coords()
+        coords[self.info.pp_mesh_dim] += delta
+        return int(mesh.mesh.mesh[tuple(coords)])
+
+    def _local_shape(self, name):
+        val = self.info.meta_vals[name]
+        shape = list(val.shape)
+        pl = self.info.boundary_placements.get(name)
+        if pl:
+            from ...parallel.device_mesh import get_device_mesh
+            mesh = get_device_mesh()
+            for d, p in enumerate(pl):
+                if p.is_shard():
+                    shape[p.dim] //= mesh.shape[d]
+        return shape
 
     def _recv(self, names: List[str], m: int):
         tensors = []
         for name in names:
             val = self.info.meta_vals[name]
-            t = torch.empty(tuple(val.shape), dtype=val.dtype,
+            t = torch.empty(tuple(self._local_shape(name)), dtype=val.dtype,
                             device=self.device)
             tensors.append(t)
         if tensors:
@@ -316,6 +357,15 @@ class PipelineStage:
     def _exchange_returns(self, rets):
         """All ranks end up with every user return.
         reference: runtime.py:487-507 (all_gather_object)."""
+        # hybrid: resolve spmd placement of each return first (PARTIAL
+        # loss -> all_reduce over the spmd dims, SHARD -> all_gather)
+        if self.info.ret_placements:
+            from ...parallel.device_mesh import get_device_mesh
+            from ...runtime.compiled_func import unshard_tensor
+            mesh = get_device_mesh()
+            rets = {k: (unshard_tensor(v, self.info.ret_placements[k], mesh)
+                        if k in self.info.ret_placements else v)
+                    for k, v in rets.items()}
         cpu_rets = {k: v.detach().cpu() for k, v in rets.items()}
         gathered: List = [None] * dist.get_world_size(self.group)
         dist.all_gather_object(gathered, cpu_rets, group=self.group)
