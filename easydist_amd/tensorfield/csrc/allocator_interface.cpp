@@ -4,6 +4,10 @@
 //
 // Capability parity with reference ``easydist/torch/tensorfield/csrc/
 // allocator_interface.cpp`` (my_malloc/my_free 69-137, socket protocol
+// NOTE: this hipIpc-based client needs a driver that allows
+// hipIpcOpenMemHandle; the current pool driver is dmabuf-only (open
+// returns hipErrorInvalidValue), so the python client path (torch CUDA
+// IPC reductions) is the production transport there.
 // 13-68) re-written for HIP: same text protocol as server.py, dmabuf
 // IPC (HSA_ENABLE_IPC_MODE_LEGACY=0).
 #include <hip/hip_runtime.h>

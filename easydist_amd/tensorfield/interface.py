@@ -1,11 +1,10 @@
 """TFieldClient: python client of the tfield memory-pool server.
 
 Capability parity with reference ``easydist/torch/tensorfield/
-interface.py`` (TFieldClient:18, param-group APIs 56-120). Maps served
-buffers with hipIpcOpenMemHandle via ctypes and moves data with
-hipMemcpy, so tests can verify true cross-process sharing without the
-C allocator in the loop (that client lives in csrc/
-allocator_interface.cpp).
+interface.py`` (TFieldClient:18, param-group APIs 56-120). Buffers are
+mapped via torch's CUDA IPC reductions (the dmabuf path this host
+driver supports); a "handle" on the wire is 8 key bytes + the pickled
+``reduce_tensor`` payload.
 """
 from __future__ import annotations
 
@@ -57,7 +56,6 @@ class TFieldClient:
         self.sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
         self.sock.connect(path)
         self.f = self.sock.makefile("rw")
-        self.hip = _HipClient()
         import os
         self._cmd(f"hello {os.getpid()}")
 
@@ -99,21 +97,19 @@ class TFieldClient:
         self._cmd("quit")
 
     # ----------------------------------------------------- data helpers ----
+    def map_tensor(self, handle: bytes):
+        """Rebuild the shared CUDA tensor in THIS process."""
+        import pickle
+        fn, args = pickle.loads(handle[8:])
+        return fn(*args)
+
     def write_bytes(self, handle: bytes, offset: int, data: bytes):
-        ptr = self.hip.open_handle(handle)
-        try:
-            buf = ctypes.create_string_buffer(data, len(data))
-            self.hip.memcpy(ptr + offset, ctypes.addressof(buf), len(data),
-                            hipMemcpyHostToDevice)
-        finally:
-            self.hip.close_handle(ptr)
+        import torch
+        t = self.map_tensor(handle)
+        src = torch.frombuffer(bytearray(data), dtype=torch.uint8)
+        t[offset:offset + len(data)].copy_(src.to(t.device))
+        torch.cuda.synchronize()
 
     def read_bytes(self, handle: bytes, offset: int, size: int) -> bytes:
-        ptr = self.hip.open_handle(handle)
-        try:
-            buf = ctypes.create_string_buffer(size)
-            self.hip.memcpy(ctypes.addressof(buf), ptr + offset, size,
-                            hipMemcpyDeviceToHost)
-            return buf.raw
-        finally:
-            self.hip.close_handle(ptr)
+        t = self.map_tensor(handle)
+        return bytes(t[offset:offset + size].cpu().numpy().tobytes())

@@ -1,11 +1,12 @@
-"""tfield-server: cross-process GPU memory pool over hipIpc.
+"""tfield-server: cross-process GPU memory pool.
 
 Capability parity with reference ``easydist/torch/tensorfield/``
 (server.py:55-160 IPCMemoryPool, mem_pool.py, param-group registry
-server.py:16-48). MI355X re-design: no cupy — the pool talks to the HIP
-runtime directly through ctypes (hipMalloc / hipIpcGetMemHandle), and
-clients map buffers with hipIpcOpenMemHandle (dmabuf IPC mode,
-HSA_ENABLE_IPC_MODE_LEGACY=0).
+server.py:16-48). MI355X re-design: buffers are exported through
+torch's CUDA IPC reductions — the dmabuf transport this host driver
+supports (raw hipIpcOpenMemHandle returns hipErrorInvalidValue here;
+see csrc/allocator_interface.cpp for the hipIpc-based C client kept for
+drivers that allow it).
 
 Run: ``python -m easydist_amd.tensorfield.server [--socket PATH]``.
 
@@ -34,63 +35,45 @@ HIP_IPC_HANDLE_SIZE = 64
 DEFAULT_SOCKET = "/tmp/easydist_tfield.sock"
 
 
-class Hip:
-    def __init__(self):
-        self.lib = ctypes.CDLL("libamdhip64.so")
-        self.lib.hipMalloc.argtypes = [ctypes.POINTER(ctypes.c_void_p),
-                                       ctypes.c_size_t]
-        self.lib.hipFree.argtypes = [ctypes.c_void_p]
-        self.lib.hipIpcGetMemHandle.argtypes = [ctypes.c_char_p,
-                                                ctypes.c_void_p]
-        self.lib.hipSetDevice.argtypes = [ctypes.c_int]
-
-    def malloc(self, size: int) -> int:
-        p = ctypes.c_void_p()
-        rc = self.lib.hipMalloc(ctypes.byref(p), size)
-        if rc != 0:
-            raise MemoryError(f"hipMalloc({size}) rc={rc}")
-        return p.value
-
-    def free(self, ptr: int):
-        self.lib.hipFree(ctypes.c_void_p(ptr))
-
-    def ipc_handle(self, ptr: int) -> bytes:
-        buf = ctypes.create_string_buffer(HIP_IPC_HANDLE_SIZE)
-        rc = self.lib.hipIpcGetMemHandle(buf, ctypes.c_void_p(ptr))
-        if rc != 0:
-            raise RuntimeError(f"hipIpcGetMemHandle rc={rc}")
-        return buf.raw
-
-
 class IPCMemoryPool:
-    """One hipMalloc per allocation, exported as an IPC handle.
+    """torch-allocated slabs exported via torch's CUDA IPC reductions.
 
-    (The reference leaned on cupy's pool; allocation rate here is
-    model-startup-scale, so direct hipMalloc is the simpler honest
-    design. A slab layer can be added behind the same protocol.)"""
+    The raw hipIpcGetMemHandle/hipIpcOpenMemHandle pair is unavailable on
+    this host driver (dmabuf-only IPC: get succeeds, open returns
+    hipErrorInvalidValue); torch's multiprocessing reductions ride the
+    supported dmabuf path, so the pool allocates torch CUDA tensors and
+    ships their pickled `reduce_tensor` payloads as the "handle"."""
 
     def __init__(self, device: int = 0):
-        self.hip = Hip()
-        self.hip.lib.hipSetDevice(device)
-        self.allocs: Dict[bytes, Tuple[int, int]] = {}   # handle -> ptr,size
+        import torch
+        self.torch = torch
+        torch.cuda.set_device(device)
+        torch.cuda.init()
+        self.allocs: Dict[bytes, Tuple[object, int]] = {}  # key->tensor,size
         self.lock = threading.Lock()
+        self._next = 0
 
     def alloc(self, size: int) -> Tuple[bytes, int, int]:
-        ptr = self.hip.malloc(size)
-        h = self.hip.ipc_handle(ptr)
+        import pickle
+
+        from torch.multiprocessing.reductions import reduce_tensor
+        t = self.torch.empty(size, dtype=self.torch.uint8, device="cuda")
+        payload = pickle.dumps(reduce_tensor(t))
         with self.lock:
-            self.allocs[h] = (ptr, size)
-        return h, 0, size
+            key = self._next.to_bytes(8, "little")
+            self._next += 1
+            self.allocs[key] = (t, size, payload)
+        return key + payload, 0, size
 
     def free(self, handle: bytes):
+        key = handle[:8]
         with self.lock:
-            ent = self.allocs.pop(handle, None)
-        if ent:
-            self.hip.free(ent[0])
+            self.allocs.pop(key, None)
 
     def stat(self):
         with self.lock:
-            return len(self.allocs), sum(s for _, s in self.allocs.values())
+            return len(self.allocs), sum(e[1]
+                                         for e in self.allocs.values())
 
 
 class ParamGroupStore:
