@@ -98,6 +98,13 @@ def _compile_pp(func, tracing_mode, args, kwargs, module, opt,
                 if pls:
                     info.boundary_placements[name] = pls[0]
 
+    qualnames = (list(params.keys()) + list(buffers.keys())
+                 + [f"{pn}.{k}" for pn, st in named_states.items()
+                    for k, v in st.items()
+                    if isinstance(v, __import__("torch").Tensor)])
+    for ph, qn in zip(info.ph_names, qualnames):
+        info.ph_qualnames[ph] = qn
+
     ph_values = dict(zip(info.ph_names, flat_inputs))
 
     world = dist.get_world_size() if dist.is_initialized() else 1

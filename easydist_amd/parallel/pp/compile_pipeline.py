@@ -83,6 +83,9 @@ class PipelineInfo:
     n_params: int
     param_names: List[str]       # torch param qualnames, traced order
     meta_vals: Dict[str, object]  # node name -> fake val (shape/dtype)
+    # placeholder name -> user-facing qualified name (params: module
+    # qualname; buffers likewise; optimizer states: "<param>.<state_key>")
+    ph_qualnames: Dict[str, str] = field(default_factory=dict)
     # hybrid pp x spmd: per-placeholder / per-ret spmd placement vectors
     ph_placements: Dict[str, list] = field(default_factory=dict)
     ret_placements: Dict[str, list] = field(default_factory=dict)

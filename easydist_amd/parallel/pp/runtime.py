@@ -223,6 +223,24 @@ class LocalPipelineRuntime:
             out.update(ex.state_env)
         return out
 
+    def state_dict(self):
+        """Qualified-name state (params+buffers+opt states), full values.
+        reference: pp/runtime.py:509-544."""
+        out = {}
+        for ex in self.execs:
+            for ph, t in ex.state_env.items():
+                qn = self.info.ph_qualnames.get(ph, ph)
+                out[qn] = t.detach().clone()
+        return out
+
+    def load_state_dict(self, sd):
+        for ex in self.execs:
+            for ph in list(ex.state_env):
+                qn = self.info.ph_qualnames.get(ph, ph)
+                if qn in sd:
+                    ex.state_env[ph] = sd[qn].detach().clone().to(
+                        self.device)
+
 
 # --------------------------------------------------------- distributed mode --
 class PipelineStage:
@@ -393,3 +411,39 @@ class PipelineStage:
 
     def get_state(self):
         return dict(self.ex.state_env)
+
+    def state_dict(self):
+        """Gather the FULL training state to every rank (cross-stage
+        all_gather_object + spmd unshard). reference: runtime.py:509-544,
+        compile_pipeline.py:484-583."""
+        from ...parallel.device_mesh import get_device_mesh
+        from ...runtime.compiled_func import unshard_tensor
+        mesh = get_device_mesh()
+        local = {}
+        for ph, t in self.ex.state_env.items():
+            qn = self.info.ph_qualnames.get(ph, ph)
+            pl = self.info.ph_placements.get(ph)
+            if pl is not None and mesh is not None:
+                t = unshard_tensor(t, pl, mesh)
+            local[qn] = t.detach().cpu()
+        gathered = [None] * dist.get_world_size(self.group)
+        dist.all_gather_object(gathered, local, group=self.group)
+        out = {}
+        for d in gathered:
+            out.update(d)
+        return {k: v.to(self.device) for k, v in out.items()}
+
+    def load_state_dict(self, sd):
+        """Load full values; each stage re-shards its own slice."""
+        from ...parallel.device_mesh import get_device_mesh
+        from ...runtime.compiled_func import shard_tensor_local
+        mesh = get_device_mesh()
+        for ph in list(self.ex.state_env):
+            qn = self.info.ph_qualnames.get(ph, ph)
+            if qn not in sd:
+                continue
+            t = sd[qn].detach().to(self.device)
+            pl = self.info.ph_placements.get(ph)
+            if pl is not None and mesh is not None:
+                t = shard_tensor_local(t, pl, mesh)
+            self.ex.state_env[ph] = t.clone()

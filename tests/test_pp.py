@@ -129,3 +129,27 @@ def test_pp_gpt_local():
 @pytest.mark.world2
 def test_pp_gpt_ws2_dapple():
     spawn(_gpt_body, args=(2, "dapple"), world_size=2, port=29545)
+
+
+def test_pp_state_dict_roundtrip():
+    """Checkpoint parity: save at step 2, train to 4, load, rerun — losses
+    must repeat exactly (reference: pp/runtime.py:509-544)."""
+    from easydist_amd import easydist_compile
+
+    init_single_process()
+    torch.manual_seed(42)
+    model = MLP4()
+    opt = torch.optim.Adam(model.parameters(), lr=1e-2)
+    compiled = easydist_compile(train_step, parallel_mode="pp",
+                                cuda_graph=False, split_points={"fc2"},
+                                nchunks=2)
+    torch.manual_seed(7)
+    batches = [(torch.randn(8, 16), torch.randn(8, 16)) for _ in range(4)]
+    for x, y in batches[:2]:
+        compiled(model, opt, x, y)
+    rt = list(compiled.compiled.values())[0]
+    ckpt = rt.state_dict()
+    later = [float(compiled(model, opt, x, y)) for x, y in batches[2:]]
+    rt.load_state_dict(ckpt)
+    replay = [float(compiled(model, opt, x, y)) for x, y in batches[2:]]
+    assert later == replay, (later, replay)
