@@ -230,6 +230,9 @@ def _strategy_cache_path(gm, mesh) -> Optional[str]:
     h = hashlib.sha256()
     h.update(str(gm.graph).encode())
     h.update(repr(mesh.shape).encode())
+    import easydist_amd
+    h.update(easydist_amd.__version__.encode())
+    h.update(torch.__version__.encode())
     d = os.path.join(os.path.expanduser("~"), ".easydist_amd",
                      "compile_cache")
     os.makedirs(d, exist_ok=True)

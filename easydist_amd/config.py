@@ -39,7 +39,7 @@ easydist_device = os.environ.get("EASYDIST_DEVICE", "cuda")
 forced_compile = _env_flag("EASYDIST_FORCED_COMPILE")
 
 # ------------------------------------------------------------- compile cache --
-enable_compile_cache = _env_flag("EASYDIST_COMPILE_CACHE")
+enable_compile_cache = _env_flag("EASYDIST_COMPILE_CACHE", True)
 compile_cache_dir = os.environ.get("EASYDIST_COMPILE_CACHE_DIR", "./md_compiled")
 
 # --------------------------------------------------------- rule discovery ----

@@ -124,22 +124,46 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
       chunks.push_back(c);
     }
   }
-  // upload the table; when a hipGraph capture is active the pinned host
-  // buffer must outlive the graph (the captured H2D reads it on replay)
-  auto host = at::empty({(long)(chunks.size() * sizeof(AdamChunk))},
-                        at::TensorOptions().dtype(at::kByte).pinned_memory(true));
-  memcpy(host.data_ptr(), chunks.data(), chunks.size() * sizeof(AdamChunk));
-  at::Tensor dev_chunks = host.to(params[0].device(), /*non_blocking=*/true);
-  if (at::cuda::currentStreamCaptureStatusMayInitCtx() !=
-      at::cuda::CaptureStatus::None) {
-    static std::vector<at::Tensor> capture_keepalive;
-    capture_keepalive.push_back(host);
+  // capture-safe table upload: PERSISTENT pinned staging + device table,
+  // (re)allocated only OUTSIDE capture (the eager warmup sizes them);
+  // inside capture we only host-write the pinned buffer (recorded values
+  // stay valid: the graph pool re-serves the same output pointers on
+  // every replay) and record one async H2D memcpy node.
+  static at::Tensor g_pinned, g_dev;
+  static size_t g_capacity = 0;
+  const size_t bytes = chunks.size() * sizeof(AdamChunk);
+  bool capturing = at::cuda::currentStreamCaptureStatusMayInitCtx() !=
+                   at::cuda::CaptureStatus::None;
+  if (bytes > g_capacity) {
+    TORCH_CHECK(!capturing,
+                "fused_adam_step: staging buffers must be warmed up (one "
+                "eager call) before hipGraph capture");
+    g_pinned = at::empty({(long)bytes},
+                         at::TensorOptions().dtype(at::kByte)
+                             .pinned_memory(true));
+    g_dev = at::empty({(long)bytes},
+                      at::TensorOptions().dtype(at::kByte)
+                          .device(params[0].device()));
+    g_capacity = bytes;
   }
+  memcpy(g_pinned.data_ptr(), chunks.data(), bytes);
   auto stream = at::cuda::getCurrentCUDAStream();
+  C10_CUDA_CHECK(hipMemcpyAsync(g_dev.data_ptr(), g_pinned.data_ptr(),
+                                bytes, hipMemcpyHostToDevice, stream));
   hipLaunchKernelGGL(fused_adam_kernel, dim3(chunks.size()), dim3(256), 0,
-      stream, (const AdamChunk*)dev_chunks.data_ptr(), (int)chunks.size(),
+      stream, (const AdamChunk*)g_dev.data_ptr(), (int)chunks.size(),
       steps_flat.data_ptr<float>(), (float)lr, (float)beta1, (float)beta2,
       (float)weight_decay, (float)eps);
+  if (capturing) {
+    // the recorded H2D re-reads THIS pinned buffer on every replay:
+    // retire the pair so no later call can overwrite it
+    static std::vector<at::Tensor> keepalive;
+    keepalive.push_back(g_pinned);
+    keepalive.push_back(g_dev);
+    g_pinned = at::Tensor();
+    g_dev = at::Tensor();
+    g_capacity = 0;
+  }
   // bump steps on host-free path: out_step = step + 1
   for (int t = 0; t < T; ++t)
     out_steps[t] = steps[t] + 1;

@@ -44,12 +44,10 @@ def _adam_aten(params, grads, exp_avgs, exp_avg_sqs, steps, lr, beta1, beta2,
 def _adam_cuda(params, grads, exp_avgs, exp_avg_sqs, steps, lr, beta1, beta2,
                weight_decay, eps):
     ext = load_extension()
-    # under hipGraph capture the chunk-table upload (pinned host alloc +
-    # H2D) would invalidate the capture; the decomposed aten math records
-    # fine, and replay erases its launch overhead anyway
-    capturing = torch.cuda.is_current_stream_capturing()
-    if ext is not None and not capturing \
-            and all(p.dtype == torch.float32 for p in params):
+    # capture-safe: the kernel stages its chunk table through persistent
+    # pinned+device buffers sized by the eager warmup call, so hipGraph
+    # capture records one async H2D + one kernel launch
+    if ext is not None and all(p.dtype == torch.float32 for p in params):
         return ext.fused_adam_step(list(params), list(grads), list(exp_avgs),
                                    list(exp_avg_sqs), list(steps), lr, beta1,
                                    beta2, weight_decay, eps)
