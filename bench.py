@@ -27,7 +27,8 @@ def main():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="gpt2-small",
                    choices=["gpt2-small", "gpt2-medium", "gpt2-1.3b",
-                            "gpt-1l-12288"])
+                            "gpt-1l-12288", "vit-large", "mixtral-4l",
+                            "mixtral-small"])
     p.add_argument("--per-gpu-batch", type=int, default=8)
     p.add_argument("--seq", type=int, default=1024)
     p.add_argument("--parallel", default="auto")
@@ -63,37 +64,78 @@ def main():
     easydist_setup(backend="torch", device="cuda" if use_cuda else "cpu")
     set_device_mesh(list(range(world_size)), ["spmd0"])
 
-    cfg = {
-        "gpt2-small": gptm.GPT2_SMALL,
-        "gpt2-medium": gptm.GPT2_MEDIUM,
-        "gpt2-1.3b": gptm.GPT2_1_3B,
-        "gpt-1l-12288": gptm.GPT_BENCH_1L,
-    }[args.model]
-    if args.seq != cfg.block_size:
-        from dataclasses import replace
-        cfg = replace(cfg, block_size=args.seq)
-
-    torch.manual_seed(1234 + 0)   # same weights everywhere
-    model = gptm.GPT(cfg).to(device)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
-
     global_batch = args.per_gpu_batch * n_gpus
+    torch.manual_seed(1234 + 0)   # same weights everywhere
+    g = torch.Generator(device="cpu").manual_seed(7)     # same on all ranks
 
-    def train_step(model, opt, idx, targets):
-        return gptm.gpt_train_step(model, opt, idx, targets)
+    if args.model == "vit-large":
+        # BASELINE config #4: ViT-Large, mixed DP+TP reshard
+        from easydist_amd.models import vit as vitm
+        model = vitm.ViT(vitm.VIT_LARGE).to(device)
+        opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
 
-    compiled = easydist_compile(train_step, parallel_mode=args.parallel,
-                                cuda_graph=args.hipgraph)
+        def train_step(model, opt, x, y):
+            return vitm.vit_train_step(model, opt, x, y)
+        idx = torch.randn(global_batch, 3, 224, 224, generator=g).to(device)
+        tg = torch.randint(0, 1000, (global_batch,), generator=g).to(device)
+        tokens_per_sample = (224 // 16) ** 2 + 1
+    elif args.model.startswith("mixtral"):
+        # BASELINE config #5: Mixtral-shape MoE, expert all-to-all; module
+        # EP over the world group when world>1
+        from easydist_amd.models import moe as moem
+        cfg = (moem.MIXTRAL_BENCH_4L if args.model == "mixtral-4l"
+               else moem.MIXTRAL_SMALL)
+        if args.seq != cfg.block_size:
+            from dataclasses import replace
+            cfg = replace(cfg, block_size=args.seq)
+        ep_group = dist.group.WORLD if world_size > 1 else None
+        model = moem.MoEGPT(cfg, ep_group=ep_group).to(device)
+        opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
 
-    def batch():
-        g = torch.Generator(device="cpu").manual_seed(7)   # same on all ranks
+        def train_step(model, opt, idx, targets):
+            return moem.moe_train_step(model, opt, idx, targets)
+        seq = cfg.block_size
+        idx = torch.randint(0, cfg.vocab_size, (global_batch, seq),
+                            generator=g).to(device)
+        tg = torch.randint(0, cfg.vocab_size, (global_batch, seq),
+                           generator=g).to(device)
+        tokens_per_sample = seq
+    else:
+        cfg = {
+            "gpt2-small": gptm.GPT2_SMALL,
+            "gpt2-medium": gptm.GPT2_MEDIUM,
+            "gpt2-1.3b": gptm.GPT2_1_3B,
+            "gpt-1l-12288": gptm.GPT_BENCH_1L,
+        }[args.model]
+        if args.seq != cfg.block_size:
+            from dataclasses import replace
+            cfg = replace(cfg, block_size=args.seq)
+        model = gptm.GPT(cfg).to(device)
+        opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
+
+        def train_step(model, opt, idx, targets):
+            return gptm.gpt_train_step(model, opt, idx, targets)
         idx = torch.randint(0, cfg.vocab_size, (global_batch, args.seq),
-                            generator=g, device="cpu").to(device)
+                            generator=g).to(device)
         tg = torch.randint(0, cfg.vocab_size, (global_batch, args.seq),
-                           generator=g, device="cpu").to(device)
-        return idx, tg
+                           generator=g).to(device)
+        tokens_per_sample = args.seq
 
-    idx, tg = batch()
+    # MoE EP runs module-parallel (each rank computes its shard directly)
+    moe_ep = args.model.startswith("mixtral") and world_size > 1
+    if moe_ep:
+        class _Direct:
+            def __init__(self):
+                self.compiled = {}
+
+            def __call__(self, model, opt, i, t):
+                local_i = torch.chunk(i, world_size, 0)[rank]
+                local_t = torch.chunk(t, world_size, 0)[rank]
+                return train_step(model, opt, local_i, local_t)
+        compiled = _Direct()
+    else:
+        compiled = easydist_compile(train_step, parallel_mode=args.parallel,
+                                    cuda_graph=args.hipgraph)
 
     t_compile = time.time()
     for _ in range(max(args.warmup, 1)):
@@ -127,8 +169,11 @@ def main():
     samples_per_sec = global_batch * args.steps / elapsed
 
     if rank == 0:
+        family = ("vit" if args.model.startswith("vit")
+                  else "mixtral_moe" if args.model.startswith("mixtral")
+                  else "gpt2")
         out = {
-            "metric": "gpt2_train_samples_per_sec",
+            "metric": f"{family}_train_samples_per_sec",
             "value": samples_per_sec,
             "unit": "samples/sec",
             "n_gpus": n_gpus,

@@ -107,14 +107,11 @@ class MoELayer(nn.Module):
         # [E, cap, C] is identical on every rank => equal-split all-to-all
         cap = _capacity(N, E, K, self.cfg.capacity_factor)
 
-        # scatter tokens into per-expert bins (dropping overflow, standard
-        # capacity routing)
+        # STATIC-SHAPE capacity routing: every (token, k) pair writes a
+        # slot; overflow pairs land in a per-expert TRASH slot (index
+        # `cap`) with gate 0. No data-dependent shapes -> whole-graph
+        # traceable and hipGraph-capturable.
         flat_expert = topi.reshape(-1)                 # [N*K]
-        bins = tokens.new_zeros(E, cap, C)
-        gates = tokens.new_zeros(E, cap)
-        src_index = torch.full((E, cap), -1, dtype=torch.long,
-                               device=x.device)
-        # position of each (token, k) within its expert's bin
         order = torch.argsort(flat_expert, stable=True)
         sorted_e = flat_expert[order]
         seg_start = torch.searchsorted(sorted_e, torch.arange(
@@ -122,14 +119,25 @@ class MoELayer(nn.Module):
         pos_in_seg = torch.arange(N * K, device=x.device) - \
             seg_start[sorted_e]
         keep = pos_in_seg < cap
-        kept_pairs = order[keep]                       # indices into N*K
-        kept_e = sorted_e[keep]
-        kept_pos = pos_in_seg[keep]
-        tok_idx = kept_pairs // K
-        k_idx = kept_pairs % K
-        bins[kept_e, kept_pos] = tokens[tok_idx]
-        gates[kept_e, kept_pos] = topv[tok_idx, k_idx]
-        src_index[kept_e, kept_pos] = tok_idx
+        slot = torch.where(keep, pos_in_seg,
+                           torch.full_like(pos_in_seg, cap))
+        tok_idx = order // K
+        k_idx = order % K
+        bins_x = tokens.new_zeros(E, cap + 1, C)
+        gates_x = topv.new_zeros(E, cap + 1)
+        src_x = torch.zeros(E, cap + 1, dtype=torch.long, device=x.device)
+        valid_x = torch.zeros(E, cap + 1, dtype=torch.bool,
+                              device=x.device)
+        bins_x[sorted_e, slot] = tokens[tok_idx]
+        gates_x[sorted_e, slot] = torch.where(
+            keep, topv[tok_idx, k_idx], torch.zeros_like(pos_in_seg,
+                                                         dtype=topv.dtype))
+        src_x[sorted_e, slot] = tok_idx
+        valid_x[sorted_e, slot] = keep
+        bins = bins_x[:, :cap].contiguous()
+        gates = gates_x[:, :cap]
+        src_index = src_x[:, :cap]
+        valid = valid_x[:, :cap]
 
         if self.ep_group is not None and self.ep_world > 1:
             from ..parallel import comm
@@ -149,12 +157,13 @@ class MoELayer(nn.Module):
         else:
             out_bins = self.experts(bins).reshape(E, cap, C)
 
-        # combine: weighted scatter-add back to token positions
+        # combine: weighted scatter-add back to token positions (invalid
+        # slots add zeros to token 0 — static shapes, no boolean select)
         out = tokens.new_zeros(N, C)
-        valid = src_index >= 0
-        flat_src = src_index[valid]
-        contrib = out_bins[valid] * gates[valid].unsqueeze(-1)
-        out.index_add_(0, flat_src, contrib.to(out.dtype))
+        contrib = out_bins * gates.unsqueeze(-1) \
+            * valid.unsqueeze(-1).to(out_bins.dtype)
+        out.index_add_(0, src_index.reshape(-1),
+                       contrib.reshape(-1, C).to(out.dtype))
         return out.reshape(B, T, C)
 
 

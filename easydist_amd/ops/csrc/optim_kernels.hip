@@ -131,6 +131,7 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
   // every replay) and record one async H2D memcpy node.
   static at::Tensor g_pinned, g_dev;
   static size_t g_capacity = 0;
+  static hipEvent_t g_h2d_done = nullptr;
   const size_t bytes = chunks.size() * sizeof(AdamChunk);
   bool capturing = at::cuda::currentStreamCaptureStatusMayInitCtx() !=
                    at::cuda::CaptureStatus::None;
@@ -146,10 +147,19 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                           .device(params[0].device()));
     g_capacity = bytes;
   }
+  // the PREVIOUS call's async H2D must have consumed the pinned buffer
+  // before the host overwrites it (race -> garbage chunk table -> NaNs)
+  if (g_h2d_done == nullptr)
+    C10_CUDA_CHECK(hipEventCreateWithFlags(&g_h2d_done,
+                                           hipEventDisableTiming));
+  else if (!capturing)
+    C10_CUDA_CHECK(hipEventSynchronize(g_h2d_done));
   memcpy(g_pinned.data_ptr(), chunks.data(), bytes);
   auto stream = at::cuda::getCurrentCUDAStream();
   C10_CUDA_CHECK(hipMemcpyAsync(g_dev.data_ptr(), g_pinned.data_ptr(),
                                 bytes, hipMemcpyHostToDevice, stream));
+  if (!capturing)
+    C10_CUDA_CHECK(hipEventRecord(g_h2d_done, stream));
   hipLaunchKernelGGL(fused_adam_kernel, dim3(chunks.size()), dim3(256), 0,
       stream, (const AdamChunk*)g_dev.data_ptr(), (int)chunks.size(),
       steps_flat.data_ptr<float>(), (float)lr, (float)beta1, (float)beta2,
