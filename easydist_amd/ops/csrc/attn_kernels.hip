@@ -38,11 +38,13 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   const int wave = threadIdx.x / WAVE;
   const int qr0 = qb0 + wave * 16;          // this wave's first q row
 
-  // LDS: K tile [KB][D] + V tile [KB][D] + P strips [4 waves][16][KB]
+  // LDS: K tile [KB][D] + V^T tile [D][KB] + P strips [4 waves][16][KB]
+  // (V transposed at stage time so the PV B-operand is a contiguous
+  // 16-byte vector load instead of 8 scalar gathers)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* k_lds = reinterpret_cast<bf16*>(smem);
-  bf16* v_lds = k_lds + KB * D;
-  bf16* p_lds = v_lds + KB * D + wave * 16 * KB;
+  bf16* vt_lds = k_lds + KB * D;
+  bf16* p_lds = vt_lds + KB * D + wave * 16 * KB;
 
   const int fr = lane & 15;        // fragment row/col index
   const int fg = lane >> 4;        // fragment k-group (8 contiguous)
@@ -70,8 +72,11 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       for (int e = threadIdx.x * 8; e < elems; e += 256 * 8) {
         *reinterpret_cast<bf16x8*>(&k_lds[e]) =
             *reinterpret_cast<const bf16x8*>(&k[(long)kv0 * D + e]);
-        *reinterpret_cast<bf16x8*>(&v_lds[e]) =
-            *reinterpret_cast<const bf16x8*>(&v[(long)kv0 * D + e]);
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(&v[(long)kv0 * D + e]);
+        const int row = e / D, col = e % D;
+        #pragma unroll
+        for (int i = 0; i < 8; ++i)
+          vt_lds[(col + i) * KB + row] = vv.v[i];
       }
     }
     __syncthreads();
@@ -154,16 +159,13 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     for (int ks = 0; ks < KB / 32; ++ks)
       pf[ks] = *reinterpret_cast<const bf16x8v*>(
           &p_lds[fr * KB + ks * 32 + fg * 8]);
-    // B operand: lane holds V[kv0 + 8*fg + jj][d0 + fr] — column reads
+    // B operand from V^T: contiguous vector load per mfma
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
       #pragma unroll
       for (int ks = 0; ks < KB / 32; ++ks) {
-        bf16x8v vf;
-        #pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          vf[jj] = *reinterpret_cast<const __bf16*>(
-              &v_lds[(ks * 32 + fg * 8 + jj) * D + j * 16 + fr]);
+        bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
+            &vt_lds[(j * 16 + fr) * KB + ks * 32 + fg * 8]);
         o_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf[ks], vf,
                                                            o_acc[j], 0, 0, 0);
       }
@@ -254,7 +256,8 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* k_lds = reinterpret_cast<bf16*>(smem);
   bf16* v_lds = k_lds + KB * D;
-  bf16* s_lds = v_lds + KB * D + wave * 16 * KB;   // wave-private strip
+  bf16* kt_lds = v_lds + KB * D;                   // K^T [D][KB]
+  bf16* s_lds = kt_lds + KB * D + wave * 16 * KB;  // wave-private strip
 
   // A-operand fragments for this wave's 16 q rows
   bf16x8v qf[D / 32], dof[D / 32];
@@ -282,10 +285,14 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
     __syncthreads();
     for (int e = threadIdx.x * 8; e < KB * D; e += 256 * 8) {
-      *reinterpret_cast<bf16x8*>(&k_lds[e]) =
-          *reinterpret_cast<const bf16x8*>(&k[(long)kv0 * D + e]);
+      bf16x8 kk = *reinterpret_cast<const bf16x8*>(&k[(long)kv0 * D + e]);
+      *reinterpret_cast<bf16x8*>(&k_lds[e]) = kk;
       *reinterpret_cast<bf16x8*>(&v_lds[e]) =
           *reinterpret_cast<const bf16x8*>(&v[(long)kv0 * D + e]);
+      const int row = e / D, col = e % D;
+      #pragma unroll
+      for (int i = 0; i < 8; ++i)
+        kt_lds[(col + i) * KB + row] = kk.v[i];
     }
     __syncthreads();
 
@@ -333,16 +340,13 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     for (int ks = 0; ks < KB / 32; ++ks)
       dsf[ks] = *reinterpret_cast<const bf16x8v*>(
           &s_lds[fr * KB + ks * 32 + fg * 8]);
-    // dq += dS @ K: B[n=d][k=c] = K[c][d] (column reads)
+    // dq += dS @ K: B-operand from K^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
       #pragma unroll
       for (int ks = 0; ks < KB / 32; ++ks) {
-        bf16x8v kcol;
-        #pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          kcol[jj] = *reinterpret_cast<const __bf16*>(
-              &k_lds[(ks * 32 + fg * 8 + jj) * D + j * 16 + fr]);
+        bf16x8v kcol = *reinterpret_cast<const bf16x8v*>(
+            &kt_lds[(j * 16 + fr) * KB + ks * 32 + fg * 8]);
         dq_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf[ks], kcol,
                                                             dq_acc[j], 0, 0,
                                                             0);
@@ -388,7 +392,9 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* q_lds = reinterpret_cast<bf16*>(smem);
   bf16* do_lds = q_lds + QB * D;
-  bf16* s_lds = do_lds + QB * D + wave * 16 * QB;
+  bf16* qt_lds = do_lds + QB * D;                  // Q^T [D][QB]
+  bf16* dot_lds = qt_lds + QB * D;                 // dO^T [D][QB]
+  bf16* s_lds = dot_lds + QB * D + wave * 16 * QB;
 
   bf16x8v kf[D / 32], vf[D / 32];
   #pragma unroll
@@ -410,10 +416,16 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   for (int q0 = q_start; q0 < S; q0 += QB) {
     __syncthreads();
     for (int e = threadIdx.x * 8; e < QB * D; e += 256 * 8) {
-      *reinterpret_cast<bf16x8*>(&q_lds[e]) =
-          *reinterpret_cast<const bf16x8*>(&q[(long)q0 * D + e]);
-      *reinterpret_cast<bf16x8*>(&do_lds[e]) =
-          *reinterpret_cast<const bf16x8*>(&dO_[(long)q0 * D + e]);
+      bf16x8 qq = *reinterpret_cast<const bf16x8*>(&q[(long)q0 * D + e]);
+      bf16x8 dd = *reinterpret_cast<const bf16x8*>(&dO_[(long)q0 * D + e]);
+      *reinterpret_cast<bf16x8*>(&q_lds[e]) = qq;
+      *reinterpret_cast<bf16x8*>(&do_lds[e]) = dd;
+      const int row = e / D, col = e % D;
+      #pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        qt_lds[(col + i) * QB + row] = qq.v[i];
+        dot_lds[(col + i) * QB + row] = dd.v[i];
+      }
     }
     __syncthreads();
 
@@ -459,16 +471,13 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     for (int ks = 0; ks < QB / 32; ++ks)
       ptf[ks] = *reinterpret_cast<const bf16x8v*>(
           &s_lds[fr * QB + ks * 32 + fg * 8]);
-    // dV += P^T @ dO: B[n=d][k=q] = dO[q][d] (column reads)
+    // dV += P^T @ dO: B-operand from dO^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
       #pragma unroll
       for (int ks = 0; ks < QB / 32; ++ks) {
-        bf16x8v docol;
-        #pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          docol[jj] = *reinterpret_cast<const __bf16*>(
-              &do_lds[(ks * 32 + fg * 8 + jj) * D + j * 16 + fr]);
+        bf16x8v docol = *reinterpret_cast<const bf16x8v*>(
+            &dot_lds[(j * 16 + fr) * QB + ks * 32 + fg * 8]);
         dv_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf[ks], docol,
                                                             dv_acc[j], 0, 0,
                                                             0);
@@ -492,16 +501,13 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     for (int ks = 0; ks < QB / 32; ++ks)
       dstf[ks] = *reinterpret_cast<const bf16x8v*>(
           &s_lds[fr * QB + ks * 32 + fg * 8]);
-    // dK += dS^T @ Q: B[n=d][k=q] = Q[q][d] (column reads)
+    // dK += dS^T @ Q: B-operand from Q^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
       #pragma unroll
       for (int ks = 0; ks < QB / 32; ++ks) {
-        bf16x8v qcolf;
-        #pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          qcolf[jj] = *reinterpret_cast<const __bf16*>(
-              &q_lds[(ks * 32 + fg * 8 + jj) * D + j * 16 + fr]);
+        bf16x8v qcolf = *reinterpret_cast<const bf16x8v*>(
+            &qt_lds[(j * 16 + fr) * QB + ks * 32 + fg * 8]);
         dk_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf[ks], qcolf,
                                                             dk_acc[j], 0, 0,
                                                             0);
@@ -538,7 +544,8 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
   auto dv = at::empty_like(v);
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid(S / QB, B * H), block(256);
-  size_t lds = (2 * KB * (size_t)D + 4 * 16 * KB) * 2;
+  size_t lds = (3 * KB * (size_t)D + 4 * 16 * KB) * 2;      // dq kernel
+  size_t lds_kv = (4 * KB * (size_t)D + 4 * 16 * KB) * 2;   // dkv kernel
   float scale = 1.f / sqrtf((float)D);
   if (D == 64) {
     hipLaunchKernelGGL(flash_bwd_dq_kernel<64>, grid, block, lds, stream,
@@ -546,7 +553,8 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
         (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
         lse.data_ptr<float>(), delta.data_ptr<float>(),
         (bf16*)dq.data_ptr(), B, H, S, causal, scale);
-    hipLaunchKernelGGL(flash_bwd_dkv_kernel<64>, grid, block, lds, stream,
+    hipLaunchKernelGGL(flash_bwd_dkv_kernel<64>, grid, block, lds_kv,
+        stream,
         (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
         (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
         lse.data_ptr<float>(), delta.data_ptr<float>(),
@@ -557,7 +565,8 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
         (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
         lse.data_ptr<float>(), delta.data_ptr<float>(),
         (bf16*)dq.data_ptr(), B, H, S, causal, scale);
-    hipLaunchKernelGGL(flash_bwd_dkv_kernel<128>, grid, block, lds, stream,
+    hipLaunchKernelGGL(flash_bwd_dkv_kernel<128>, grid, block, lds_kv,
+        stream,
         (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
         (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
         lse.data_ptr<float>(), delta.data_ptr<float>(),
