@@ -210,6 +210,8 @@ def shard_graph(gm, mesh, io_map, ret_names, device, fix_rets=True):
         _save_strategy_cache(cache_path,
                              (strategies_per_dim, search_time, solve_time))
 
+    _verify_strategy_agreement(strategies_per_dim)
+
     from ..utils.dumps import dump_graph, dump_strategies
     dump_strategies(strategies_per_dim, "auto")
     dump_graph(gm, "auto_pre_shard")
@@ -252,14 +254,53 @@ def _load_strategy_cache(path):
 
 
 def _save_strategy_cache(path, payload):
+    import os
     import pickle
     if path is None:
         return
     try:
-        with open(path, "wb") as f:
+        tmp = f"{path}.{os.getpid()}.tmp"
+        with open(tmp, "wb") as f:
             pickle.dump(payload, f)
+        os.replace(tmp, path)   # atomic: concurrent ranks never read partial
     except Exception as e:   # noqa: BLE001
         logger.warning("strategy cache write failed: %s", e)
+
+
+def _verify_strategy_agreement(strategies_per_dim):
+    """All ranks must have solved to the SAME strategy (the MILP is
+    deterministic, but a divergence would deadlock the collectives —
+    fail loudly instead). reference control plane: compile_auto.py:517
+    (rank0 + RPC broadcast); here every rank solves and we cross-check a
+    hash."""
+    import hashlib
+
+    import torch.distributed as dist
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        return
+    digest = hashlib.sha256(
+        repr([sorted((k, repr(v)) for k, v in d.items())
+              for d in strategies_per_dim]).encode()).digest()[:8]
+    t = torch.tensor(list(digest), dtype=torch.uint8)
+    if torch.cuda.is_available():
+        t = t.cuda()
+    ref = t.clone()
+    dist.broadcast(ref, src=0)
+    mismatch = torch.tensor(
+        [0 if torch.equal(t, ref) else 1],
+        device=t.device)
+    dist.all_reduce(mismatch, op=dist.ReduceOp.MAX)
+    if int(mismatch):
+        # resolve by adopting rank 0's strategies (mirrors the reference's
+        # rank0-solves + broadcast control plane); ALL ranks join the
+        # object broadcast
+        if not torch.equal(t, ref):
+            logger.warning("strategy solve diverged from rank 0: adopting "
+                           "rank 0's result")
+        payload = [strategies_per_dim]
+        dist.broadcast_object_list(payload, src=0)
+        strategies_per_dim.clear()
+        strategies_per_dim.extend(payload[0])
 
 
 def _adam_positions(params, buffers, named_states):
