@@ -17,6 +17,10 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 
 #define QB 64     // q rows per workgroup (16 per wave)
 #define KB 64     // kv tile
+#define KBP (KB + 8)  // padded stride for transposed tiles: row stride
+                      // 144 B spreads the 64 LDS banks (unpadded 128 B
+                      // collapses onto 2 banks -> 8-way conflicts)
+#define QBP (QB + 8)
 
 template <int D>
 __global__ void __launch_bounds__(256)
@@ -44,7 +48,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* k_lds = reinterpret_cast<bf16*>(smem);
   bf16* vt_lds = k_lds + KB * D;
-  bf16* p_lds = vt_lds + KB * D + wave * 16 * KB;
+  bf16* p_lds = vt_lds + KBP * D + wave * 16 * KB;
 
   const int fr = lane & 15;        // fragment row/col index
   const int fg = lane >> 4;        // fragment k-group (8 contiguous)
@@ -76,7 +80,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         const int row = e / D, col = e % D;
         #pragma unroll
         for (int i = 0; i < 8; ++i)
-          vt_lds[(col + i) * KB + row] = vv.v[i];
+          vt_lds[(col + i) * KBP + row] = vv.v[i];
       }
     }
     __syncthreads();
@@ -165,7 +169,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       #pragma unroll
       for (int ks = 0; ks < KB / 32; ++ks) {
         bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
-            &vt_lds[(j * 16 + fr) * KB + ks * 32 + fg * 8]);
+            &vt_lds[(j * 16 + fr) * KBP + ks * 32 + fg * 8]);
         o_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf[ks], vf,
                                                            o_acc[j], 0, 0, 0);
       }
@@ -201,7 +205,7 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid(S / QB, B * H), block(256);
-  size_t lds = (2 * KB * D + 4 * 16 * KB) * 2;
+  size_t lds = ((size_t)KB * D + (size_t)KBP * D + 4 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
   if (D == 64)
     hipLaunchKernelGGL(flash_fwd_kernel<64>, grid, block, lds, stream,
@@ -256,8 +260,8 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* k_lds = reinterpret_cast<bf16*>(smem);
   bf16* v_lds = k_lds + KB * D;
-  bf16* kt_lds = v_lds + KB * D;                   // K^T [D][KB]
-  bf16* s_lds = kt_lds + KB * D + wave * 16 * KB;  // wave-private strip
+  bf16* kt_lds = v_lds + KB * D;                   // K^T [D][KBP]
+  bf16* s_lds = kt_lds + KBP * D + wave * 16 * KB;  // wave-private strip
 
   // A-operand fragments for this wave's 16 q rows
   bf16x8v qf[D / 32], dof[D / 32];
@@ -292,7 +296,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i)
-        kt_lds[(col + i) * KB + row] = kk.v[i];
+        kt_lds[(col + i) * KBP + row] = kk.v[i];
     }
     __syncthreads();
 
@@ -346,7 +350,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < KB / 32; ++ks) {
         bf16x8v kcol = *reinterpret_cast<const bf16x8v*>(
-            &kt_lds[(j * 16 + fr) * KB + ks * 32 + fg * 8]);
+            &kt_lds[(j * 16 + fr) * KBP + ks * 32 + fg * 8]);
         dq_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf[ks], kcol,
                                                             dq_acc[j], 0, 0,
                                                             0);
@@ -392,9 +396,9 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* q_lds = reinterpret_cast<bf16*>(smem);
   bf16* do_lds = q_lds + QB * D;
-  bf16* qt_lds = do_lds + QB * D;                  // Q^T [D][QB]
-  bf16* dot_lds = qt_lds + QB * D;                 // dO^T [D][QB]
-  bf16* s_lds = dot_lds + QB * D + wave * 16 * QB;
+  bf16* qt_lds = do_lds + QB * D;                  // Q^T [D][QBP]
+  bf16* dot_lds = qt_lds + QBP * D;                // dO^T [D][QBP]
+  bf16* s_lds = dot_lds + QBP * D + wave * 16 * QB;
 
   bf16x8v kf[D / 32], vf[D / 32];
   #pragma unroll
@@ -423,8 +427,8 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        qt_lds[(col + i) * QB + row] = qq.v[i];
-        dot_lds[(col + i) * QB + row] = dd.v[i];
+        qt_lds[(col + i) * QBP + row] = qq.v[i];
+        dot_lds[(col + i) * QBP + row] = dd.v[i];
       }
     }
     __syncthreads();
@@ -477,7 +481,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < QB / 32; ++ks) {
         bf16x8v docol = *reinterpret_cast<const bf16x8v*>(
-            &dot_lds[(j * 16 + fr) * QB + ks * 32 + fg * 8]);
+            &dot_lds[(j * 16 + fr) * QBP + ks * 32 + fg * 8]);
         dv_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf[ks], docol,
                                                             dv_acc[j], 0, 0,
                                                             0);
@@ -507,7 +511,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < QB / 32; ++ks) {
         bf16x8v qcolf = *reinterpret_cast<const bf16x8v*>(
-            &qt_lds[(j * 16 + fr) * QB + ks * 32 + fg * 8]);
+            &qt_lds[(j * 16 + fr) * QBP + ks * 32 + fg * 8]);
         dk_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf[ks], qcolf,
                                                             dk_acc[j], 0, 0,
                                                             0);
@@ -544,8 +548,9 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
   auto dv = at::empty_like(v);
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid(S / QB, B * H), block(256);
-  size_t lds = (3 * KB * (size_t)D + 4 * 16 * KB) * 2;      // dq kernel
-  size_t lds_kv = (4 * KB * (size_t)D + 4 * 16 * KB) * 2;   // dkv kernel
+  size_t lds = (2 * KB * (size_t)D + (size_t)KBP * D + 4 * 16 * KB) * 2;
+  size_t lds_kv = (2 * QB * (size_t)D + 2 * (size_t)QBP * D
+                   + 4 * 16 * QB) * 2;
   float scale = 1.f / sqrtf((float)D);
   if (D == 64) {
     hipLaunchKernelGGL(flash_bwd_dq_kernel<64>, grid, block, lds, stream,
