@@ -42,13 +42,14 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   const int wave = threadIdx.x / WAVE;
   const int qr0 = qb0 + wave * 16;          // this wave's first q row
 
-  // LDS: K tile [KB][D] + V^T tile [D][KB] + P strips [4 waves][16][KB]
-  // (V transposed at stage time so the PV B-operand is a contiguous
-  // 16-byte vector load instead of 8 scalar gathers)
+  // LDS: DOUBLE-BUFFERED K [KB][D] + V^T [D][KBP] tile pairs + P strips.
+  // Global loads for tile t+1 land in registers while tile t's MFMA loop
+  // runs; ONE barrier per tile.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* k_lds = reinterpret_cast<bf16*>(smem);
-  bf16* vt_lds = k_lds + KB * D;
-  bf16* p_lds = vt_lds + KBP * D + wave * 16 * KB;
+  constexpr int TILE_K = KB * D;
+  constexpr int TILE_VT = KBP * D;
+  bf16* smem_b = reinterpret_cast<bf16*>(smem);
+  bf16* p_lds = smem_b + 2 * (TILE_K + TILE_VT) + wave * 16 * KB;
 
   const int fr = lane & 15;        // fragment row/col index
   const int fg = lane >> 4;        // fragment k-group (8 contiguous)
@@ -68,22 +69,42 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   for (int j = 0; j < D / 16; ++j) o_acc[j] = {0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = causal ? min(S, qb0 + QB) : S;
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
-    // ---- stage K/V tile: 256 threads, 16B each --------------------------
-    __syncthreads();
-    {
-      const int elems = KB * D;               // tile elements
-      for (int e = threadIdx.x * 8; e < elems; e += 256 * 8) {
-        *reinterpret_cast<bf16x8*>(&k_lds[e]) =
-            *reinterpret_cast<const bf16x8*>(&k[(long)kv0 * D + e]);
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(&v[(long)kv0 * D + e]);
-        const int row = e / D, col = e % D;
-        #pragma unroll
-        for (int i = 0; i < 8; ++i)
-          vt_lds[(col + i) * KBP + row] = vv.v[i];
-      }
+  const int n_tiles = (kv_end + KB - 1) / KB;
+  constexpr int PF = TILE_K / (256 * 8);     // 16B vectors per thread
+  bf16x8 kreg[PF], vreg[PF];
+
+  auto load_tile = [&](int t) {
+    #pragma unroll
+    for (int pi = 0; pi < PF; ++pi) {
+      const int e = threadIdx.x * 8 + pi * 2048;
+      kreg[pi] = *reinterpret_cast<const bf16x8*>(&k[(long)t * TILE_K + e]);
+      vreg[pi] = *reinterpret_cast<const bf16x8*>(&v[(long)t * TILE_K + e]);
     }
-    __syncthreads();
+  };
+  auto store_tile = [&](int t) {
+    bf16* kb = smem_b + (t & 1) * (TILE_K + TILE_VT);
+    bf16* vb = kb + TILE_K;
+    #pragma unroll
+    for (int pi = 0; pi < PF; ++pi) {
+      const int e = threadIdx.x * 8 + pi * 2048;
+      *reinterpret_cast<bf16x8*>(&kb[e]) = kreg[pi];
+      const int row = e / D, col = e % D;
+      #pragma unroll
+      for (int i = 0; i < 8; ++i)
+        vb[(col + i) * KBP + row] = vreg[pi].v[i];
+    }
+  };
+
+  load_tile(0);
+  store_tile(0);
+  __syncthreads();
+
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * KB;
+    const bf16* k_lds = smem_b + (t & 1) * (TILE_K + TILE_VT);
+    const bf16* vt_lds = k_lds + TILE_K;
+    if (t + 1 < n_tiles)
+      load_tile(t + 1);          // global loads overlap the MFMA loop
 
     // ---- S = Q K^T ------------------------------------------------------
     // 4 col-blocks of 16 keys; B operand: lane holds K[kv0+fr][8*fg..]
@@ -174,6 +195,10 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                                                            o_acc[j], 0, 0, 0);
       }
     }
+
+    if (t + 1 < n_tiles)
+      store_tile(t + 1);         // write the prefetched tile
+    __syncthreads();
   }
 
   // ---- epilogue --------------------------------------------------------
@@ -205,7 +230,7 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid(S / QB, B * H), block(256);
-  size_t lds = ((size_t)KB * D + (size_t)KBP * D + 4 * 16 * KB) * 2;
+  size_t lds = (2 * ((size_t)KB * D + (size_t)KBP * D) + 4 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
   if (D == 64)
     hipLaunchKernelGGL(flash_fwd_kernel<64>, grid, block, lds, stream,
