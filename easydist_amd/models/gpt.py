@@ -111,6 +111,11 @@ class GPT(nn.Module):
     def forward(self, idx):
         B, T = idx.shape
         x = self.wte(idx) + self.wpe(self.pos[:T])
+        if torch.is_autocast_enabled(x.device.type):
+            # embeddings are not on autocast's cast list: without this the
+            # whole residual stream runs fp32 (fp32 norms/adds + a cast
+            # pair around every matmul)
+            x = x.to(torch.bfloat16)
         for blk in self.h:
             x = blk(x)
         x = self.ln_f(x)

@@ -205,6 +205,8 @@ class MoEGPT(nn.Module):
     def forward(self, idx):
         B, T = idx.shape
         x = self.wte(idx) + self.wpe(self.pos[:T])
+        if torch.is_autocast_enabled(x.device.type):
+            x = x.to(torch.bfloat16)   # keep the residual stream bf16
         for blk in self.h:
             x = blk(x)
         return self.lm_head(self.ln_f(x))

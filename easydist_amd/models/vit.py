@@ -85,7 +85,8 @@ class ViT(nn.Module):
     def forward(self, x):
         B = x.shape[0]
         x = self.patch_embed(x).flatten(2).transpose(1, 2)   # [B, N, C]
-        x = torch.cat([self.cls.expand(B, -1, -1), x], dim=1) + self.pos
+        x = torch.cat([self.cls.expand(B, -1, -1).to(x.dtype), x],
+                      dim=1) + self.pos.to(x.dtype)
         for blk in self.blocks:
             x = blk(x)
         return self.head(self.ln_f(x)[:, 0])

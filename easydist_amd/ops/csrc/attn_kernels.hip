@@ -176,7 +176,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         p_lds[(4 * fg + r) * KB + j * 16 + fr] = f2bf(s_acc[j][r]);
     // wave-private strip; cross-lane visibility within the wave needs a
     // data-share sync, which s_waitcnt lgkmcnt(0) provides per-wave
-    __builtin_amdgcn_s_waitcnt(0);
+    lds_fence();
 
     // A operand: lane holds P[fr][8*fg..+8]
     bf16x8v pf[KB / 32];
@@ -363,7 +363,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int r = 0; r < 4; ++r)
         s_lds[(4 * fg + r) * KB + j * 16 + fr] = f2bf(s_acc[j][r]);
-    __builtin_amdgcn_s_waitcnt(0);
+    lds_fence();
     bf16x8v dsf[KB / 32];
     #pragma unroll
     for (int ks = 0; ks < KB / 32; ++ks)
@@ -494,7 +494,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
         s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(p);
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    lds_fence();
     bf16x8v ptf[QB / 32];
     #pragma unroll
     for (int ks = 0; ks < QB / 32; ++ks)
@@ -513,7 +513,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       }
     }
     // dS^T into the strip for the dK mfma
-    __builtin_amdgcn_s_waitcnt(0);
+    lds_fence();
     #pragma unroll
     for (int j = 0; j < QB / 16; ++j) {
       int qcol = q0 + j * 16 + fr;
@@ -524,7 +524,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
         s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(ds);
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    lds_fence();
     bf16x8v dstf[QB / 32];
     #pragma unroll
     for (int ks = 0; ks < QB / 32; ++ks)

@@ -30,3 +30,10 @@ DEVINL float wave_max(float x) {
     x = fmaxf(x, __shfl_down(x, off, WAVE));
   return __shfl(x, 0, WAVE);
 }
+
+
+// wait for LDS (lgkm) traffic only — unlike s_waitcnt(0) this does NOT
+// flush vmcnt, so in-flight global prefetch loads keep overlapping.
+// imm encoding (gfx90a+): vmcnt=63 (bits 3:0 + 15:14), expcnt=7 (6:4),
+// lgkmcnt=0 (13:8)
+DEVINL void lds_fence() { __builtin_amdgcn_s_waitcnt(0xC07F); }
