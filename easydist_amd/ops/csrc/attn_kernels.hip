@@ -28,8 +28,12 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                  const bf16* __restrict__ V, bf16* __restrict__ O,
                  float* __restrict__ LSE, int B, int H, int S, bool causal,
                  float scale) {
-  // grid: (S/QB, B*H)
-  const int qb0 = blockIdx.x * QB;
+  // grid: (ceil(S/QBLK), B*H); QBLK = 128 q rows per workgroup — each of
+  // the 4 waves owns TWO 16-row fragments (RF=2), doubling the MFMA work
+  // per staged K/V tile (arithmetic intensity) at the same LDS footprint.
+  constexpr int RF = 2;
+  constexpr int QBLK = 4 * 16 * RF;
+  const int qb0 = blockIdx.x * QBLK;
   const int bh = blockIdx.y;
   const long base = (long)bh * S * D;
   const bf16* q = Q + base;
@@ -40,11 +44,8 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
 
   const int lane = threadIdx.x % WAVE;
   const int wave = threadIdx.x / WAVE;
-  const int qr0 = qb0 + wave * 16;          // this wave's first q row
 
   // LDS: DOUBLE-BUFFERED K [KB][D] + V^T [D][KBP] tile pairs + P strips.
-  // Global loads for tile t+1 land in registers while tile t's MFMA loop
-  // runs; ONE barrier per tile.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr int TILE_K = KB * D;
   constexpr int TILE_VT = KBP * D;
@@ -54,21 +55,38 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   const int fr = lane & 15;        // fragment row/col index
   const int fg = lane >> 4;        // fragment k-group (8 contiguous)
 
-  // Q fragments: A operand rows = q rows; lane holds q[qr0+fr][8*fg..+8]
-  bf16x8v qf[D / 32];
+  // per-fragment first q row (two contiguous 16-row fragments per wave)
+  int qr0[RF];
   #pragma unroll
-  for (int ks = 0; ks < D / 32; ++ks)
-    qf[ks] = *reinterpret_cast<const bf16x8v*>(
-        &q[(long)(qr0 + fr) * D + ks * 32 + fg * 8]);
+  for (int rf = 0; rf < RF; ++rf) qr0[rf] = qb0 + wave * 16 * RF + rf * 16;
 
-  // online state: rows 4*fg..4*fg+3 of this wave's 16 (per C-layout)
-  float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
-  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
-  f32x4 o_acc[D / 16];
+  // Q fragments (A operand): lane holds q[qr0+fr][8*fg..+8]
+  bf16x8v qf[RF][D / 32];
   #pragma unroll
-  for (int j = 0; j < D / 16; ++j) o_acc[j] = {0.f, 0.f, 0.f, 0.f};
+  for (int rf = 0; rf < RF; ++rf) {
+    const int row = qr0[rf] + fr;
+    #pragma unroll
+    for (int ks = 0; ks < D / 32; ++ks) {
+      if (row < S)
+        qf[rf][ks] = *reinterpret_cast<const bf16x8v*>(
+            &q[(long)row * D + ks * 32 + fg * 8]);
+      else
+        #pragma unroll
+        for (int u = 0; u < 8; ++u) qf[rf][ks][u] = (__bf16)0.f;
+    }
+  }
 
-  const int kv_end = causal ? min(S, qb0 + QB) : S;
+  float m_run[RF][4], l_run[RF][4];
+  f32x4 o_acc[RF][D / 16];
+  #pragma unroll
+  for (int rf = 0; rf < RF; ++rf) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) { m_run[rf][r] = -1e30f; l_run[rf][r] = 0.f; }
+    #pragma unroll
+    for (int j = 0; j < D / 16; ++j) o_acc[rf][j] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int kv_end = causal ? min(S, qb0 + QBLK) : S;
   const int n_tiles = (kv_end + KB - 1) / KB;
   constexpr int PF = TILE_K / (256 * 8);     // 16B vectors per thread
   bf16x8 kreg[PF], vreg[PF];
@@ -106,93 +124,87 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     if (t + 1 < n_tiles)
       load_tile(t + 1);          // global loads overlap the MFMA loop
 
-    // ---- S = Q K^T ------------------------------------------------------
-    // 4 col-blocks of 16 keys; B operand: lane holds K[kv0+fr][8*fg..]
-    f32x4 s_acc[KB / 16];
     #pragma unroll
-    for (int j = 0; j < KB / 16; ++j) {
-      s_acc[j] = {0.f, 0.f, 0.f, 0.f};
-      #pragma unroll
-      for (int ks = 0; ks < D / 32; ++ks) {
-        bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
-            &k_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
-        s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], kf,
-                                                           s_acc[j], 0, 0, 0);
-      }
-    }
-
-    // ---- online softmax --------------------------------------------------
-    // lane holds S[4*fg+r][j*16 + fr] for r=0..3
-    float m_new[4], row_max[4];
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float mx = -1e30f;
+    for (int rf = 0; rf < RF; ++rf) {
+      // ---- S = Q K^T ----------------------------------------------------
+      f32x4 s_acc[KB / 16];
       #pragma unroll
       for (int j = 0; j < KB / 16; ++j) {
-        float sv = s_acc[j][r] * scale;
-        int kcol = kv0 + j * 16 + fr;
-        int qrow = qr0 + 4 * fg + r;
-        if (causal && kcol > qrow) sv = -1e30f;
-        else if (kcol >= S) sv = -1e30f;
-        s_acc[j][r] = sv;
-        mx = fmaxf(mx, sv);
+        s_acc[j] = {0.f, 0.f, 0.f, 0.f};
+        #pragma unroll
+        for (int ks = 0; ks < D / 32; ++ks) {
+          bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
+              &k_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+          s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[rf][ks], kf,
+                                                             s_acc[j], 0, 0,
+                                                             0);
+        }
       }
-      // max across the 16 lanes of this fragment group
-      #pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
-      row_max[r] = mx;
-      m_new[r] = fmaxf(m_run[r], mx);
-    }
-    float p_sum[4] = {0.f, 0.f, 0.f, 0.f};
-    #pragma unroll
-    for (int j = 0; j < KB / 16; ++j) {
+
+      // ---- online softmax ----------------------------------------------
+      float m_new[4];
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float p = __expf(s_acc[j][r] - m_new[r]);
-        s_acc[j][r] = p;
-        p_sum[r] += p;
+        float mx = -1e30f;
+        #pragma unroll
+        for (int j = 0; j < KB / 16; ++j) {
+          float sv = s_acc[j][r] * scale;
+          int kcol = kv0 + j * 16 + fr;
+          int qrow = qr0[rf] + 4 * fg + r;
+          if (causal && kcol > qrow) sv = -1e30f;
+          else if (kcol >= S) sv = -1e30f;
+          s_acc[j][r] = sv;
+          mx = fmaxf(mx, sv);
+        }
+        #pragma unroll
+        for (int off = 8; off > 0; off >>= 1)
+          mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+        m_new[r] = fmaxf(m_run[rf][r], mx);
       }
-    }
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
+      float p_sum[4] = {0.f, 0.f, 0.f, 0.f};
       #pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        p_sum[r] += __shfl_xor(p_sum[r], off, WAVE);
-      float alpha = __expf(m_run[r] - m_new[r]);
-      l_run[r] = l_run[r] * alpha + p_sum[r];
-      m_run[r] = m_new[r];
-      // rescale o accumulator rows
+      for (int j = 0; j < KB / 16; ++j) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float pp = __expf(s_acc[j][r] - m_new[r]);
+          s_acc[j][r] = pp;
+          p_sum[r] += pp;
+        }
+      }
       #pragma unroll
-      for (int j = 0; j < D / 16; ++j) o_acc[j][r] *= alpha;
-    }
+      for (int r = 0; r < 4; ++r) {
+        #pragma unroll
+        for (int off = 8; off > 0; off >>= 1)
+          p_sum[r] += __shfl_xor(p_sum[r], off, WAVE);
+        float alpha = __expf(m_run[rf][r] - m_new[r]);
+        l_run[rf][r] = l_run[rf][r] * alpha + p_sum[r];
+        m_run[rf][r] = m_new[r];
+        #pragma unroll
+        for (int j = 0; j < D / 16; ++j) o_acc[rf][j][r] *= alpha;
+      }
 
-    // ---- P to LDS strip (re-fragment), then P @ V -----------------------
-    // write: P[4*fg+r][j*16+fr]
-    #pragma unroll
-    for (int j = 0; j < KB / 16; ++j)
+      // ---- P through the wave strip, then P @ V -------------------------
       #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        p_lds[(4 * fg + r) * KB + j * 16 + fr] = f2bf(s_acc[j][r]);
-    // wave-private strip; cross-lane visibility within the wave needs a
-    // data-share sync, which s_waitcnt lgkmcnt(0) provides per-wave
-    lds_fence();
-
-    // A operand: lane holds P[fr][8*fg..+8]
-    bf16x8v pf[KB / 32];
-    #pragma unroll
-    for (int ks = 0; ks < KB / 32; ++ks)
-      pf[ks] = *reinterpret_cast<const bf16x8v*>(
-          &p_lds[fr * KB + ks * 32 + fg * 8]);
-    // B operand from V^T: contiguous vector load per mfma
-    #pragma unroll
-    for (int j = 0; j < D / 16; ++j) {
+      for (int j = 0; j < KB / 16; ++j)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          p_lds[(4 * fg + r) * KB + j * 16 + fr] = f2bf(s_acc[j][r]);
+      lds_fence();
+      bf16x8v pf[KB / 32];
       #pragma unroll
-      for (int ks = 0; ks < KB / 32; ++ks) {
-        bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
-            &vt_lds[(j * 16 + fr) * KBP + ks * 32 + fg * 8]);
-        o_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf[ks], vf,
-                                                           o_acc[j], 0, 0, 0);
+      for (int ks = 0; ks < KB / 32; ++ks)
+        pf[ks] = *reinterpret_cast<const bf16x8v*>(
+            &p_lds[fr * KB + ks * 32 + fg * 8]);
+      lds_fence();   // strip is reused by the next fragment
+      #pragma unroll
+      for (int j = 0; j < D / 16; ++j) {
+        #pragma unroll
+        for (int ks = 0; ks < KB / 32; ++ks) {
+          bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
+              &vt_lds[(j * 16 + fr) * KBP + ks * 32 + fg * 8]);
+          o_acc[rf][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pf[ks], vf, o_acc[rf][j], 0, 0, 0);
+        }
       }
     }
 
@@ -203,15 +215,18 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
 
   // ---- epilogue --------------------------------------------------------
   #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int qrow = qr0 + 4 * fg + r;
-    if (qrow >= S) continue;
-    float inv_l = 1.f / l_run[r];
+  for (int rf = 0; rf < RF; ++rf) {
     #pragma unroll
-    for (int j = 0; j < D / 16; ++j)
-      o[(long)qrow * D + j * 16 + fr] = f2bf(o_acc[j][r] * inv_l);
-    if (fr == 0)
-      lse[qrow] = m_run[r] + __logf(l_run[r]);
+    for (int r = 0; r < 4; ++r) {
+      int qrow = qr0[rf] + 4 * fg + r;
+      if (qrow >= S) continue;
+      float inv_l = 1.f / l_run[rf][r];
+      #pragma unroll
+      for (int j = 0; j < D / 16; ++j)
+        o[(long)qrow * D + j * 16 + fr] = f2bf(o_acc[rf][j][r] * inv_l);
+      if (fr == 0)
+        lse[qrow] = m_run[rf][r] + __logf(l_run[rf][r]);
+    }
   }
 }
 
@@ -229,7 +244,7 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
   auto out = at::empty_like(q);
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
-  dim3 grid(S / QB, B * H), block(256);
+  dim3 grid((S + 127) / 128, B * H), block(256);
   size_t lds = (2 * ((size_t)KB * D + (size_t)KBP * D) + 4 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
   if (D == 64)
