@@ -142,24 +142,45 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       }
 
       // ---- online softmax ----------------------------------------------
+      // interior tiles (fully below the causal diagonal, fully in-range)
+      // skip the per-element mask compares — most tiles at long S
+      const bool need_mask = (causal && kv0 + KB - 1 > qr0[rf] + 4 * fg)
+                             || (kv0 + KB > S);
       float m_new[4];
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float mx = -1e30f;
+      if (need_mask) {
         #pragma unroll
-        for (int j = 0; j < KB / 16; ++j) {
-          float sv = s_acc[j][r] * scale;
-          int kcol = kv0 + j * 16 + fr;
-          int qrow = qr0[rf] + 4 * fg + r;
-          if (causal && kcol > qrow) sv = -1e30f;
-          else if (kcol >= S) sv = -1e30f;
-          s_acc[j][r] = sv;
-          mx = fmaxf(mx, sv);
+        for (int r = 0; r < 4; ++r) {
+          float mx = -1e30f;
+          #pragma unroll
+          for (int j = 0; j < KB / 16; ++j) {
+            float sv = s_acc[j][r] * scale;
+            int kcol = kv0 + j * 16 + fr;
+            int qrow = qr0[rf] + 4 * fg + r;
+            if (causal && kcol > qrow) sv = -1e30f;
+            else if (kcol >= S) sv = -1e30f;
+            s_acc[j][r] = sv;
+            mx = fmaxf(mx, sv);
+          }
+          #pragma unroll
+          for (int off = 8; off > 0; off >>= 1)
+            mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+          m_new[r] = fmaxf(m_run[rf][r], mx);
         }
+      } else {
         #pragma unroll
-        for (int off = 8; off > 0; off >>= 1)
-          mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
-        m_new[r] = fmaxf(m_run[rf][r], mx);
+        for (int r = 0; r < 4; ++r) {
+          float mx = -1e30f;
+          #pragma unroll
+          for (int j = 0; j < KB / 16; ++j) {
+            float sv = s_acc[j][r] * scale;
+            s_acc[j][r] = sv;
+            mx = fmaxf(mx, sv);
+          }
+          #pragma unroll
+          for (int off = 8; off > 0; off >>= 1)
+            mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+          m_new[r] = fmaxf(m_run[rf][r], mx);
+        }
       }
       float p_sum[4] = {0.f, 0.f, 0.f, 0.f};
       #pragma unroll

@@ -13,6 +13,29 @@ DEVINL float loadf(const T* p, int i);
 template <> DEVINL float loadf<bf16>(const bf16* p, int i) { return bf2f(p[i]); }
 template <> DEVINL float loadf<float>(const float* p, int i) { return p[i]; }
 
+// 8-element vector load/store (one 16B dword4 for bf16, two for fp32)
+DEVINL void vload8(const bf16* p, int i, float* out) {
+  bf16x8 v = *reinterpret_cast<const bf16x8*>(&p[i]);
+  #pragma unroll
+  for (int u = 0; u < 8; ++u) out[u] = bf2f(v.v[u]);
+}
+DEVINL void vload8(const float* p, int i, float* out) {
+  f32x4v a = *reinterpret_cast<const f32x4v*>(&p[i]);
+  f32x4v b = *reinterpret_cast<const f32x4v*>(&p[i + 4]);
+  out[0] = a.x; out[1] = a.y; out[2] = a.z; out[3] = a.w;
+  out[4] = b.x; out[5] = b.y; out[6] = b.z; out[7] = b.w;
+}
+DEVINL void vstore8(bf16* p, int i, const bf16* v) {
+  bf16x8 o;
+  #pragma unroll
+  for (int u = 0; u < 8; ++u) o.v[u] = v[u];
+  *reinterpret_cast<bf16x8*>(&p[i]) = o;
+}
+DEVINL void vstore8(float* p, int i, const float* v) {
+  #pragma unroll
+  for (int u = 0; u < 8; ++u) p[i + u] = v[u];
+}
+
 template <typename T>
 __global__ void ce_fwd_kernel(const T* __restrict__ logits,
                               const long* __restrict__ targets,
@@ -29,8 +52,7 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
   const int V8 = (V / 8) * 8;
   for (int i = lane * 8; i < V8; i += WAVE * 8) {
     float v8[8];
-    #pragma unroll
-    for (int u = 0; u < 8; ++u) v8[u] = loadf(lrow, i + u);
+    vload8(lrow, i, v8);
     #pragma unroll
     for (int u = 0; u < 8; ++u) {
       float m2 = fmaxf(m, v8[u]);
@@ -72,16 +94,17 @@ __global__ void ce_bwd_kernel(const float* __restrict__ grad_scalar,
   const long tg = targets[row];
   const int V8 = (V / 8) * 8;
   for (int i = lane * 8; i < V8; i += WAVE * 8) {
+    float v8[8];
+    vload8(lrow, i, v8);
     T out8[8];
     #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      float p = __expf(loadf(lrow, i + u) - l);
+      float p = __expf(v8[u] - l);
       if (i + u == tg) p -= 1.f;
       if constexpr (sizeof(T) == 2) out8[u] = f2bf(p * scale);
       else out8[u] = p * scale;
     }
-    #pragma unroll
-    for (int u = 0; u < 8; ++u) drow[i + u] = out8[u];
+    vstore8(drow, i, out8);
   }
   for (int i = V8 + lane; i < V; i += WAVE) {
     float p = __expf(loadf(lrow, i) - l);
