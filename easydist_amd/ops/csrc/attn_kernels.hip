@@ -381,16 +381,29 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       }
     }
     // dS = P * (dP - delta) * scale, P = exp(S*scale - lse)
-    #pragma unroll
-    for (int j = 0; j < KB / 16; ++j) {
+    const bool need_mask = (causal && kv0 + KB - 1 > qr0 + 4 * fg)
+                           || (kv0 + KB > S);
+    if (need_mask) {
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int kcol = kv0 + j * 16 + fr;
-        int qrow = qr0 + 4 * fg + r;
-        float p = 0.f;
-        if (!(causal && kcol > qrow) && kcol < S)
-          p = __expf(s_acc[j][r] * scale - lse_r[r]);
-        s_acc[j][r] = p * (dp_acc[j][r] - dlt_r[r]) * scale;
+      for (int j = 0; j < KB / 16; ++j) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int kcol = kv0 + j * 16 + fr;
+          int qrow = qr0 + 4 * fg + r;
+          float p = 0.f;
+          if (!(causal && kcol > qrow) && kcol < S)
+            p = __expf(s_acc[j][r] * scale - lse_r[r]);
+          s_acc[j][r] = p * (dp_acc[j][r] - dlt_r[r]) * scale;
+        }
+      }
+    } else {
+      #pragma unroll
+      for (int j = 0; j < KB / 16; ++j) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float p = __expf(s_acc[j][r] * scale - lse_r[r]);
+          s_acc[j][r] = p * (dp_acc[j][r] - dlt_r[r]) * scale;
+        }
       }
     }
     // re-fragment dS through the wave strip
@@ -516,18 +529,33 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     }
     // P^T = exp(S^T*scale - lse[qcol]); dS^T = P^T (dP^T - delta[qcol]) scale
     // first pass: P^T into the strip for the dV mfma
-    #pragma unroll
-    for (int j = 0; j < QB / 16; ++j) {
-      int qcol = q0 + j * 16 + fr;
-      float l = lse[qcol];
+    // no-mask fast path: whole wave's keys are <= every q col in tile
+    const bool need_mask = (causal && kr0 + 15 > q0) || (q0 + QB > S);
+    if (need_mask) {
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int krow = kr0 + 4 * fg + r;
-        float p = 0.f;
-        if (!(causal && krow > qcol) && qcol < S)
-          p = __expf(st_acc[j][r] * scale - l);
-        st_acc[j][r] = p;
-        s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(p);
+      for (int j = 0; j < QB / 16; ++j) {
+        int qcol = q0 + j * 16 + fr;
+        float l = lse[qcol];
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int krow = kr0 + 4 * fg + r;
+          float p = 0.f;
+          if (!(causal && krow > qcol) && qcol < S)
+            p = __expf(st_acc[j][r] * scale - l);
+          st_acc[j][r] = p;
+          s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(p);
+        }
+      }
+    } else {
+      #pragma unroll
+      for (int j = 0; j < QB / 16; ++j) {
+        float l = lse[q0 + j * 16 + fr];
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float p = __expf(st_acc[j][r] * scale - l);
+          st_acc[j][r] = p;
+          s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(p);
+        }
       }
     }
     lds_fence();

@@ -14,6 +14,8 @@ pytestmark = pytest.mark.gpu
 SCRIPT = r"""
 import json, os, torch
 os.environ["EASYDIST_MEM_OPT"] = "1"
+os.environ["MASTER_ADDR"] = "127.0.0.1"
+os.environ["MASTER_PORT"] = "29613"   # parent pytest holds the default port
 import easydist_amd.config as cfg
 cfg.enable_memory_opt = True
 from easydist_amd import easydist_compile, easydist_setup, set_device_mesh
