@@ -83,6 +83,7 @@ def init_single_process(backend: str = None, port: int = 29599):
     if dist.is_initialized():
         return
     backend = backend or ("nccl" if torch.cuda.is_available() else "gloo")
+    port = int(os.environ.get("MASTER_PORT", port))   # env wins
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", str(port))
     dist.init_process_group(backend=backend, rank=0, world_size=1,
