@@ -60,19 +60,26 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   #pragma unroll
   for (int rf = 0; rf < RF; ++rf) qr0[rf] = qb0 + wave * 16 * RF + rf * 16;
 
-  // Q fragments (A operand): lane holds q[qr0+fr][8*fg..+8]
+  // Q fragments (A operand), PRE-SCALED by scale*log2(e): the softmax
+  // runs in exp2 domain (v_exp2 without the fused multiply) and the hot
+  // loop loses its per-element scale multiply
+  const float qscale = scale * 1.44269504088896340736f;
   bf16x8v qf[RF][D / 32];
   #pragma unroll
   for (int rf = 0; rf < RF; ++rf) {
     const int row = qr0[rf] + fr;
     #pragma unroll
     for (int ks = 0; ks < D / 32; ++ks) {
-      if (row < S)
-        qf[rf][ks] = *reinterpret_cast<const bf16x8v*>(
+      if (row < S) {
+        bf16x8v raw = *reinterpret_cast<const bf16x8v*>(
             &q[(long)row * D + ks * 32 + fg * 8]);
-      else
+        #pragma unroll
+        for (int u = 0; u < 8; ++u)
+          qf[rf][ks][u] = (__bf16)(bf2f((bf16)raw[u]) * qscale);
+      } else {
         #pragma unroll
         for (int u = 0; u < 8; ++u) qf[rf][ks][u] = (__bf16)0.f;
+      }
     }
   }
 
@@ -153,7 +160,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
           float mx = -1e30f;
           #pragma unroll
           for (int j = 0; j < KB / 16; ++j) {
-            float sv = s_acc[j][r] * scale;
+            float sv = s_acc[j][r];        // already scale*log2e scaled
             int kcol = kv0 + j * 16 + fr;
             int qrow = qr0[rf] + 4 * fg + r;
             if (causal && kcol > qrow) sv = -1e30f;
@@ -169,13 +176,8 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       } else {
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          float mx = -1e30f;
-          #pragma unroll
-          for (int j = 0; j < KB / 16; ++j) {
-            float sv = s_acc[j][r] * scale;
-            s_acc[j][r] = sv;
-            mx = fmaxf(mx, sv);
-          }
+          float mx = fmaxf(fmaxf(s_acc[0][r], s_acc[1][r]),
+                           fmaxf(s_acc[2][r], s_acc[3][r]));
           #pragma unroll
           for (int off = 8; off > 0; off >>= 1)
             mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
@@ -187,7 +189,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       for (int j = 0; j < KB / 16; ++j) {
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          float pp = __expf(s_acc[j][r] - m_new[r]);
+          float pp = exp2f(s_acc[j][r] - m_new[r]);
           s_acc[j][r] = pp;
           p_sum[r] += pp;
         }
@@ -197,7 +199,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         #pragma unroll
         for (int off = 8; off > 0; off >>= 1)
           p_sum[r] += __shfl_xor(p_sum[r], off, WAVE);
-        float alpha = __expf(m_run[rf][r] - m_new[r]);
+        float alpha = exp2f(m_run[rf][r] - m_new[r]);
         l_run[rf][r] = l_run[rf][r] * alpha + p_sum[r];
         m_run[rf][r] = m_new[r];
         #pragma unroll
@@ -245,8 +247,9 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       #pragma unroll
       for (int j = 0; j < D / 16; ++j)
         o[(long)qrow * D + j * 16 + fr] = f2bf(o_acc[rf][j][r] * inv_l);
-      if (fr == 0)
-        lse[qrow] = m_run[rf][r] + __logf(l_run[rf][r]);
+      if (fr == 0)   // m_run is log2-domain: lse = ln(sum exp) = ln2*(m2 + log2 l)
+        lse[qrow] = 0.69314718055994530942f
+                    * (m_run[rf][r] + __log2f(l_run[rf][r]));
     }
   }
 }
