@@ -189,7 +189,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       for (int j = 0; j < KB / 16; ++j) {
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          float pp = exp2f(s_acc[j][r] - m_new[r]);
+          float pp = __builtin_amdgcn_exp2f(s_acc[j][r] - m_new[r]);
           s_acc[j][r] = pp;
           p_sum[r] += pp;
         }
@@ -199,7 +199,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         #pragma unroll
         for (int off = 8; off > 0; off >>= 1)
           p_sum[r] += __shfl_xor(p_sum[r], off, WAVE);
-        float alpha = exp2f(m_run[rf][r] - m_new[r]);
+        float alpha = __builtin_amdgcn_exp2f(m_run[rf][r] - m_new[r]);
         l_run[rf][r] = l_run[rf][r] * alpha + p_sum[r];
         m_run[rf][r] = m_new[r];
         #pragma unroll
