@@ -99,14 +99,6 @@ def lower_layer_norm(gm: fx.GraphModule) -> int:
     return n_lowered
 
 
-def _dtype_of_item(tuple_node: fx.Node, k: int) -> Optional[torch.dtype]:
-    for u in tuple_node.users:
-        if u.op == "call_function" and u.target is operator.getitem \
-                and u.args[1] == k:
-            return _dtype_of(u)
-    return None
-
-
 def _rewire_tuple(graph: fx.Graph, old: fx.Node, new: fx.Node,
                   kernel_dtypes):
     """Point every getitem(old, k) at getitem(new, k). A consumer that is

@@ -8,47 +8,8 @@ driver supports); a "handle" on the wire is 8 key bytes + the pickled
 """
 from __future__ import annotations
 
-import ctypes
 import socket
 from typing import Optional, Tuple
-
-HIP_IPC_HANDLE_SIZE = 64
-hipMemcpyHostToDevice = 1
-hipMemcpyDeviceToHost = 2
-hipMemcpyDeviceToDevice = 3
-
-
-class _HipClient:
-    def __init__(self, device: int = 0):
-        self.lib = ctypes.CDLL("libamdhip64.so")
-        self.lib.hipIpcOpenMemHandle.argtypes = [
-            ctypes.POINTER(ctypes.c_void_p), ctypes.c_char_p,
-            ctypes.c_uint]
-        self.lib.hipIpcCloseMemHandle.argtypes = [ctypes.c_void_p]
-        self.lib.hipMemcpy.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
-                                       ctypes.c_size_t, ctypes.c_int]
-        self.lib.hipSetDevice.argtypes = [ctypes.c_int]
-        # opening an IPC handle needs an active device context
-        rc = self.lib.hipSetDevice(device)
-        if rc != 0:
-            raise RuntimeError(f"hipSetDevice({device}) rc={rc}")
-
-    def open_handle(self, handle: bytes) -> int:
-        p = ctypes.c_void_p()
-        # 1 == hipIpcMemLazyEnablePeerAccess
-        rc = self.lib.hipIpcOpenMemHandle(ctypes.byref(p), handle, 1)
-        if rc != 0:
-            raise RuntimeError(f"hipIpcOpenMemHandle rc={rc}")
-        return p.value
-
-    def close_handle(self, ptr: int):
-        self.lib.hipIpcCloseMemHandle(ctypes.c_void_p(ptr))
-
-    def memcpy(self, dst: int, src: int, size: int, kind: int):
-        rc = self.lib.hipMemcpy(ctypes.c_void_p(dst), ctypes.c_void_p(src),
-                                size, kind)
-        if rc != 0:
-            raise RuntimeError(f"hipMemcpy rc={rc}")
 
 
 class TFieldClient:
