@@ -80,3 +80,49 @@ def test_dp_ws1(mode):
 @pytest.mark.parametrize("mode", ["ddp", "zero2", "zero3"])
 def test_dp_ws2(mode):
     spawn(_run_golden, args=(2, mode), world_size=2, port=29536)
+
+
+def _gpt_zero_body(world_size, mode):
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+    from easydist_amd.models.gpt import GPT, GPTConfig
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+    torch.manual_seed(0)
+    cfg = GPTConfig(vocab_size=128, n_layer=2, n_head=2, n_embd=32,
+                    block_size=16)
+    model = GPT(cfg)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+    opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-3, fused=True)
+
+    def step(model, opt, idx, tg):
+        loss = model.loss(idx, tg)
+        loss.backward()
+        opt.step()
+        opt.zero_grad(True)
+        return loss
+
+    compiled = easydist_compile(step, parallel_mode=mode, cuda_graph=False)
+    torch.manual_seed(3)
+    for i in range(3):
+        idx = torch.randint(0, 128, (4, 16))
+        tg = torch.randint(0, 128, (4, 16))
+        dist.broadcast(idx, src=0)
+        dist.broadcast(tg, src=0)
+        loss = compiled(model, opt, idx, tg)
+        ref = step(model_ref, opt_ref, idx, tg)
+        assert abs(float(loss) - float(ref)) < 5e-4, \
+            (i, float(loss), float(ref))
+
+
+@pytest.mark.world2
+@pytest.mark.parametrize("mode", ["zero2", "zero3"])
+def test_gpt_zero_ws2(mode):
+    spawn(_gpt_zero_body, args=(2, mode), world_size=2,
+          port=29547 + (mode == "zero3"))

@@ -153,3 +153,12 @@ def test_pp_state_dict_roundtrip():
     rt.load_state_dict(ckpt)
     replay = [float(compiled(model, opt, x, y)) for x, y in batches[2:]]
     assert later == replay, (later, replay)
+
+
+def test_pp_nchunks_edge_cases():
+    """nchunks == 1 (no pipelining) and nchunks > nstages."""
+    init_single_process()
+    for nchunks in (1, 8):
+        _golden_loop(MLP4,
+                     lambda m: torch.optim.Adam(m.parameters(), lr=1e-2),
+                     dict(split_points={"fc2"}, nchunks=nchunks), steps=2)
