@@ -41,9 +41,12 @@ def _to_real(meta: torch.Tensor, device=None, promote_fp64=True) -> torch.Tensor
         return t.to(dt)
     if meta.dtype == torch.bool:
         return torch.rand(meta.shape, device=device) > 0.5
-    # integer tensors: small non-negative values (safe for index ops)
-    hi = 2
-    return torch.randint(0, hi, meta.shape, dtype=meta.dtype, device=device)
+    # integer tensors: ZEROS. Index 0 is in-bounds for every non-empty
+    # dim even after the probe shards a tensor down (a size-2 dim sharded
+    # 2-ways leaves size-1 shards where index 1 hardware-faults the
+    # aten index kernels — observed as HSA_STATUS_ERROR_EXCEPTION on the
+    # MoE routing graph).
+    return torch.zeros(meta.shape, dtype=meta.dtype, device=device)
 
 
 def _sig_of(node: fx.Node):
