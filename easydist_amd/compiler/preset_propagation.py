@@ -191,3 +191,62 @@ def _register_pp_markers():
 
 
 _register_pp_markers()
+
+
+# ---------------------------------------------------------- matmul family ----
+# Analytic rules: execution-based discovery on the matmul family costs the
+# bulk of compile time (fp64 probes of [*,14336,4096] bmms on MoE graphs);
+# the algebra is closed-form. reference kept these in the DFS; we preset.
+def _reduce_add():
+    import operator as _operator
+    return functools.partial(CombinationFunc.reduce, ops=_operator.add)
+
+
+@register_preset(aten.mm.default)
+def _mm_rule(input_shapes, args, kwargs):
+    (M, K), (K2, N) = input_shapes[0], input_shapes[1]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    ann[0][0] = ShardDim.get_shard_dim(1)          # M
+    combs[1] = _gather(0)
+    ann[1][1] = ShardDim.get_shard_dim(2)          # N
+    combs[2] = _gather(1)
+    ann[0][1] = ShardDim.get_shard_dim(3)          # K (both operands)
+    ann[1][0] = ShardDim.get_shard_dim(3)
+    combs[3] = _reduce_add()
+    return ann, combs
+
+
+@register_preset(aten.addmm.default)
+def _addmm_rule(input_shapes, args, kwargs):
+    # (bias, a[M,K], b[K,N])
+    bias = input_shapes[0]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    ann[1][0] = ShardDim.get_shard_dim(1)          # M
+    if len(bias) == 2 and bias[0] == input_shapes[1][0]:
+        ann[0][0] = ShardDim.get_shard_dim(1)
+    combs[1] = _gather(0)
+    ann[2][1] = ShardDim.get_shard_dim(2)          # N
+    if len(bias) >= 1 and bias[-1] == input_shapes[2][1]:
+        ann[0][len(bias) - 1] = ShardDim.get_shard_dim(2)
+    combs[2] = _gather(1)
+    # K-shard omitted: PARTIAL would double-count the bias term
+    return ann, combs
+
+
+@register_preset(aten.bmm.default)
+def _bmm_rule(input_shapes, args, kwargs):
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    ann[0][0] = ShardDim.get_shard_dim(1)          # batch
+    ann[1][0] = ShardDim.get_shard_dim(1)
+    combs[1] = _gather(0)
+    ann[0][1] = ShardDim.get_shard_dim(2)          # M
+    combs[2] = _gather(1)
+    ann[1][2] = ShardDim.get_shard_dim(3)          # N
+    combs[3] = _gather(2)
+    ann[0][2] = ShardDim.get_shard_dim(4)          # K
+    ann[1][1] = ShardDim.get_shard_dim(4)
+    combs[4] = _reduce_add()
+    return ann, combs
