@@ -169,7 +169,8 @@ def shard_graph(gm, mesh, io_map, ret_names, device, fix_rets=True):
         logger.info("strategy cache hit: %s", cache_path)
     else:
         t0 = time.time()
-        sharding_info = EDTorchShardingAnn(gm, device=device).run()
+        sharding_info = EDTorchShardingAnn(
+            gm, device=mdconfig.discovery_device or device).run()
         search_time = time.time() - t0
         logger.info("sharding discovery: %d annotated ops (%.2fs)",
                     len(sharding_info), search_time)

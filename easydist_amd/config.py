@@ -126,3 +126,10 @@ def get_solver_time_limit() -> float:
     if max_seconds_same_incumbent != float("inf"):
         limit = min(limit, max_seconds_same_incumbent)
     return limit
+
+# where ShardCombine probes execute. Discovery is SEMANTIC (which
+# recombination reproduces the global output) — results are device
+# independent. CPU avoids thousands of tiny launch+sync round trips
+# (GPU discovery on the MoE graph is minutes; CPU is seconds) and can
+# never hardware-fault the GPU with degenerate probe shapes.
+discovery_device = os.environ.get("EASYDIST_DISCOVERY_DEVICE", "cpu")
