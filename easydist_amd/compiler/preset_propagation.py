@@ -250,3 +250,63 @@ def _bmm_rule(input_shapes, args, kwargs):
     ann[1][1] = ShardDim.get_shard_dim(4)
     combs[4] = _reduce_add()
     return ann, combs
+
+
+# ------------------------------------------------------------- elementwise ---
+# Same-shape elementwise ops: every dim shards with gather recombination.
+# Probing these by execution is pure waste (they dominate op counts in
+# decomposed optimizer + MoE graphs).
+_EW_UNARY = [
+    aten.neg.default, aten.sqrt.default, aten.rsqrt.default,
+    aten.exp.default, aten.log.default, aten.relu.default,
+    aten.gelu.default, aten.silu.default, aten.tanh.default,
+    aten.sigmoid.default, aten.reciprocal.default, aten.abs.default,
+    aten.sin.default, aten.cos.default, aten.erf.default,
+    aten.logical_not.default, aten.sgn.default, aten.sign.default,
+    aten.floor.default, aten.ceil.default, aten.round.default,
+    aten.pow.Tensor_Scalar, aten.clamp.default, aten.clamp_min.default,
+    aten.clamp_max.default, aten.leaky_relu.default, aten.elu.default,
+    aten.hardtanh.default, aten.gelu_backward.default,
+    aten.threshold_backward.default, aten.tanh_backward.default,
+    aten.sigmoid_backward.default, aten.silu_backward.default,
+    aten.leaky_relu_backward.default, aten.elu_backward.default,
+]
+_EW_BINARY = [
+    aten.add.Tensor, aten.sub.Tensor, aten.mul.Tensor, aten.div.Tensor,
+    aten.maximum.default, aten.minimum.default, aten.fmod.Tensor,
+    aten.remainder.Tensor, aten.atan2.default, aten.pow.Tensor_Tensor,
+    aten.add.Scalar, aten.sub.Scalar, aten.mul.Scalar, aten.div.Scalar,
+    aten.rsub.Scalar, aten.fmod.Scalar,
+    aten.eq.Tensor, aten.ne.Tensor, aten.lt.Tensor, aten.le.Tensor,
+    aten.gt.Tensor, aten.ge.Tensor, aten.eq.Scalar, aten.ne.Scalar,
+    aten.lt.Scalar, aten.le.Scalar, aten.gt.Scalar, aten.ge.Scalar,
+    aten.logical_and.default, aten.logical_or.default,
+    aten.bitwise_and.Tensor, aten.bitwise_or.Tensor,
+    aten.addcmul.default, aten.addcdiv.default, aten.lerp.Scalar,
+    aten.lerp.Tensor, aten.where.self,
+]
+
+
+@register_preset(*(_EW_UNARY + _EW_BINARY))
+def _elementwise_rule(input_shapes, args, kwargs):
+    if not input_shapes:
+        return None
+    shape0 = input_shapes[0]
+    # every TENSOR input must have the exact same shape (no broadcasting:
+    # broadcast patterns go to execution-based discovery)
+    for sh in input_shapes[1:]:
+        if tuple(sh) != tuple(shape0):
+            return None
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(shape0)):
+        if shape0[d] <= 1:
+            continue
+        for i in range(len(input_shapes)):
+            ann[i][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    if not combs:
+        return None
+    return ann, combs
