@@ -69,7 +69,7 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
             ret_names.add(o.name)
 
     gm, out_pl_env, search_time, solve_time = shard_graph(
-        gm, mesh, io_map, ret_names, device)
+        gm, mesh, io_map, ret_names, device, n_state=n_state)
 
     # ---- 5b. comm overlap: widen every start/wait window -----------------
     from .passes.comm_optimize import comm_optimize
@@ -155,7 +155,8 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     return compiled
 
 
-def shard_graph(gm, mesh, io_map, ret_names, device, fix_rets=True):
+def shard_graph(gm, mesh, io_map, ret_names, device, fix_rets=True,
+                n_state=None):
     """Discovery + per-mesh-dim MILP solve + sharding transform.
 
     Shards along the mesh's spmd dims only (a 'pp' dim is skipped), so
@@ -219,7 +220,8 @@ def shard_graph(gm, mesh, io_map, ret_names, device, fix_rets=True):
 
     gm, out_pl_env = sharding_transform(gm, strategies_per_dim, mesh.shape)
     gm = _fix_output_reshard(gm, out_pl_env, io_map,
-                             ret_names if fix_rets else set(), mesh)
+                             ret_names if fix_rets else set(), mesh,
+                             n_state=n_state)
     return gm, out_pl_env, search_time, solve_time
 
 
@@ -339,10 +341,14 @@ def _adam_positions(params, buffers, named_states):
     return out
 
 
-def _fix_output_reshard(gm, out_pl_env, io_map, ret_names, mesh):
+def _fix_output_reshard(gm, out_pl_env, io_map, ret_names, mesh,
+                        n_state=None):
     """Reshard outputs that must land at a fixed placement: user returns to
     REPLICATE; state outputs back to their input placeholder's placement
-    (reference behavior: sharding.py:920-949)."""
+    (reference behavior: sharding.py:920-949). State round-trips resolve
+    through io_map when the trace used copy_ AND positionally (output i of
+    the state block is placeholder i) for in-place-op traces whose io_map
+    is empty after fix_inplace."""
     from ..runtime import comm_runtime as crt
     from .passes.sharding import ShardingTransform
     graph = gm.graph
@@ -354,6 +360,7 @@ def _fix_output_reshard(gm, out_pl_env, io_map, ret_names, mesh):
             ph_pl[n.name] = out_pl_env[n.name][0]
     src_to_ph = {v: k for k, v in io_map.items() if v is not None}
 
+    placeholders = [n for n in graph.nodes if n.op == "placeholder"]
     tr = ShardingTransform(gm, [], mesh.shape)
     tr.out_pl = out_pl_env
     changed = False
@@ -367,6 +374,10 @@ def _fix_output_reshard(gm, out_pl_env, io_map, ret_names, mesh):
                 want = [R] * mesh.ndim
             elif o.name in src_to_ph and src_to_ph[o.name] in ph_pl:
                 want = ph_pl[src_to_ph[o.name]]
+            elif (n_state is not None and i < n_state
+                  and i < len(placeholders)
+                  and o.name != placeholders[i].name):
+                want = ph_pl.get(placeholders[i].name, [R] * mesh.ndim)
             if want is None:
                 continue
             if all(repr(c) == repr(w) for c, w in zip(cur, want)):

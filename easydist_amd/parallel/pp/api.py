@@ -74,7 +74,8 @@ def _compile_pp(func, tracing_mode, args, kwargs, module, opt,
         from ...compiler.passes.functionalize import canonicalize
         gm, io_map = canonicalize(gm)
         gm, out_pl_env, _st, _sv = shard_graph(gm, mesh, io_map, set(),
-                                               device, fix_rets=False)
+                                               device, fix_rets=False,
+                                               n_state=n_state)
 
     info = compile_pipeline(gm, flat_inputs, n_params, n_state,
                             list(params.keys()), io_map=io_map)
@@ -87,6 +88,11 @@ def _compile_pp(func, tracing_mode, args, kwargs, module, opt,
             pls = out_pl_env.get(name)
             if pls:
                 info.ph_placements[name] = pls[0]
+        import os as _os
+        if _os.environ.get("EASYDIST_DEBUG_PP"):
+            logger.error("ph_placements sample: %s",
+                         {k: repr(v) for k, v in
+                          list(info.ph_placements.items())[:6]})
         # ret + boundary placements for the runtime's spmd fixes
         for sg in info.stages:
             for name in sg.ret_names:

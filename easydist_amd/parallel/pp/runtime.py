@@ -72,6 +72,12 @@ class StageExecutor:
             # clone: the runtime owns its state (mutated in place by
             # step_gm); never alias the user's module tensors
             self.state_env[name] = t.clone()
+        import os as _os
+        if _os.environ.get("EASYDIST_DEBUG_PP"):
+            import torch as _t
+            logger.error("init_state stage %s: %s", self.sg.stage_idx,
+                         {k: tuple(v.shape) for k, v in
+                          list(self.state_env.items())[:6]})
 
     def reset_step(self, nchunks: int):
         self.stash = [dict() for _ in range(nchunks)]
@@ -90,7 +96,15 @@ class StageExecutor:
 
     def run_fw(self, m: int, data_chunks) -> None:
         args = [self._lookup(n, m, data_chunks) for n in self.sg.fw_inputs]
-        outs = self.sg.fw_gm(*args)
+        try:
+            outs = self.sg.fw_gm(*args)
+        except Exception:
+            import torch as _t
+            shapes = {n: (tuple(a.shape) if isinstance(a, _t.Tensor)
+                          else a) for n, a in zip(self.sg.fw_inputs, args)}
+            logger.error("stage %d fw failed; inputs: %s", self.sg.stage_idx,
+                         shapes)
+            raise
         for name, val in zip(self.sg.fw_outputs, outs):
             self.stash[m][name] = val
             if name in self.sg.ret_names:
