@@ -40,9 +40,17 @@ def _ln_fwd_cuda(x, w, b, eps):
 
 
 def _ln_bwd_aten(grad, x, mean, rstd, w, mask):
+    """Emulates the HIP kernel contract on aten: fp32 math, dx in x's
+    dtype, dw/db fp32 (the CPU aten kernel also rejects mixed
+    bf16-input/fp32-stat calls that CUDA accepts)."""
     d = x.shape[-1]
-    return torch.ops.aten.native_layer_norm_backward(
-        grad, x, [d], mean, rstd, w, None, list(mask))
+    bias = (torch.zeros(d, dtype=torch.float32, device=x.device)
+            if len(mask) > 2 and mask[2] else None)
+    wf = w.float() if w is not None else None
+    dx, dw, db = torch.ops.aten.native_layer_norm_backward(
+        grad.float(), x.float(), [d], mean.float(), rstd.float(), wf, bias,
+        list(mask))
+    return (dx.to(x.dtype) if dx is not None else None, dw, db)
 
 
 def _ln_bwd_cuda(grad, x, mean, rstd, w, mask):
