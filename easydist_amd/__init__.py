@@ -14,7 +14,7 @@ from .compiler.api import easydist_compile, register_parallel_method
 from .parallel.device_mesh import (NDDeviceMesh, get_device_mesh,
                                    set_device_mesh)
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"   # salts the strategy cache
 
 logger = logging.getLogger("easydist_amd")
 

@@ -62,3 +62,30 @@ def test_gpt_ws1():
 @pytest.mark.world2
 def test_gpt_ws2():
     spawn(_run_gpt_golden, args=(2,), world_size=2, port=29537)
+
+
+def test_gpt_autocast_cpu_compile():
+    """The bench train step (autocast bf16 + fused CE + lowered norms)
+    through the auto pipeline on CPU — locks in the aten fallbacks of
+    every lowered kernel."""
+    import torch
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+    from easydist_amd.models import gpt as gptm
+    from easydist_amd.utils.testing import init_single_process
+
+    init_single_process()
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([0], ["spmd0"])
+    torch.manual_seed(0)
+    cfg = gptm.GPTConfig(vocab_size=128, n_layer=2, n_head=2, n_embd=32,
+                         block_size=16)
+    model = gptm.GPT(cfg)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+    compiled = easydist_compile(gptm.gpt_train_step, cuda_graph=False)
+    idx = torch.randint(0, 128, (4, 16))
+    tg = torch.randint(0, 128, (4, 16))
+    losses = [float(compiled(model, opt, idx, tg)) for _ in range(4)]
+    assert all(map(lambda x: x == x, losses)), losses   # no NaNs
+    assert losses[-1] < losses[0], losses
