@@ -147,6 +147,12 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     compiled.init_named_states = named_states
     compiled.ret_out_positions = list(range(n_state_outs + n_grads,
                                             len(flat_outs)))
+    # qualified names for checkpoint APIs: position -> user-facing name
+    qualnames = (list(params.keys()) + list(buffers.keys())
+                 + [f"{pn}.{k}" for pn, st in named_states.items()
+                    for k, v in st.items()
+                    if isinstance(v, torch.Tensor)])
+    compiled.state_qualnames = {i: qn for i, qn in enumerate(qualnames)}
     compiled.meta = {
         "search_time": search_time, "solve_time": solve_time,
         "n_nodes": len(gm.graph.nodes), "out_spec": gm._out_spec

@@ -227,6 +227,34 @@ class EDCompiledFunc:
     def local_state(self):
         return dict(self.state)
 
+    # ------------------------------------------------- checkpointing -------
+    def state_dict(self) -> Dict[str, torch.Tensor]:
+        """FULL training state under user-facing qualified names
+        (params, buffers, '<param>.<opt_state>'), gathered across the
+        mesh (reference state access: compile_auto.py:778-815)."""
+        mesh = get_device_mesh()
+        names = getattr(self, "state_qualnames", {})
+        out = {}
+        for pos, t in self.state.items():
+            pl = self.input_placements[pos] or []
+            full = unshard_tensor(t, pl, mesh)
+            out[names.get(pos, str(pos))] = full.detach().clone()
+        return out
+
+    def load_state_dict(self, sd: Dict[str, torch.Tensor]):
+        """Load full values; every rank re-shards its local slice."""
+        mesh = get_device_mesh()
+        names = getattr(self, "state_qualnames", {})
+        for pos in list(self.state):
+            qn = names.get(pos, str(pos))
+            if qn not in sd:
+                continue
+            t = sd[qn].detach().to(self.device)
+            pl = self.input_placements[pos] or []
+            self.state[pos] = shard_tensor_local(t, pl, mesh).clone()
+        # invalidate captured graphs: buffer addresses changed
+        self._graph = None
+
     def run_graph(self, flat_inputs: List):
         """hipGraph capture + replay (torch.cuda.CUDAGraph is hipGraph on
         ROCm). Non-state inputs are copied into static buffers each step."""

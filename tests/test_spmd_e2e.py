@@ -82,3 +82,29 @@ def test_auto_spmd_ws1():
 @pytest.mark.world2
 def test_auto_spmd_ws2():
     spawn(_run_golden, args=(2,), world_size=2, port=29532)
+
+
+def test_auto_state_dict_roundtrip():
+    """Checkpoint-and-replay: losses after reload must match exactly."""
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+
+    init_single_process()
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([0], ["spmd0"])
+    torch.manual_seed(0)
+    model = MLP()
+    opt = torch.optim.Adam(model.parameters(), lr=1e-2, fused=True)
+    compiled = easydist_compile(train_step, cuda_graph=False)
+    torch.manual_seed(3)
+    batches = [(torch.randn(8, 16), torch.randn(8, 16)) for _ in range(4)]
+    for x, y in batches[:2]:
+        compiled(model, opt, x, y)
+    rt = list(compiled.compiled.values())[0]
+    ckpt = rt.state_dict()
+    assert any(k == "fc1.weight" for k in ckpt), list(ckpt)[:4]
+    assert any(k.endswith(".exp_avg") for k in ckpt), list(ckpt)[:8]
+    later = [float(compiled(model, opt, x, y)) for x, y in batches[2:]]
+    rt.load_state_dict(ckpt)
+    replay = [float(compiled(model, opt, x, y)) for x, y in batches[2:]]
+    assert later == replay, (later, replay)
