@@ -244,6 +244,9 @@ def _strategy_cache_path(gm, mesh) -> Optional[str]:
     import easydist_amd
     h.update(easydist_amd.__version__.encode())
     h.update(torch.__version__.encode())
+    # rule changes must invalidate cached strategies
+    from .preset_propagation import _PRESET_REGISTRY
+    h.update(repr(sorted(str(k) for k in _PRESET_REGISTRY)).encode())
     d = os.path.join(os.path.expanduser("~"), ".easydist_amd",
                      "compile_cache")
     os.makedirs(d, exist_ok=True)

@@ -310,3 +310,28 @@ def _elementwise_rule(input_shapes, args, kwargs):
     if not combs:
         return None
     return ann, combs
+
+
+# -------------------------------------------------- value-dependent ops ------
+# Ops whose semantics depend on tensor VALUES (indices/orderings): probe
+# executions use zero-filled integer tensors (OOB-safety), which makes
+# sharded-recombination checks pass COINCIDENTALLY and mints false rules.
+# Force replicate: (None, {}) = "no sharding rule".
+_VALUE_DEPENDENT = [
+    aten.index.Tensor, aten.index_put.default, aten.index_add.default,
+    aten.index_select.default, aten.index_copy.default,
+    aten.scatter.src, aten.scatter.value, aten.scatter_add.default,
+    aten.scatter_reduce.two, aten.gather.default,
+    aten.argsort.default, aten.sort.default, aten.topk.default,
+    aten.searchsorted.Tensor, aten.searchsorted.Scalar,
+    aten.unique_consecutive.default, aten.nonzero.default,
+    aten.masked_select.default, aten.take.default, aten.bucketize.Tensor,
+    aten.argmax.default, aten.argmin.default, aten.bincount.default,
+    aten.mode.default, aten.kthvalue.default, aten.median.default,
+    aten.index_put_.default,
+]
+
+
+@register_preset(*_VALUE_DEPENDENT)
+def _value_dependent_rule(input_shapes, args, kwargs):
+    return (None, {})

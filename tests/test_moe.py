@@ -117,3 +117,56 @@ def _ep_grad_body(world_size):
 @pytest.mark.world2
 def test_moe_ep_grads_ws2():
     spawn(_ep_grad_body, args=(2,), world_size=2, port=29562)
+
+
+def _moe_auto_body(world_size):
+    """Full auto-SPMD compile of the MoE model, golden vs vanilla."""
+    import copy
+
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+    from easydist_amd.models.moe import MIXTRAL_SMALL, MoEGPT
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+    torch.manual_seed(0)
+    cfg = MIXTRAL_SMALL
+    model = MoEGPT(cfg)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+    opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-3,
+                               fused=True)
+
+    def step(model, opt, idx, tg):
+        loss = model.loss(idx, tg)
+        loss.backward()
+        opt.step()
+        opt.zero_grad(True)
+        return loss
+
+    compiled = easydist_compile(step, cuda_graph=False)
+    torch.manual_seed(5)
+    for i in range(2):
+        idx = torch.randint(0, cfg.vocab_size, (4, cfg.block_size))
+        tg = torch.randint(0, cfg.vocab_size, (4, cfg.block_size))
+        dist.broadcast(idx, src=0)
+        dist.broadcast(tg, src=0)
+        loss = compiled(model, opt, idx, tg)
+        ref = step(model_ref, opt_ref, idx, tg)
+        assert abs(float(loss) - float(ref)) < 5e-3, \
+            (i, float(loss), float(ref))
+
+
+def test_moe_auto_ws1():
+    from easydist_amd.utils.testing import init_single_process
+    init_single_process()
+    _moe_auto_body(1)
+
+
+@pytest.mark.world2
+def test_moe_auto_ws2():
+    spawn(_moe_auto_body, args=(2,), world_size=2, port=29565)
