@@ -60,7 +60,10 @@ class AutoFlowSolver1D:
             in_vars = set()
             for st in cc.strategies:
                 in_vars.update(st.in_placements.keys())
-            for v in in_vars:
+            # sorted: set iteration is PYTHONHASHSEED-ordered — edge order
+            # must be identical across processes/runs or MILP tie-breaking
+            # diverges
+            for v in sorted(in_vars):
                 cp = self.var_producer_cluster.get(v)
                 if cp is None or cp is cc:
                     continue

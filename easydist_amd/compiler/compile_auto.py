@@ -88,7 +88,8 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
         if ppos:
             nfused = fuse_optimizer(gm, flat_outs_f,
                                     [n for n in gm.graph.nodes
-                                     if n.op == "placeholder"], ppos, opt)
+                                     if n.op == "placeholder"], ppos, opt,
+                                    pl_env=out_pl_env)
             if nfused:
                 out_node_f.args = (pytree.tree_unflatten(flat_outs_f,
                                                          spec_f),)

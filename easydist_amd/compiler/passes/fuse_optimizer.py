@@ -79,7 +79,7 @@ def _match_param_chain(p_ph, ea_ph, eas_ph, step_ph, p_new, ea_new,
 
 def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
                    param_positions: Dict[int, Dict[str, int]],
-                   opt) -> int:
+                   opt, pl_env=None) -> int:
     """Fuse matching per-param Adam chains. Returns #fused params.
 
     param_positions: param flat-input position -> {'step': out_pos,
@@ -101,6 +101,21 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
     if wd != 0.0:
         return 0   # chain shape differs; fuse later
 
+    def _all_replicate(*nodes):
+        """The fused call wires these nodes DIRECTLY, bypassing any
+        reshard the transform placed between them and the decomposed
+        chain — only safe when every placement is replicate."""
+        if not pl_env:
+            return True
+        for n in nodes:
+            pls = pl_env.get(n.name)
+            if not pls:
+                continue
+            for pl in pls[0]:
+                if not pl.is_replicate():
+                    return False
+        return True
+
     matched = []
     for p_pos, outs in param_positions.items():
         p_ph = placeholders[p_pos]
@@ -114,6 +129,8 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
         g = _match_param_chain(p_ph, ea_ph, eas_ph, step_ph, p_new, ea_new,
                                eas_new, step_new)
         if g is None:
+            continue
+        if not _all_replicate(p_ph, ea_ph, eas_ph, g):
             continue
         matched.append((p_pos, outs, p_ph, g, ea_ph, eas_ph, step_ph))
     if not matched:
