@@ -37,16 +37,23 @@ GPT_BENCH_1L = GPTConfig(n_layer=1, n_head=48, n_embd=12288, block_size=1024)
 
 
 class CausalSelfAttention(nn.Module):
+    """Sequence-parallel ready: with ``sp_group`` set (see
+    GPT.enable_sequence_parallel) each rank holds a sequence shard and
+    attention runs as exact ring attention over the xGMI neighbors
+    (ops/ring_attention.py); otherwise the gfx950 flash kernel handles
+    the full sequence locally."""
+
     def __init__(self, cfg: GPTConfig):
         super().__init__()
         assert cfg.n_embd % cfg.n_head == 0
         self.n_head = cfg.n_head
         self.n_embd = cfg.n_embd
+        self.sp_group = None
         self.c_attn = nn.Linear(cfg.n_embd, 3 * cfg.n_embd, bias=cfg.bias)
         self.c_proj = nn.Linear(cfg.n_embd, cfg.n_embd, bias=cfg.bias)
 
     def forward(self, x):
-        from ..ops import attention
+        from ..ops import attention, ring_attention
         B, T, C = x.shape
         qkv = self.c_attn(x)
         q, k, v = qkv.split(self.n_embd, dim=2)
@@ -54,7 +61,11 @@ class CausalSelfAttention(nn.Module):
         q = q.view(B, T, self.n_head, hd).transpose(1, 2)
         k = k.view(B, T, self.n_head, hd).transpose(1, 2)
         v = v.view(B, T, self.n_head, hd).transpose(1, 2)
-        y = attention.scaled_dot_product_attention(q, k, v, causal=True)
+        if self.sp_group is not None:
+            y = ring_attention.ring_attention(q, k, v, group=self.sp_group,
+                                              causal=True)
+        else:
+            y = attention.scaled_dot_product_attention(q, k, v, causal=True)
         y = y.transpose(1, 2).reshape(B, T, C)
         return self.c_proj(y)
 
@@ -108,9 +119,22 @@ class GPT(nn.Module):
         elif isinstance(m, nn.Embedding):
             nn.init.normal_(m.weight, std=0.02)
 
+    def enable_sequence_parallel(self, group):
+        """Sequence-parallel mode: every rank feeds its OWN sequence
+        shard (token slice) of the batch; attention rings over the
+        group; positions offset by the rank's slice start."""
+        import torch.distributed as dist
+        self._sp_group = group
+        self._sp_rank = dist.get_rank(group)
+        for blk in self.h:
+            blk.attn.sp_group = group
+        return self
+
     def forward(self, idx):
         B, T = idx.shape
-        x = self.wte(idx) + self.wpe(self.pos[:T])
+        off = getattr(self, "_sp_rank", 0) * T \
+            if getattr(self, "_sp_group", None) is not None else 0
+        x = self.wte(idx) + self.wpe(self.pos[off:off + T])
         if torch.is_autocast_enabled(x.device.type):
             # embeddings are not on autocast's cast list: without this the
             # whole residual stream runs fp32 (fp32 norms/adds + a cast
