@@ -43,9 +43,28 @@ def _tensor_scalar(n) -> Tuple[Optional[fx.Node], Optional[float]]:
     return None, None
 
 
+def _strip_l2(g, p_ph, wd):
+    """Coupled (Adam) weight decay traces as g' = add(raw, mul(p, wd))
+    (decomp.py:_single_adam). Return the RAW grad node — the kernel
+    re-applies ``grad += wd*p`` itself — or None if the shape differs."""
+    if not (isinstance(g, fx.Node) and g.target is aten.add.Tensor):
+        return None
+    for a, b in (g.args, g.args[::-1]):
+        t, s = _tensor_scalar(b)
+        if t is p_ph and s is not None and abs(s - wd) < 1e-12 \
+                and isinstance(a, fx.Node):
+            return a
+    return None
+
+
 def _match_param_chain(p_ph, ea_ph, eas_ph, step_ph, p_new, ea_new,
-                       eas_new, step_new):
-    """Verify the decomposed no-amsgrad Adam shape; return the grad node."""
+                       eas_new, step_new, wd=0.0, decay_scale=None):
+    """Verify the decomposed Adam/AdamW shape; return the grad node.
+
+    ``wd``: coupled L2 decay (Adam) — the grad feeding the moments is
+    add(raw_g, wd*p); we unwrap it. ``decay_scale``: decoupled decay
+    (AdamW) — p_new subtracts from mul(p_ph, 1-lr*wd) instead of p_ph.
+    """
     # step_new = add(step_ph, 1)
     if not (isinstance(step_new, fx.Node)
             and step_new.target in (aten.add.Tensor, aten.add.Scalar)
@@ -65,14 +84,26 @@ def _match_param_chain(p_ph, ea_ph, eas_ph, step_ph, p_new, ea_new,
         g = t0
     else:
         return None
+    if wd != 0.0:
+        g = _strip_l2(g, p_ph, wd)
+        if g is None:
+            return None
     # eas_new = add(mul(eas_ph, b2), mul(mul(g, 1-b2), g))
     if not (isinstance(eas_new, fx.Node)
             and eas_new.target is aten.add.Tensor):
         return None
-    # p_new = sub(p_ph, ...)
+    # p_new = sub(base, ...); base = p_ph (Adam) or mul(p_ph, 1-lr*wd)
+    # (AdamW decomp always emits the mul, even at wd=0 -> scalar 1.0)
     if not (isinstance(p_new, fx.Node)
-            and p_new.target is aten.sub.Tensor
-            and p_new.args[0] is p_ph):
+            and p_new.target is aten.sub.Tensor):
+        return None
+    base = p_new.args[0]
+    if decay_scale is not None:
+        t, s = _tensor_scalar(base)
+        if not (t is p_ph and s is not None
+                and abs(s - decay_scale) < 1e-12):
+            return None
+    elif base is not p_ph:
         return None
     return g
 
@@ -89,17 +120,21 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
     """
     if opt is None or type(opt).__name__ not in ("Adam", "AdamW"):
         return 0
-    if type(opt).__name__ == "AdamW":
-        return 0   # different decomposition; fuse later
+    decoupled = type(opt).__name__ == "AdamW"
     groups = opt.param_groups
-    if len(groups) != 1 or groups[0].get("amsgrad"):
+    if len(groups) != 1 or groups[0].get("amsgrad") \
+            or groups[0].get("maximize"):
         return 0
     lr = float(groups[0]["lr"])
     beta1, beta2 = map(float, groups[0]["betas"])
     eps = float(groups[0]["eps"])
     wd = float(groups[0].get("weight_decay", 0.0))
-    if wd != 0.0:
-        return 0   # chain shape differs; fuse later
+    # decoupled (AdamW): p is pre-scaled by (1-lr*wd) before the Adam
+    # update; we emit ONE _foreach_mul for the whole bank and hand the
+    # kernel wd=0. Coupled (Adam): the kernel applies grad += wd*p itself.
+    decay_scale = (1.0 - lr * wd) if decoupled else None
+    kernel_wd = 0.0 if decoupled else wd
+    chain_wd = 0.0 if decoupled else wd
 
     def _all_replicate(*nodes):
         """The fused call wires these nodes DIRECTLY, bypassing any
@@ -127,7 +162,8 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
         eas_new = flat_outs[outs["exp_avg_sq"]]
         step_new = flat_outs[outs["step"]]
         g = _match_param_chain(p_ph, ea_ph, eas_ph, step_ph, p_new, ea_new,
-                               eas_new, step_new)
+                               eas_new, step_new, wd=chain_wd,
+                               decay_scale=decay_scale)
         if g is None:
             continue
         if not _all_replicate(p_ph, ea_ph, eas_ph, g):
@@ -139,14 +175,21 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
     graph = gm.graph
     out_node = next(n for n in graph.nodes if n.op == "output")
     with graph.inserting_before(out_node):
+        p_list = [m[2] for m in matched]
+        if decay_scale is not None:
+            # decoupled decay for the whole bank in one multi-tensor op
+            scaled = graph.call_function(
+                aten._foreach_mul.Scalar, (p_list, decay_scale))
+            p_list = [graph.call_function(operator.getitem, (scaled, i))
+                      for i in range(len(matched))]
         fused = graph.call_function(
             torch.ops.easydist_amd.fused_adam_step.default,
-            ([m[2] for m in matched],            # params
+            (p_list,                             # params (maybe pre-decayed)
              [m[3] for m in matched],            # grads
              [m[4] for m in matched],            # exp_avgs
              [m[5] for m in matched],            # exp_avg_sqs
              [m[6] for m in matched],            # steps (pre-increment)
-             lr, beta1, beta2, wd, eps))
+             lr, beta1, beta2, kernel_wd, eps))
         lists = [graph.call_function(operator.getitem, (fused, k))
                  for k in range(4)]
         for i, (p_pos, outs, *_rest) in enumerate(matched):

@@ -49,22 +49,48 @@ def warmup_optimizer(module: torch.nn.Module, opt) -> dict:
     if opt is None:
         return named_states
     params = dict(module.named_parameters())
-    with torch.no_grad():
-        snapshot = {n: p.detach().clone() for n, p in params.items()}
-        for p in params.values():
-            if p.grad is None:
-                p.grad = torch.zeros_like(p)
-    opt.step()
-    opt.zero_grad(True)
-    with torch.no_grad():
-        # the warm-up step is fake: undo any param drift (weight decay etc.)
-        for n, p in params.items():
-            p.copy_(snapshot[n])
+    fresh = {n for n, p in params.items() if not opt.state.get(p)}
+    if fresh:
+        with torch.no_grad():
+            psnap = {n: p.detach().clone() for n, p in params.items()}
+            # pre-existing states (resumed optimizer) must survive the
+            # fake step untouched
+            ssnap = {n: {k: (v.detach().clone() if torch.is_tensor(v)
+                             else v)
+                         for k, v in opt.state[p].items()}
+                     for n, p in params.items() if n not in fresh}
+            for p in params.values():
+                if p.grad is None:
+                    p.grad = torch.zeros_like(p)
+        opt.step()
+        opt.zero_grad(True)
+        with torch.no_grad():
+            # the warm-up step is fake: undo any param drift (weight
+            # decay etc.)
+            for n, p in params.items():
+                p.copy_(psnap[n])
+            for n, p in params.items():
+                st = opt.state.get(p)
+                if not st:
+                    continue
+                if n in ssnap:
+                    for k, v in ssnap[n].items():
+                        if torch.is_tensor(v) and torch.is_tensor(st.get(k)):
+                            st[k].copy_(v)
+                        else:
+                            st[k] = v
+                else:
+                    # freshly materialized: the fake step left wd/momentum
+                    # residue in the moment buffers (g was 0 but wd*p was
+                    # not) — reset to true step-0 state
+                    for k, v in list(st.items()):
+                        if k == "step":
+                            st[k] = v - 1
+                        elif torch.is_tensor(v):
+                            v.zero_()
     for n, p in params.items():
         if p in opt.state:
             named_states[n] = dict(opt.state[p])
-            if "step" in named_states[n]:
-                named_states[n]["step"] = named_states[n]["step"] - 1
     return named_states
 
 

@@ -1,0 +1,81 @@
+"""Fused optimizer pass coverage: Adam (coupled L2 decay) and AdamW
+(decoupled decay) chains must fuse into easydist_amd::fused_adam_step and
+train bit-comparably with the vanilla eager optimizer (CPU, ws1)."""
+import copy
+import logging
+
+import pytest
+import torch
+import torch.nn as nn
+
+from easydist_amd import easydist_compile, easydist_setup, set_device_mesh
+from easydist_amd.utils.testing import init_single_process
+
+
+class _Net(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.a = nn.Linear(16, 32)
+        self.b = nn.Linear(32, 16)
+
+    def forward(self, x):
+        return self.b(torch.relu(self.a(x)))
+
+
+def _step(model, opt, x, y):
+    loss = torch.nn.functional.mse_loss(model(x), y)
+    loss.backward()
+    opt.step()
+    opt.zero_grad(True)
+    return loss
+
+
+def _run(make_opt, caplog):
+    init_single_process()
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([0], ["spmd0"])
+    torch.manual_seed(0)
+    model = _Net()
+    model_ref = copy.deepcopy(model)
+    opt = make_opt(model.parameters())
+    opt_ref = make_opt(model_ref.parameters())
+
+    compiled = easydist_compile(_step, cuda_graph=False)
+    torch.manual_seed(7)
+    with caplog.at_level(
+            logging.INFO,
+            logger="easydist_amd.compiler.passes.fuse_optimizer"):
+        for i in range(3):
+            x = torch.randn(8, 16)
+            y = torch.randn(8, 16)
+            loss = compiled(model, opt, x, y)
+            ref = _step(model_ref, opt_ref, x, y)
+            assert abs(float(loss) - float(ref)) < 1e-5, \
+                (i, float(loss), float(ref))
+    assert any("fused 4/4" in r.message for r in caplog.records), \
+        [r.message for r in caplog.records]
+    # end-state parity: fused kernel math == eager torch optimizer math
+    # (module params are stale by design — the runtime owns the live
+    # training state; compare via the compiled state API)
+    live = dict(compiled.named_parameters())
+    for n, pr in model_ref.named_parameters():
+        p = live[n]
+        assert torch.allclose(p, pr, rtol=1e-5, atol=1e-6), \
+            (n, (p - pr).abs().max())
+
+
+def test_fused_adam_l2_decay(caplog):
+    _run(lambda ps: torch.optim.Adam(ps, lr=1e-2, weight_decay=0.01,
+                                     fused=True), caplog)
+
+
+def test_fused_adamw(caplog):
+    _run(lambda ps: torch.optim.AdamW(ps, lr=1e-2, weight_decay=0.05,
+                                      fused=True), caplog)
+
+
+def test_fused_adamw_zero_wd(caplog):
+    # AdamW decomp emits mul(p, 1.0) even at wd=0 — the matcher must
+    # still recognize the chain
+    _run(lambda ps: torch.optim.AdamW(ps, lr=1e-2, weight_decay=0.0,
+                                      fused=True), caplog)
