@@ -138,6 +138,10 @@ def easydist_compile(func=None, parallel_mode="auto", tracing_mode="fake",
                      cuda_graph=True, use_hint=False,
                      max_solver_time=float("inf"), compile_only=False,
                      memory_opt=False, **compile_kwargs):
+    # reference semantics (sharding_interpreter.py:270): use_hint shrinks
+    # giant-op probes during discovery; our discovery ALWAYS probes at
+    # discovery_max_dim-shrunk shapes (sharding_interpreter._shrink_map),
+    # so the flag is accepted for API parity and is a no-op
     mdconfig.use_hint = use_hint
     mdconfig.max_seconds_same_incumbent = max_solver_time
 

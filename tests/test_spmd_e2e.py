@@ -108,3 +108,45 @@ def test_auto_state_dict_roundtrip():
     rt.load_state_dict(ckpt)
     replay = [float(compiled(model, opt, x, y)) for x, y in batches[2:]]
     assert later == replay, (later, replay)
+
+
+def _run_golden_2d(world_size):
+    """2-D SPMD mesh (2x2): the per-mesh-dim solver composes placements
+    across both dims; golden vs vanilla."""
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([[0, 1], [2, 3]], ["spmd0", "spmd1"])
+
+    torch.manual_seed(42)
+    model = MLP()
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-2, fused=True)
+    opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-2, fused=True)
+    compiled = easydist_compile(train_step, parallel_mode="auto",
+                                cuda_graph=False)
+    torch.manual_seed(7)
+    for step in range(3):
+        x = torch.randn(8, 16)
+        y = torch.randn(8, 16)
+        dist.broadcast(x, src=0)
+        dist.broadcast(y, src=0)
+        loss = compiled(model, opt, x, y)
+        ref_loss = train_step(model_ref, opt_ref, x, y)
+        assert abs(float(loss) - float(ref_loss)) < 1e-4, \
+            (step, float(loss), float(ref_loss))
+    final = compiled.named_parameters()
+    for n, p_ref in model_ref.named_parameters():
+        got = final[n]
+        assert torch.allclose(got, p_ref.detach(), rtol=1e-4, atol=1e-5), \
+            (n, (got - p_ref.detach()).abs().max())
+
+
+@pytest.mark.world4
+def test_auto_spmd_2d_mesh_ws4():
+    spawn(_run_golden_2d, args=(4,), world_size=4, port=29533)
