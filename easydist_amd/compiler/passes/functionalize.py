@@ -93,10 +93,29 @@ def fix_inplace(gm: fx.GraphModule) -> fx.GraphModule:
     return gm
 
 
+def fix_overload_binding(gm: fx.GraphModule) -> fx.GraphModule:
+    """Repair traced nodes whose args don't bind to their overload.
+
+    make_fx records dropout's mask draw as ``bernoulli.default(x, p)``
+    (two positionals), but that overload's schema only takes ``self`` —
+    executing the graph raises. Rebind to the ``.p`` overload."""
+    import torch
+    for n in gm.graph.nodes:
+        if n.op == "call_function" \
+                and n.target is torch.ops.aten.bernoulli.default \
+                and len(n.args) == 2:
+            n.target = torch.ops.aten.bernoulli.p
+    return gm
+
+
 def canonicalize(gm: fx.GraphModule):
     """Run the full pre-sharding pass stack. Returns (gm, state_io_map)."""
     eliminate_detach(gm)
     fix_inplace(gm)
+    # AFTER fix_inplace: bernoulli_.float functionalizes to
+    # bernoulli.default (the packet has no .float overload), which then
+    # needs its p argument rebound to the .p overload
+    fix_overload_binding(gm)
     io_map = defunctionalize_copies(gm)
     gm.graph.eliminate_dead_code()
     gm.recompile()

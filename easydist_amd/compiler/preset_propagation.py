@@ -165,6 +165,112 @@ def _embedding_bwd_rule(input_shapes, args, kwargs):
     return ann, combs
 
 
+# ------------------------------------------------------- slice family -------
+# slice/narrow/select keep a fixed window of ONE dim: every OTHER dim
+# shards freely; the windowed dim must stay whole (a shard's local window
+# reads the wrong global elements — discovered the hard way when all-zero
+# integer probes made a dim-sharded slice of position_ids verify as
+# identity).
+@register_preset(aten.slice.Tensor)
+def _slice_rule(input_shapes, args, kwargs):
+    shape = input_shapes[0]
+    sdim = args[1] if len(args) > 1 else 0
+    sdim = sdim % len(shape) if shape else 0
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(shape)):
+        if d == sdim or shape[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.narrow.default)
+def _narrow_rule(input_shapes, args, kwargs):
+    return _slice_rule(input_shapes, args, kwargs)
+
+
+@register_preset(aten.select.int)
+def _select_rule(input_shapes, args, kwargs):
+    shape = input_shapes[0]
+    sdim = args[1] % len(shape) if shape else 0
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(shape)):
+        if d == sdim or shape[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d - (1 if d > sdim else 0))
+        sid += 1
+    return ann, combs
+
+
+# ------------------------------------------------------------- attention ----
+# Fused SDPA kernels must NEVER be probed by execution: the CPU flash
+# kernel SIGFPEs (integer div-by-zero, uncatchable) on some sharded
+# probe combos, and the algebra is closed-form anyway: batch and head
+# dims shard freely (attention mixes only within a head's sequence),
+# an attn_mask shards along iff its dim matches (1 = broadcast stays
+# replicated).
+def _sdpa_shardable_dims(shapes, n_qkv, mask_idx):
+    """Yield (dim, participating-input-indices) for batch/head dims."""
+    q = shapes[0]   # bwd's grad_out has q's [B,H,S,D] layout too
+    for dim in (0, 1):
+        size = q[dim]
+        if size <= 1:
+            continue
+        idxs = list(range(n_qkv))
+        if mask_idx is not None:
+            m = shapes[mask_idx]
+            if len(m) == 4 and m[dim] == size:
+                idxs.append(mask_idx)
+            elif len(m) == 4 and m[dim] != 1:
+                continue                          # incompatible mask
+        yield dim, idxs
+
+
+@register_preset(
+    aten._scaled_dot_product_flash_attention_for_cpu.default)
+def _sdpa_cpu_rule(input_shapes, args, kwargs):
+    # tensors: (q, k, v[, attn_mask]); outputs (out [B,H,S,D], lse [B,H,S])
+    if len(input_shapes[0]) != 4:
+        return None, {}
+    mask_idx = 3 if len(input_shapes) > 3 \
+        and len(input_shapes[3]) == 4 else None
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for dim, idxs in _sdpa_shardable_dims(input_shapes, 3, mask_idx):
+        for i in idxs:
+            ann[i][dim] = ShardDim.get_shard_dim(sid)
+        combs[sid] = [_gather(dim), _gather(dim)]
+        sid += 1
+    return ann, combs
+
+
+@register_preset(
+    aten._scaled_dot_product_flash_attention_for_cpu_backward.default)
+def _sdpa_cpu_bwd_rule(input_shapes, args, kwargs):
+    # tensors: (grad, q, k, v, out, lse[, attn_mask]); outputs (dq,dk,dv)
+    if len(input_shapes) < 6 or len(input_shapes[1]) != 4:
+        return None, {}
+    mask_idx = 6 if len(input_shapes) > 6 \
+        and len(input_shapes[6]) == 4 else None
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for dim, idxs in _sdpa_shardable_dims(input_shapes, 6, mask_idx):
+        for i in idxs:
+            ann[i][dim] = ShardDim.get_shard_dim(sid)
+        combs[sid] = [_gather(dim)] * 3
+        sid += 1
+    return ann, combs
+
+
 # ------------------------------------------------------- pipeline markers ----
 def _register_pp_markers():
     """pp_split/step_split are identities: shard through any dim.

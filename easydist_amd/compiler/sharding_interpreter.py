@@ -27,6 +27,19 @@ logger = logging.getLogger(__name__)
 _DISCOVERY_CACHE: Dict[Tuple, Tuple] = {}
 
 
+_probe_seed = [0]
+
+
+def _probe_gen():
+    """Deterministic per-probe generator: counter-seeded so (a) every rank
+    draws the same probe sequence (discovery must agree across ranks
+    without communicating) and (b) distinct args of one op still get
+    DISTINCT values (identical probes make sub(a,b)=0-style symmetric
+    outputs that verify false rules)."""
+    _probe_seed[0] += 1
+    return torch.Generator(device="cpu").manual_seed(0x9E3779B9 + _probe_seed[0])
+
+
 def _to_real(meta: torch.Tensor, device=None, promote_fp64=True) -> torch.Tensor:
     """Materialize a random tensor matching a FakeTensor's meta.
 
@@ -37,16 +50,25 @@ def _to_real(meta: torch.Tensor, device=None, promote_fp64=True) -> torch.Tensor
     device = device or meta.device
     if meta.dtype.is_floating_point:
         dt = torch.float64 if promote_fp64 else meta.dtype
-        t = torch.rand(meta.shape, dtype=torch.float64, device=device) + 0.5
-        return t.to(dt)
+        t = torch.rand(meta.shape, dtype=torch.float64,
+                       generator=_probe_gen()) + 0.5
+        return t.to(dt).to(device)
     if meta.dtype == torch.bool:
-        return torch.rand(meta.shape, device=device) > 0.5
-    # integer tensors: ZEROS. Index 0 is in-bounds for every non-empty
-    # dim even after the probe shards a tensor down (a size-2 dim sharded
-    # 2-ways leaves size-1 shards where index 1 hardware-faults the
-    # aten index kernels — observed as HSA_STATUS_ERROR_EXCEPTION on the
-    # MoE routing graph).
-    return torch.zeros(meta.shape, dtype=meta.dtype, device=device)
+        return (torch.rand(meta.shape, generator=_probe_gen()) > 0.5
+                ).to(device)
+    # integer tensors: seeded random {0,1}. Values stay in-bounds for
+    # every indexed dim of size >= 2 even after the probe shards a tensor
+    # down (larger random values hardware-faulted the aten index kernels
+    # on sharded MoE routing probes: HSA_STATUS_ERROR_EXCEPTION), while a
+    # CONSTANT fill minted false rules — e.g. all-zero position_ids made
+    # slice(pos, 1, 0, T) on a dim-1-sharded input look like identity,
+    # and rank 1 then silently read the wrong positions. Value-dependent
+    # index/sort ops never reach execution probing (preset blacklist);
+    # ops that do index with these values on a size-1 dim raise a normal
+    # (catchable) error on the CPU discovery device and just lose the
+    # candidate rule. Fixed seed: every rank must discover identically.
+    return torch.randint(0, 2, meta.shape, generator=_probe_gen(),
+                         dtype=meta.dtype, device="cpu").to(device)
 
 
 def _sig_of(node: fx.Node):
@@ -69,6 +91,7 @@ class EDTorchShardingAnn:
         self.device = device or ("cuda" if torch.cuda.is_available() else "cpu")
 
     def run(self) -> Dict[str, Tuple[Optional[ShardAnnotation], dict]]:
+        _probe_seed[0] = 0    # history-independent probe sequence
         info: Dict[str, Tuple] = {}
         for node in self.gm.graph.nodes:
             if node.op != "call_function":
