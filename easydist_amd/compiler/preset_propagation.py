@@ -397,21 +397,35 @@ _EW_BINARY = [
 def _elementwise_rule(input_shapes, args, kwargs):
     if not input_shapes:
         return None
-    shape0 = input_shapes[0]
-    # every TENSOR input must have the exact same shape (no broadcasting:
-    # broadcast patterns go to execution-based discovery)
-    for sh in input_shapes[1:]:
-        if tuple(sh) != tuple(shape0):
-            return None
+    # trailing-aligned broadcasting: an input participates in a dim's
+    # shard iff it actually has that dim at full size; size-1/absent
+    # (broadcast) inputs stay replicated, which is exactly correct
+    out_rank = max(len(s) for s in input_shapes)
+    out = []
+    for o in range(out_rank):
+        m = 1
+        for sh in input_shapes:
+            di = o - (out_rank - len(sh))
+            if di >= 0 and sh[di] != 1:
+                if m != 1 and sh[di] != m:
+                    return None           # invalid broadcast combo
+                m = sh[di]
+        out.append(m)
     ann = ShardAnnotation.init_from_input_shapes(input_shapes)
     combs = {}
     sid = 1
-    for d in range(len(shape0)):
-        if shape0[d] <= 1:
+    for o in range(out_rank):
+        if out[o] <= 1:
             continue
-        for i in range(len(input_shapes)):
-            ann[i][d] = ShardDim.get_shard_dim(sid)
-        combs[sid] = _gather(d)
+        parts = [(i, o - (out_rank - len(sh)))
+                 for i, sh in enumerate(input_shapes)
+                 if o - (out_rank - len(sh)) >= 0
+                 and sh[o - (out_rank - len(sh))] == out[o]]
+        if not parts:
+            continue
+        for i, di in parts:
+            ann[i][di] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(o)
         sid += 1
     if not combs:
         return None
@@ -441,3 +455,287 @@ _VALUE_DEPENDENT = [
 @register_preset(*_VALUE_DEPENDENT)
 def _value_dependent_rule(input_shapes, args, kwargs):
     return (None, {})
+
+
+# ---------------------------------------------------- permutation family ----
+# Pure index permutations: every dim shards; only the gather position
+# moves. Probing these by execution (fp64 copies of full activations)
+# was the bulk of the residual GPT-2 discovery time.
+def _perm_rule(input_shapes, perm_of_out):
+    """perm_of_out[out_dim] = in_dim."""
+    shape = input_shapes[0]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for o, i in enumerate(perm_of_out):
+        if shape[i] <= 1:
+            continue
+        ann[0][i] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(o)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.t.default)
+def _t_rule(input_shapes, args, kwargs):
+    if len(input_shapes[0]) != 2:
+        return None
+    return _perm_rule(input_shapes, [1, 0])
+
+
+@register_preset(aten.transpose.int)
+def _transpose_rule(input_shapes, args, kwargs):
+    r = len(input_shapes[0])
+    d0, d1 = args[1] % r, args[2] % r
+    perm = list(range(r))
+    perm[d0], perm[d1] = perm[d1], perm[d0]
+    return _perm_rule(input_shapes, perm)
+
+
+@register_preset(aten.permute.default)
+def _permute_rule(input_shapes, args, kwargs):
+    r = len(input_shapes[0])
+    return _perm_rule(input_shapes, [d % r for d in args[1]])
+
+
+@register_preset(aten.unsqueeze.default)
+def _unsqueeze_rule(input_shapes, args, kwargs):
+    shape = input_shapes[0]
+    r = len(shape)
+    d = args[1] % (r + 1)
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for i in range(r):
+        if shape[i] <= 1:
+            continue
+        ann[0][i] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(i + (1 if i >= d else 0))
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.squeeze.dim, aten.squeeze.dims, aten.squeeze.default)
+def _squeeze_rule(input_shapes, args, kwargs):
+    shape = input_shapes[0]
+    r = len(shape)
+    if len(args) < 2:
+        drop = [i for i in range(r) if shape[i] == 1]
+    else:
+        ds = args[1] if isinstance(args[1], (list, tuple)) else [args[1]]
+        drop = [d % r for d in ds if shape[d % r] == 1]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for i in range(r):
+        if shape[i] <= 1 or i in drop:
+            continue
+        o = i - sum(1 for d in drop if d < i)
+        ann[0][i] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(o)
+        sid += 1
+    return ann, combs
+
+
+# --------------------------------------------------------- reductions -------
+@register_preset(aten.sum.dim_IntList, aten.mean.dim)
+def _dim_reduce_rule(input_shapes, args, kwargs):
+    shape = input_shapes[0]
+    r = len(shape)
+    dims = args[1] if len(args) > 1 and args[1] is not None \
+        else list(range(r))
+    dims = [d % r for d in (dims if isinstance(dims, (list, tuple))
+                            else [dims])]
+    keep = bool(args[2]) if len(args) > 2 else bool(kwargs.get("keepdim"))
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for i in range(r):
+        if shape[i] <= 1:
+            continue
+        if i in dims:
+            continue   # reduced-dim shard would be PARTIAL(sum) for sum /
+                       # not representable for mean; conservatively omitted
+        o = i if keep else i - sum(1 for d in dims if d < i)
+        ann[0][i] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(o)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.pow.Scalar)
+def _pow_scalar_rule(input_shapes, args, kwargs):
+    # pow.Scalar(scalar_base, tensor_exponent): elementwise in the tensor
+    shape = input_shapes[0]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(shape)):
+        if shape[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    if not combs:
+        return None
+    return ann, combs
+
+
+# ------------------------------------------------------------ cat/split -----
+@register_preset(aten.cat.default)
+def _cat_rule(input_shapes, args, kwargs):
+    if not input_shapes:
+        return None
+    r = len(input_shapes[0])
+    cdim = (args[1] if len(args) > 1 else kwargs.get("dim", 0)) % r
+    if any(len(s) != r for s in input_shapes):
+        return None
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(r):
+        if d == cdim:
+            continue
+        if any(s[d] != input_shapes[0][d] for s in input_shapes) \
+                or input_shapes[0][d] <= 1:
+            continue
+        for i in range(len(input_shapes)):
+            ann[i][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.split.Tensor)
+def _split_rule(input_shapes, args, kwargs):
+    import math as _math
+    shape = input_shapes[0]
+    r = len(shape)
+    sdim = (args[2] if len(args) > 2 else kwargs.get("dim", 0)) % r
+    size = args[1]
+    if isinstance(size, (list, tuple)):
+        n_out = len(size)
+    else:
+        n_out = _math.ceil(shape[sdim] / size) if size else 0
+    if n_out <= 0:
+        return None
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(r):
+        if d == sdim or shape[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = [_gather(d)] * n_out
+        sid += 1
+    return ann, combs
+
+
+# ------------------------------------------------------------ layer norm ----
+@register_preset(aten.native_layer_norm.default)
+def _ln_rule(input_shapes, args, kwargs):
+    # tensors: (x[, w][, b]); normalized = trailing len(args[1]) dims
+    x = input_shapes[0]
+    n_norm = len(args[1])
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(x) - n_norm):
+        if x[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = [_gather(d), _gather(d), _gather(d)]
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.native_layer_norm_backward.default)
+def _ln_bwd_rule(input_shapes, args, kwargs):
+    # tensors: (gout, x, mean, rstd[, w][, b]); outputs (dx, dw, db);
+    # batch-dim shard -> dx gathers, dw/db are PARTIAL sums
+    x = input_shapes[1]
+    n_norm = len(args[2])
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(len(x) - n_norm):
+        if x[d] <= 1:
+            continue
+        for i in range(4):      # gout, x, mean, rstd share batch layout
+            if d < len(input_shapes[i]):
+                ann[i][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = [_gather(d), _reduce_add(), _reduce_add()]
+        sid += 1
+    return ann, combs
+
+
+# ------------------------------------------------- easydist_amd custom ops --
+def _register_custom_ops():
+    """Sharding algebra of our own gfx950 kernels is known exactly; their
+    fp64 aten fallbacks are among the most expensive things execution
+    discovery can run (full-activation flash attention, [N, vocab] CE)."""
+    try:
+        from .. import ops as _ops  # noqa: F401  (defines torch.ops.easydist_amd)
+    except Exception:               # pragma: no cover
+        return
+    ed = torch.ops.easydist_amd
+
+    @register_preset(ed.flash_attention.default)
+    def _fa_rule(input_shapes, args, kwargs):
+        # (q,k,v [B,H,S,D]) -> (out [B,H,S,D], lse [B,H,S])
+        if len(input_shapes[0]) != 4:
+            return None, {}
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        combs = {}
+        sid = 1
+        for d in (0, 1):
+            if input_shapes[0][d] <= 1:
+                continue
+            for i in range(3):
+                ann[i][d] = ShardDim.get_shard_dim(sid)
+            combs[sid] = [_gather(d), _gather(d)]
+            sid += 1
+        return ann, combs
+
+    @register_preset(ed.flash_attention_bwd.default)
+    def _fa_bwd_rule(input_shapes, args, kwargs):
+        # (grad,q,k,v,out [B,H,S,D], lse [B,H,S]) -> (dq,dk,dv)
+        if len(input_shapes) < 6 or len(input_shapes[1]) != 4:
+            return None, {}
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        combs = {}
+        sid = 1
+        for d in (0, 1):
+            if input_shapes[1][d] <= 1:
+                continue
+            for i in range(6):
+                ann[i][d] = ShardDim.get_shard_dim(sid)
+            combs[sid] = [_gather(d)] * 3
+            sid += 1
+        return ann, combs
+
+    @register_preset(ed.ce_fwd.default)
+    def _ce_rule(input_shapes, args, kwargs):
+        # (logits [N,C], targets [N]) -> (nll SUM scalar, lse [N]):
+        # row shard -> loss PARTIAL(sum), lse gathers; class shard not
+        # representable (partial logsumexp)
+        if len(input_shapes[0]) != 2 or input_shapes[0][0] <= 1:
+            return None, {}
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        ann[0][0] = ShardDim.get_shard_dim(1)
+        ann[1][0] = ShardDim.get_shard_dim(1)
+        return ann, {1: [_reduce_add(), _gather(0)]}
+
+    @register_preset(ed.ce_bwd.default)
+    def _ce_bwd_rule(input_shapes, args, kwargs):
+        # (grad scalar, logits [N,C], targets [N], lse [N]) -> dlogits
+        if len(input_shapes) < 4 or len(input_shapes[1]) != 2 \
+                or input_shapes[1][0] <= 1:
+            return None, {}
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        for i in (1, 2, 3):
+            ann[i][0] = ShardDim.get_shard_dim(1)
+        return ann, {1: _gather(0)}
+
+
+_register_custom_ops()
