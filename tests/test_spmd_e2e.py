@@ -150,3 +150,37 @@ def _run_golden_2d(world_size):
 @pytest.mark.world4
 def test_auto_spmd_2d_mesh_ws4():
     spawn(_run_golden_2d, args=(4,), world_size=4, port=29533)
+
+
+def _pure_fn_body(world_size):
+    """easydist_compile of a plain function — no module, no optimizer
+    (reference: tests/test_torch/test_simple.py's fn_2 shape)."""
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+
+    def fn(x, y):
+        return torch.mm(torch.exp(torch.tanh(x)), y)
+
+    c = easydist_compile(fn, cuda_graph=False)
+    torch.manual_seed(0)
+    x = torch.randn(8, 8)
+    y = torch.randn(8, 8)
+    dist.broadcast(x, src=0)
+    dist.broadcast(y, src=0)
+    out = c(x, y)
+    assert torch.allclose(out, fn(x, y), rtol=1e-5, atol=1e-6)
+
+
+def test_pure_function_ws1():
+    init_single_process()
+    _pure_fn_body(1)
+
+
+@pytest.mark.world2
+def test_pure_function_ws2():
+    spawn(_pure_fn_body, args=(2,), world_size=2, port=29534)
