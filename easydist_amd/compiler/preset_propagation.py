@@ -739,3 +739,142 @@ def _register_custom_ops():
 
 
 _register_custom_ops()
+
+
+# ------------------------------------------------------------ conv / bn -----
+# fp64 execution probes of full-size convolutions are the dominant
+# discovery cost on the (w)resnet family; the algebra is closed-form.
+@register_preset(aten.convolution.default)
+def _conv_rule(input_shapes, args, kwargs):
+    # (x [N,Ci,*sp], w [Co,Ci/g,*k][, bias [Co]]); groups = args[-1]
+    if len(input_shapes) < 2 or len(input_shapes[0]) < 3:
+        return None
+    x, w = input_shapes[0], input_shapes[1]
+    has_bias = len(input_shapes) > 2
+    transposed = bool(args[6]) if len(args) > 6 else False
+    groups = args[8] if len(args) > 8 else 1
+    if transposed:
+        return None
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    if x[0] > 1:                       # batch
+        ann[0][0] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(0)
+        sid += 1
+    if groups == 1 and w[0] > 1:       # out channels: w + bias
+        ann[1][0] = ShardDim.get_shard_dim(sid)
+        if has_bias:
+            ann[2][0] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(1)
+        sid += 1
+    if groups == 1 and not has_bias and x[1] > 1:   # in channels: PARTIAL
+        ann[0][1] = ShardDim.get_shard_dim(sid)
+        ann[1][1] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _reduce_add()
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.convolution_backward.default)
+def _conv_bwd_rule(input_shapes, args, kwargs):
+    # (gout [N,Co,*], x [N,Ci,*], w [Co,Ci/g,*k]) -> (dx, dw, db)
+    if len(input_shapes) < 3 or len(input_shapes[0]) < 3:
+        return None
+    transposed = bool(args[7]) if len(args) > 7 else False
+    groups = args[9] if len(args) > 9 else 1
+    if transposed:
+        return None
+    gout, x, w = input_shapes[:3]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    if x[0] > 1:                       # batch: dw/db are partial sums
+        ann[0][0] = ShardDim.get_shard_dim(sid)
+        ann[1][0] = ShardDim.get_shard_dim(sid)
+        combs[sid] = [_gather(0), _reduce_add(), _reduce_add()]
+        sid += 1
+    if groups == 1 and gout[1] > 1:    # out channels: dx partial
+        ann[0][1] = ShardDim.get_shard_dim(sid)
+        ann[2][0] = ShardDim.get_shard_dim(sid)
+        combs[sid] = [_reduce_add(), _gather(0), _gather(0)]
+        sid += 1
+    return ann, combs
+
+
+def _bn_rule_n(input_shapes, n_out):
+    # channel shard only: batch/spatial sharding changes the TRAINING
+    # batch statistics (per-shard mean != global mean), so it must not
+    # be offered. tensors: (x [N,C,*], w [C], b [C][, rm [C], rv [C]])
+    x = input_shapes[0]
+    if len(x) < 2 or x[1] <= 1:
+        return None, {}
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    ann[0][1] = ShardDim.get_shard_dim(1)
+    for i in range(1, len(input_shapes)):
+        if len(input_shapes[i]) == 1 and input_shapes[i][0] == x[1]:
+            ann[i][0] = ShardDim.get_shard_dim(1)
+    # out gathers on channel; save_mean/save_var (and the functional
+    # variant's updated running stats) gather on their dim 0
+    combs = {1: [_gather(1)] + [_gather(0)] * (n_out - 1)}
+    return ann, combs
+
+
+@register_preset(aten.native_batch_norm.default,
+                 aten._native_batch_norm_legit.default)
+def _bn_rule(input_shapes, args, kwargs):
+    return _bn_rule_n(input_shapes, 3)
+
+
+@register_preset(aten._native_batch_norm_legit_functional.default)
+def _bn_func_rule(input_shapes, args, kwargs):
+    return _bn_rule_n(input_shapes, 5)
+
+
+@register_preset(aten.native_batch_norm_backward.default)
+def _bn_bwd_rule(input_shapes, args, kwargs):
+    # (gout, x, w, rm, rv, sm, sv) -> (dx, dw, db): channel shard
+    x = input_shapes[1] if len(input_shapes) > 1 else None
+    if x is None or len(x) < 2 or x[1] <= 1:
+        return None, {}
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    ann[0][1] = ShardDim.get_shard_dim(1)
+    ann[1][1] = ShardDim.get_shard_dim(1)
+    for i in range(2, len(input_shapes)):
+        if len(input_shapes[i]) == 1 and input_shapes[i][0] == x[1]:
+            ann[i][0] = ShardDim.get_shard_dim(1)
+    return ann, {1: [_gather(1), _gather(0), _gather(0)]}
+
+
+@register_preset(aten.max_pool2d_with_indices.default)
+def _maxpool_rule(input_shapes, args, kwargs):
+    # indices are within-plane offsets: invariant under N/C sharding
+    x = input_shapes[0]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in (0, 1):
+        if d < len(x) - 2 and x[d] > 1:
+            ann[0][d] = ShardDim.get_shard_dim(sid)
+            combs[sid] = [_gather(d), _gather(d)]
+            sid += 1
+    return ann, combs
+
+
+@register_preset(aten.max_pool2d_with_indices_backward.default)
+def _maxpool_bwd_rule(input_shapes, args, kwargs):
+    # (gout, x, ..., indices) -> dx
+    x = input_shapes[1] if len(input_shapes) > 1 else None
+    if x is None:
+        return None
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in (0, 1):
+        if d < len(x) - 2 and x[d] > 1:
+            for i in range(len(input_shapes)):
+                if d < len(input_shapes[i]) - 2:
+                    ann[i][d] = ShardDim.get_shard_dim(sid)
+            combs[sid] = _gather(d)
+            sid += 1
+    return ann, combs
