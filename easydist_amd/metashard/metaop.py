@@ -49,7 +49,10 @@ def shard_tensor(tensor: torch.Tensor, dim: int, num_shards: int,
         hi = min(bounds[i + 1] + (halo if i < num_shards - 1 else 0), size)
         idx = [slice(None)] * tensor.dim()
         idx[dim] = slice(lo, hi)
-        shards.append(tensor[tuple(idx)])
+        # contiguous: some aten CPU kernels (native_layer_norm_backward)
+        # silently misread strided views — a view shard makes discovery
+        # verification compare garbage numerics
+        shards.append(tensor[tuple(idx)].contiguous())
     return shards
 
 
