@@ -78,12 +78,12 @@ def _body(world_size):
 
 @pytest.mark.world2
 def test_gpt_tp_ws2():
-    spawn(_body, args=(2,), world_size=2, port=29591)
+    spawn(_body, args=(2,), world_size=2, port=29621)
 
 
 @pytest.mark.world4
 def test_gpt_tp_ws4():
-    spawn(_body, args=(4,), world_size=4, port=29592)
+    spawn(_body, args=(4,), world_size=4, port=29622)
 
 
 def _train_body(world_size):
@@ -121,4 +121,4 @@ def _train_body(world_size):
 
 @pytest.mark.world2
 def test_gpt_tp_train_ws2():
-    spawn(_train_body, args=(2,), world_size=2, port=29593)
+    spawn(_train_body, args=(2,), world_size=2, port=29623)
