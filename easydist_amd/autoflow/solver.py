@@ -210,33 +210,45 @@ class AutoFlowSolver1D:
 
     # --------------------------------------------------------- beam search ---
     def beam_search(self) -> Dict[str, ClusterStrategy]:
-        """Greedy-ish beam over clusters in topo order."""
-        beam: List[Tuple[float, Dict[str, int], Dict[str, SPMD]]] = [(0.0, {}, {})]
+        """Beam over clusters in topo order.
+
+        Candidates are SCORED against the parent's var-placement dict
+        (no allocation); dicts are materialized only for the `width`
+        survivors of each step — the naive per-candidate dict copies
+        made this O(clusters² · width) and slower than the timed-out
+        MILP it is supposed to replace."""
         width = mdconfig.beam_width
-        edges_by_consumer: Dict[str, List] = {}
-        for cp, cc, v, nbytes in self._edges():
-            edges_by_consumer.setdefault(cc.name, []).append((cp, v, nbytes))
+        # beam entries: (cost, assign dict, var placements dict)
+        beam: List[Tuple[float, Dict[str, int], Dict[str, SPMD]]] = \
+            [(0.0, {}, {})]
+        var_bytes = {name: v.nbytes for name, v in self.graph.vars.items()}
         for c in self.clusters:
-            new_beam = []
             unary = self._unary_costs(c)
-            for cost, assign, var_pl in beam:
+            cands = []            # (new_cost, parent_idx, strategy_idx)
+            for pi, (cost, _assign, var_pl) in enumerate(beam):
+                get = var_pl.get
                 for s, st in enumerate(c.strategies):
                     add = float(unary[s])
                     for v, need in st.in_placements.items():
-                        have = var_pl.get(v)
+                        have = get(v)
                         if have is not None:
-                            nbytes = (self.graph.vars[v].nbytes
-                                      if v in self.graph.vars else 0)
-                            add += reshard_cost(have, need, nbytes, self.mesh_size)
-                    na = dict(assign)
-                    na[c.name] = s
-                    npl = dict(var_pl)
-                    for v, pl in st.out_placements.items():
-                        npl[v] = pl
-                    new_beam.append((cost + add, na, npl))
-            new_beam.sort(key=lambda t: t[0])
-            beam = new_beam[:width]
+                            add += reshard_cost(have, need,
+                                                var_bytes.get(v, 0),
+                                                self.mesh_size)
+                    cands.append((cost + add, pi, s))
+            cands.sort(key=lambda t: t[0])
+            new_beam = []
+            for cost, pi, s in cands[:width]:
+                _pc, passign, pvar = beam[pi]
+                na = dict(passign)
+                na[c.name] = s
+                npl = dict(pvar)
+                npl.update(c.strategies[s].out_placements)
+                new_beam.append((cost, na, npl))
+            beam = new_beam
         best = beam[0]
+        logger.info("beam search: %d clusters, width %d, cost %.3e",
+                    len(self.clusters), width, best[0])
         # follow-edge costs applied post-hoc are already included via
         # output_constraints unary terms; good enough for the fallback.
         return {c.name: c.strategies[best[1][c.name]] for c in self.clusters}

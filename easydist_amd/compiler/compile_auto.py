@@ -200,7 +200,14 @@ def shard_graph(gm, mesh, io_map, ret_names, device, fix_rets=True,
             solver = AutoFlowSolver1D(meta_graph, size, already_sharded,
                                       output_constraints)
             solver.add_coarsen_graph(clusters)
-            choice = (solver.beam_search() if mdconfig.solver_mode == "beam"
+            use_beam = (mdconfig.solver_mode == "beam"
+                        or len(clusters) > mdconfig.ilp_max_clusters)
+            if use_beam and mdconfig.solver_mode != "beam":
+                logger.info("%d clusters > ilp_max_clusters=%d: using beam "
+                            "search (the timed-out MILP incumbent is worse "
+                            "than the beam solution at this scale)",
+                            len(clusters), mdconfig.ilp_max_clusters)
+            choice = (solver.beam_search() if use_beam
                       else solver.ilp_solve())
             node_strats: Dict = {}
             for st in choice.values():
@@ -248,6 +255,8 @@ def _strategy_cache_path(gm, mesh) -> Optional[str]:
     # rule changes must invalidate cached strategies
     from .preset_propagation import _PRESET_REGISTRY
     h.update(repr(sorted(str(k) for k in _PRESET_REGISTRY)).encode())
+    h.update(f"{mdconfig.solver_mode}/{mdconfig.beam_width}/"
+             f"{mdconfig.ilp_max_clusters}".encode())
     d = os.path.join(os.path.expanduser("~"), ".easydist_amd",
                      "compile_cache")
     os.makedirs(d, exist_ok=True)
@@ -310,10 +319,13 @@ def _verify_strategy_agreement(strategies_per_dim):
         if not torch.equal(t, ref):
             logger.warning("strategy solve diverged from rank 0: adopting "
                            "rank 0's result")
-        payload = [strategies_per_dim]
+        # copy first: on rank 0 payload[0] IS strategies_per_dim and
+        # clear()+extend(itself) would leave rank 0 with an empty list
+        payload = [list(strategies_per_dim)]
         dist.broadcast_object_list(payload, src=0)
-        strategies_per_dim.clear()
-        strategies_per_dim.extend(payload[0])
+        if not torch.equal(t, ref):
+            strategies_per_dim.clear()
+            strategies_per_dim.extend(payload[0])
 
 
 def _adam_positions(params, buffers, named_states):

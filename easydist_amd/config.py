@@ -67,8 +67,16 @@ solver_time_limit = _env_float("EASYDIST_SOLVER_TIME_LIMIT", 120.0)
 max_seconds_same_incumbent = float("inf")
 enable_graph_coarsen = _env_flag("EASYDIST_ENABLE_GRAPH_COARSEN", True)
 coarsen_level = _env_int("EASYDIST_COARSEN_LEVEL", 1)
+# cap on nodes merged into one cone cluster: bigger cones = smaller MILP
+# (fewer reshard boundaries); GPT-2-small at bench shape needs >12 to
+# keep the MILP under the time limit
+coarsen_max_cluster = _env_int("EASYDIST_COARSEN_MAX_CLUSTER", 12)
 solver_mode = os.environ.get("EASYDIST_SOLVER_MODE", "ilp")  # ilp | beam
-beam_width = _env_int("EASYDIST_BEAM_WIDTH", 1024)
+beam_width = _env_int("EASYDIST_BEAM_WIDTH", 64)
+# above this cluster count the MILP no longer converges inside the time
+# limit (its timed-out incumbent is half-replicated and rank-divergent);
+# beam search finds the clean data/tensor-parallel assignment in seconds
+ilp_max_clusters = _env_int("EASYDIST_ILP_MAX_CLUSTERS", 1200)
 all_to_all_punish_factor = _env_float("EASYDIST_A2A_PUNISH", 1.5)
 liveness_only_input = False
 

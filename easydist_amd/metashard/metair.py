@@ -303,9 +303,10 @@ class MetaGraph:
             in_b = sum(v.nbytes for v in node.invars if v is not None)
             return out_b < in_b
 
+        from .. import config as mdconfig
         cluster_of: Dict[str, MetaNodeCluster] = {}
         clusters: List[MetaNodeCluster] = []
-        MAX_CLUSTER = 12
+        MAX_CLUSTER = mdconfig.coarsen_max_cluster
         # reverse topo: consumers first
         for node in reversed(self.nodes):
             target_cluster = None

@@ -184,3 +184,45 @@ def test_pure_function_ws1():
 @pytest.mark.world2
 def test_pure_function_ws2():
     spawn(_pure_fn_body, args=(2,), world_size=2, port=29534)
+
+
+def _beam_body(world_size):
+    """Beam-search solver path (used automatically above
+    ilp_max_clusters): golden vs vanilla on gloo ws2."""
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, mdconfig, \
+        set_device_mesh
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+    mdconfig.solver_mode = "beam"
+    mdconfig.enable_compile_cache = False   # force a real beam solve
+    try:
+        torch.manual_seed(42)
+        model = MLP()
+        for p in model.parameters():
+            dist.broadcast(p.data, src=0)
+        model_ref = copy.deepcopy(model)
+        opt = torch.optim.Adam(model.parameters(), lr=1e-2, fused=True)
+        opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-2,
+                                   fused=True)
+        compiled = easydist_compile(train_step, cuda_graph=False)
+        torch.manual_seed(7)
+        for step in range(3):
+            x = torch.randn(8, 16)
+            y = torch.randn(8, 16)
+            dist.broadcast(x, src=0)
+            dist.broadcast(y, src=0)
+            loss = compiled(model, opt, x, y)
+            ref = train_step(model_ref, opt_ref, x, y)
+            assert abs(float(loss) - float(ref)) < 1e-4, \
+                (step, float(loss), float(ref))
+    finally:
+        mdconfig.solver_mode = "ilp"
+        mdconfig.enable_compile_cache = True
+
+
+@pytest.mark.world2
+def test_beam_solver_ws2():
+    spawn(_beam_body, args=(2,), world_size=2, port=29535)
