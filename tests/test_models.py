@@ -84,3 +84,63 @@ def test_resnet18_compiled_golden():
         ref = step(model_ref, opt_ref, x, y)
         assert abs(float(loss) - float(ref)) < 1e-3, \
             (i, float(loss), float(ref))
+
+
+def test_gat_compiled_golden():
+    """GAT through auto-SPMD (masked softmax, dense adjacency, nan_to_num
+    — value-heavy ops must not mint false rules)."""
+    import copy
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+    from easydist_amd.models.gat import GAT, gat_train_step
+    from easydist_amd.utils.testing import init_single_process
+
+    init_single_process()
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([0], ["spmd0"])
+    torch.manual_seed(0)
+    model = GAT(in_dim=32, hidden=16, n_classes=8)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+    opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-3, fused=True)
+    compiled = easydist_compile(gat_train_step, cuda_graph=False)
+    torch.manual_seed(3)
+    x = torch.randn(24, 32)
+    adj = (torch.rand(24, 24) < 0.3).float()
+    adj.fill_diagonal_(1)
+    y = torch.randint(0, 8, (24,))
+    for i in range(2):
+        loss = compiled(model, opt, x, adj, y)
+        ref = gat_train_step(model_ref, opt_ref, x, adj, y)
+        assert abs(float(loss) - float(ref)) < 1e-4, \
+            (i, float(loss), float(ref))
+
+
+def test_vit_compiled_golden():
+    import copy
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+    from easydist_amd.models.vit import ViT, ViTConfig, vit_train_step
+    from easydist_amd.utils.testing import init_single_process
+
+    init_single_process()
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([0], ["spmd0"])
+    torch.manual_seed(0)
+    cfg = ViTConfig(image_size=32, patch_size=8, n_layer=2, n_head=2,
+                    n_embd=32, n_classes=10)
+    model = ViT(cfg)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+    opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-3, fused=True)
+    compiled = easydist_compile(vit_train_step, cuda_graph=False)
+    torch.manual_seed(3)
+    x = torch.randn(4, 3, 32, 32)
+    y = torch.randint(0, 10, (4,))
+    for i in range(2):
+        loss = compiled(model, opt, x, y)
+        ref = vit_train_step(model_ref, opt_ref, x, y)
+        assert abs(float(loss) - float(ref)) < 1e-4, \
+            (i, float(loss), float(ref))
