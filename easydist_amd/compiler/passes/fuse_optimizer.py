@@ -137,19 +137,26 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
     chain_wd = 0.0 if decoupled else wd
 
     def _all_replicate(*nodes):
-        """The fused call wires these nodes DIRECTLY, bypassing any
-        reshard the transform placed between them and the decomposed
-        chain — only safe when every placement is replicate."""
+        """The fused call wires these nodes DIRECTLY. Safe when every
+        placement is replicate, OR when ALL of them carry the SAME
+        placement vector: the Adam update is elementwise, so a uniform
+        sharding of param/grad/moments is exactly the local update on
+        each shard (the ZeRO-like assignments the solver picks at
+        world>1). Mixed placements would need the reshards the direct
+        wiring bypasses — rejected."""
         if not pl_env:
             return True
+        seen = []
         for n in nodes:
             pls = pl_env.get(n.name)
             if not pls:
                 continue
-            for pl in pls[0]:
-                if not pl.is_replicate():
-                    return False
-        return True
+            seen.append(tuple(repr(p) for p in pls[0]))
+        if not seen:
+            return True
+        if all(all(p == "R" for p in v) for v in seen):
+            return True
+        return len(set(seen)) == 1 and len(seen) == len(nodes)
 
     matched = []
     for p_pos, outs in param_positions.items():
