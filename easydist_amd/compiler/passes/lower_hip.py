@@ -83,7 +83,23 @@ def lower_layer_norm(gm: fx.GraphModule) -> int:
             if _dtype_of(grad) == torch.bfloat16:
                 g_bf = grad
             if g_bf is None or x_bf is None:
-                continue      # kernel is bf16-x only
+                # bf16 kernel doesn't apply — but the raw aten node can
+                # still receive mixed fp32/bf16 args at RUNTIME once
+                # resharding comm nodes (no meta) and fwd-side cast
+                # folding detach runtime dtypes from the trace's; aten
+                # hard-rejects that. Use the dtype-tolerant fp32
+                # reference op and cast outputs back to the trace's
+                # dtypes.
+                with graph.inserting_before(n):
+                    new = graph.call_function(
+                        torch.ops.easydist_amd.ln_bwd_ref.default,
+                        (grad, x, list(shape), mean, rstd, w, b,
+                         list(mask)))
+                _rewire_tuple(graph, n, new,
+                              kernel_dtypes=[torch.float32, torch.float32,
+                                             torch.float32])
+                n_lowered += 1
+                continue
             with graph.inserting_before(n):
                 new = graph.call_function(
                     torch.ops.easydist_amd.layer_norm_bwd.default,

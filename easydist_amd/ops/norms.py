@@ -17,6 +17,9 @@ lib.define("layer_norm_fwd(Tensor x, Tensor? w, Tensor? b, float eps) "
            "-> (Tensor, Tensor, Tensor)")
 lib.define("layer_norm_bwd(Tensor grad, Tensor x, Tensor mean, Tensor rstd, "
            "Tensor? w, bool[3] mask) -> (Tensor, Tensor, Tensor)")
+lib.define("ln_bwd_ref(Tensor grad, Tensor x, SymInt[] norm_shape, "
+           "Tensor mean, Tensor rstd, Tensor? w, Tensor? b, bool[] mask) "
+           "-> (Tensor, Tensor, Tensor)")
 lib.define("rms_norm_fwd(Tensor x, Tensor? w, float eps) -> (Tensor, Tensor)")
 lib.define("rms_norm_bwd(Tensor grad, Tensor x, Tensor rstd, Tensor? w) "
            "-> (Tensor, Tensor)")
@@ -37,6 +40,39 @@ def _ln_fwd_cuda(x, w, b, eps):
         b = b.to(x.dtype).contiguous() if b is not None else None
         return ext.layer_norm_fwd(x.contiguous(), w, b, eps)
     return _ln_fwd_aten(x, w, b, eps)
+
+
+def _ln_bwd_ref(grad, x, norm_shape, mean, rstd, w, b, mask):
+    """Dtype-tolerant native_layer_norm_backward: the sharded graph can
+    hand the un-lowered aten node mixed fp32/bf16 args after cast
+    folding around its (lowered) forward — aten rejects that. Compute
+    fully in fp32 and let the lowering pass cast outputs back to the
+    dtypes the graph's meta expects."""
+    f = lambda t: None if t is None else t.float().contiguous()
+    n = len(norm_shape)
+    # stats have one entry per non-normalized row: [*batch_dims, 1...]
+    ms = [int(d) for d in grad.shape[:grad.dim() - n]] + [1] * n
+    outs = torch.ops.aten.native_layer_norm_backward.default(
+        f(grad), f(x), list(norm_shape),
+        f(mean).reshape(ms), f(rstd).reshape(ms),
+        f(w), f(b), list(mask))
+    z = lambda k, like: (outs[k] if outs[k] is not None
+                         else torch.zeros_like(like, dtype=torch.float32))
+    dw_like = w if w is not None else grad.new_empty(norm_shape)
+    return (outs[0] if outs[0] is not None else torch.zeros_like(grad,
+                                                                 dtype=torch.float32),
+            z(1, dw_like), z(2, dw_like))
+
+
+lib.impl("ln_bwd_ref", _ln_bwd_ref, "CompositeExplicitAutograd")
+
+
+@torch.library.register_fake("easydist_amd::ln_bwd_ref")
+def _ln_bwd_ref_fake(grad, x, norm_shape, mean, rstd, w, b, mask):
+    shp = list(norm_shape)
+    return (torch.empty_like(grad, dtype=torch.float32),
+            torch.empty(shp, dtype=torch.float32, device=grad.device),
+            torch.empty(shp, dtype=torch.float32, device=grad.device))
 
 
 def _ln_bwd_aten(grad, x, mean, rstd, w, mask):
