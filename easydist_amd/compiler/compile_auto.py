@@ -84,7 +84,10 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
         from .passes.fuse_optimizer import fuse_optimizer
         out_node_f = next(n for n in gm.graph.nodes if n.op == "output")
         flat_outs_f, spec_f = pytree.tree_flatten(out_node_f.args[0])
-        ppos = _adam_positions(params, buffers, named_states)
+        if type(opt).__name__ == "SGD":
+            ppos = _sgd_positions(params, buffers, named_states)
+        else:
+            ppos = _adam_positions(params, buffers, named_states)
         if ppos:
             nfused = fuse_optimizer(gm, flat_outs_f,
                                     [n for n in gm.graph.nodes
@@ -360,6 +363,31 @@ def _adam_positions(params, buffers, named_states):
             offset += 1
         if {"step", "exp_avg", "exp_avg_sq"} <= set(entry):
             out[p_pos] = entry
+    return out
+
+
+def _sgd_positions(params, buffers, named_states):
+    """Map each momentum-SGD param to its flat positions (momentum_buffer
+    state key); same flat-layout contract as _adam_positions."""
+    param_names = list(params.keys())
+    offset = len(params) + len(buffers)
+    out = {}
+    for pname, st in named_states.items():
+        n_t = len([v for v in st.values() if isinstance(v, torch.Tensor)])
+        if pname not in param_names or "momentum_buffer" not in st \
+                or not isinstance(st.get("momentum_buffer"), torch.Tensor):
+            offset += n_t
+            continue
+        p_pos = param_names.index(pname)
+        entry = {"param": p_pos}
+        for key, val in st.items():
+            if not isinstance(val, torch.Tensor):
+                continue
+            if key == "momentum_buffer":
+                entry["buf"] = offset
+                entry["buf_in"] = offset
+            offset += 1
+        out[p_pos] = entry
     return out
 
 

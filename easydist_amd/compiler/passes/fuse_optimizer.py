@@ -118,7 +118,12 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
     (positions into flat_outs; the same indexes are placeholder input
     positions for the state tensors).
     """
-    if opt is None or type(opt).__name__ not in ("Adam", "AdamW"):
+    if opt is None:
+        return 0
+    if type(opt).__name__ == "SGD":
+        return _fuse_sgd(gm, flat_outs, placeholders, param_positions, opt,
+                         pl_env)
+    if type(opt).__name__ not in ("Adam", "AdamW"):
         return 0
     decoupled = type(opt).__name__ == "AdamW"
     groups = opt.param_groups
@@ -207,5 +212,109 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
             flat_outs[outs["exp_avg_sq"]] = items[2]
             flat_outs[outs["step"]] = items[3]
     logger.info("fuse_optimizer: fused %d/%d parameter Adam chains",
+                len(matched), len(param_positions))
+    return len(matched)
+
+
+def _match_sgd_chain(p_ph, buf_ph, p_new, buf_new, momentum, wd, nesterov):
+    """Verify the decomposed momentum-SGD shape (decomp.py:
+    fused_sgd_decomp, is_first_step=False); return the RAW grad node."""
+    # buf_new = add(mul(buf_ph, momentum), mul(g', 1-dampening))
+    if not (isinstance(buf_new, fx.Node)
+            and buf_new.target is aten.add.Tensor):
+        return None
+    t0, s0 = _tensor_scalar(buf_new.args[0])
+    t1, s1 = _tensor_scalar(buf_new.args[1])
+    if t0 is buf_ph:
+        g = t1
+    elif t1 is buf_ph:
+        g = t0
+    else:
+        return None
+    if g is None:
+        # dampening == 0 traces `1 * g` away: the addend IS g'
+        g = buf_new.args[1] if t0 is buf_ph else buf_new.args[0]
+    if not isinstance(g, fx.Node):
+        return None
+    if wd != 0.0:
+        g = _strip_l2(g, p_ph, wd)
+        if g is None:
+            return None
+    # p_new = sub(p_ph, mul(upd, lr))
+    if not (isinstance(p_new, fx.Node)
+            and p_new.target is aten.sub.Tensor
+            and p_new.args[0] is p_ph):
+        return None
+    upd, _lr = _tensor_scalar(p_new.args[1])
+    if upd is None:
+        return None
+    if nesterov:
+        # upd = add(g', mul(buf_new, momentum))
+        if not (isinstance(upd, fx.Node)
+                and upd.target is aten.add.Tensor):
+            return None
+    elif upd is not buf_new:
+        return None
+    return g
+
+
+def _fuse_sgd(gm, flat_outs, placeholders, param_positions, opt, pl_env):
+    groups = opt.param_groups
+    if len(groups) != 1:
+        return 0
+    lr = float(groups[0]["lr"])
+    momentum = float(groups[0].get("momentum", 0.0))
+    dampening = float(groups[0].get("dampening", 0.0))
+    wd = float(groups[0].get("weight_decay", 0.0))
+    nesterov = bool(groups[0].get("nesterov", False))
+    if momentum == 0.0 or groups[0].get("maximize"):
+        return 0     # no state to fuse / unsupported
+
+    def _ok_placements(*nodes):
+        if not pl_env:
+            return True
+        seen = []
+        for n in nodes:
+            pls = pl_env.get(n.name)
+            if pls:
+                seen.append(tuple(repr(p) for p in pls[0]))
+        if not seen:
+            return True
+        if all(all(p == "R" for p in v) for v in seen):
+            return True
+        return len(set(seen)) == 1 and len(seen) == len(nodes)
+
+    matched = []
+    for p_pos, outs in param_positions.items():
+        p_ph = placeholders[p_pos]
+        buf_ph = placeholders[outs["buf_in"]]
+        p_new = flat_outs[outs["param"]]
+        buf_new = flat_outs[outs["buf"]]
+        g = _match_sgd_chain(p_ph, buf_ph, p_new, buf_new, momentum, wd,
+                             nesterov)
+        if g is None:
+            continue
+        if not _ok_placements(p_ph, buf_ph, g):
+            continue
+        matched.append((p_pos, outs, p_ph, g, buf_ph))
+    if not matched:
+        return 0
+
+    graph = gm.graph
+    out_node = next(n for n in graph.nodes if n.op == "output")
+    with graph.inserting_before(out_node):
+        fused = graph.call_function(
+            torch.ops.easydist_amd.fused_sgd_step.default,
+            ([m[2] for m in matched], [m[3] for m in matched],
+             [m[4] for m in matched], lr, momentum, dampening, wd,
+             nesterov))
+        lists = [graph.call_function(operator.getitem, (fused, k))
+                 for k in range(2)]
+        for i, (p_pos, outs, *_rest) in enumerate(matched):
+            flat_outs[outs["param"]] = graph.call_function(
+                operator.getitem, (lists[0], i))
+            flat_outs[outs["buf"]] = graph.call_function(
+                operator.getitem, (lists[1], i))
+    logger.info("fuse_optimizer: fused %d/%d parameter SGD chains",
                 len(matched), len(param_positions))
     return len(matched)

@@ -25,6 +25,10 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                 std::vector<at::Tensor> exp_avg_sqs,
                 std::vector<at::Tensor> steps, double lr, double beta1,
                 double beta2, double weight_decay, double eps);
+std::tuple<std::vector<at::Tensor>, std::vector<at::Tensor>>
+fused_sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+               std::vector<at::Tensor> bufs, double lr, double momentum,
+               double dampening, double weight_decay, bool nesterov);
 at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& bt,
                    const std::optional<at::Tensor>& bias);
 std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
@@ -56,6 +60,14 @@ adam_wrap(at::TensorList params, at::TensorList grads, at::TensorList exp_avgs,
                          weight_decay, eps);
 }
 
+static std::tuple<std::vector<at::Tensor>, std::vector<at::Tensor>>
+sgd_wrap(at::TensorList params, at::TensorList grads, at::TensorList bufs,
+         double lr, double momentum, double dampening, double weight_decay,
+         bool nesterov) {
+  return fused_sgd_step(params.vec(), grads.vec(), bufs.vec(), lr, momentum,
+                        dampening, weight_decay, nesterov);
+}
+
 TORCH_LIBRARY(easydist_amd_hip, m) {
   m.def("layer_norm_fwd(Tensor x, Tensor? w, Tensor? b, float eps) "
         "-> (Tensor, Tensor, Tensor)");
@@ -71,6 +83,9 @@ TORCH_LIBRARY(easydist_amd_hip, m) {
         "Tensor[] exp_avg_sqs, Tensor[] steps, float lr, float beta1, "
         "float beta2, float weight_decay, float eps) "
         "-> (Tensor[], Tensor[], Tensor[], Tensor[])");
+  m.def("fused_sgd_step(Tensor[] params, Tensor[] grads, Tensor[] bufs, "
+        "float lr, float momentum, float dampening, float weight_decay, "
+        "bool nesterov) -> (Tensor[], Tensor[])");
   m.def("gemm_nt(Tensor a, Tensor bt, Tensor? bias) -> Tensor");
   m.def("flash_attn_fwd(Tensor q, Tensor k, Tensor v, bool causal) "
         "-> (Tensor, Tensor)");
@@ -86,6 +101,7 @@ TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
   m.impl("ce_fwd", ce_fwd);
   m.impl("ce_bwd", ce_bwd);
   m.impl("fused_adam_step", adam_wrap);
+  m.impl("fused_sgd_step", sgd_wrap);
   m.impl("gemm_nt", gemm_nt);
   m.impl("flash_attn_fwd", flash_attn_fwd);
   m.impl("flash_attn_bwd", flash_attn_bwd);

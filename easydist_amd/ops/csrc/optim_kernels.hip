@@ -179,3 +179,109 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
     out_steps[t] = steps[t] + 1;
   return {out_p, out_ea, out_eas, out_steps};
 }
+
+// ---------------------------------------------------------------- SGD -------
+// Fused multi-tensor SGD(momentum) step: same chunk-table scheme as Adam
+// (own persistent staging — the two kernels must not clobber each
+// other's capture-recorded buffers).
+struct SgdChunk {
+  const float* p;
+  const float* g;
+  const float* buf;   // nullptr when momentum == 0
+  float* out_p;
+  float* out_buf;     // nullptr when momentum == 0
+  int n;
+};
+
+__global__ void fused_sgd_kernel(const SgdChunk* __restrict__ chunks,
+                                 int n_chunks, float lr, float momentum,
+                                 float dampening, float weight_decay,
+                                 int nesterov) {
+  const int c = blockIdx.x;
+  if (c >= n_chunks) return;
+  const SgdChunk ch = chunks[c];
+  for (int i = threadIdx.x; i < ch.n; i += blockDim.x) {
+    float grad = ch.g[i];
+    if (weight_decay != 0.f) grad += weight_decay * ch.p[i];
+    float upd = grad;
+    if (momentum != 0.f) {
+      float nb = momentum * ch.buf[i] + (1.f - dampening) * grad;
+      ch.out_buf[i] = nb;
+      upd = nesterov ? grad + momentum * nb : nb;
+    }
+    ch.out_p[i] = ch.p[i] - lr * upd;
+  }
+}
+
+std::tuple<std::vector<at::Tensor>, std::vector<at::Tensor>>
+fused_sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+               std::vector<at::Tensor> bufs, double lr, double momentum,
+               double dampening, double weight_decay, bool nesterov) {
+  const int T = params.size();
+  const bool has_m = momentum != 0.0;
+  TORCH_CHECK(!has_m || (int)bufs.size() == T,
+              "momentum needs one buffer per param");
+  std::vector<at::Tensor> out_p, out_buf;
+  out_p.reserve(T);
+  for (int t = 0; t < T; ++t) {
+    TORCH_CHECK(params[t].scalar_type() == at::kFloat, "fp32 params only");
+    TORCH_CHECK(params[t].is_contiguous() && grads[t].is_contiguous());
+    out_p.push_back(at::empty_like(params[t]));
+    if (has_m) out_buf.push_back(at::empty_like(bufs[t]));
+  }
+  std::vector<SgdChunk> chunks;
+  for (int t = 0; t < T; ++t) {
+    long n = params[t].numel();
+    for (long off = 0; off < n; off += CHUNK_ELEMS) {
+      SgdChunk c;
+      c.p = params[t].data_ptr<float>() + off;
+      c.g = grads[t].data_ptr<float>() + off;
+      c.buf = has_m ? bufs[t].data_ptr<float>() + off : nullptr;
+      c.out_p = out_p[t].data_ptr<float>() + off;
+      c.out_buf = has_m ? out_buf[t].data_ptr<float>() + off : nullptr;
+      c.n = (int)std::min<long>(CHUNK_ELEMS, n - off);
+      chunks.push_back(c);
+    }
+  }
+  static at::Tensor g_pinned, g_dev;
+  static size_t g_capacity = 0;
+  static hipEvent_t g_h2d_done = nullptr;
+  const size_t bytes = chunks.size() * sizeof(SgdChunk);
+  bool capturing = at::cuda::currentStreamCaptureStatusMayInitCtx() !=
+                   at::cuda::CaptureStatus::None;
+  if (bytes > g_capacity) {
+    TORCH_CHECK(!capturing,
+                "fused_sgd_step: warm up once eagerly before capture");
+    g_pinned = at::empty({(long)bytes},
+                         at::TensorOptions().dtype(at::kByte)
+                             .pinned_memory(true));
+    g_dev = at::empty({(long)bytes},
+                      at::TensorOptions().dtype(at::kByte)
+                          .device(params[0].device()));
+    g_capacity = bytes;
+  }
+  if (g_h2d_done == nullptr)
+    C10_CUDA_CHECK(hipEventCreateWithFlags(&g_h2d_done,
+                                           hipEventDisableTiming));
+  else if (!capturing)
+    C10_CUDA_CHECK(hipEventSynchronize(g_h2d_done));
+  memcpy(g_pinned.data_ptr(), chunks.data(), bytes);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  C10_CUDA_CHECK(hipMemcpyAsync(g_dev.data_ptr(), g_pinned.data_ptr(),
+                                bytes, hipMemcpyHostToDevice, stream));
+  if (!capturing)
+    C10_CUDA_CHECK(hipEventRecord(g_h2d_done, stream));
+  hipLaunchKernelGGL(fused_sgd_kernel, dim3(chunks.size()), dim3(256), 0,
+      stream, (const SgdChunk*)g_dev.data_ptr(), (int)chunks.size(),
+      (float)lr, (float)momentum, (float)dampening, (float)weight_decay,
+      nesterov ? 1 : 0);
+  if (capturing) {
+    static std::vector<at::Tensor> keepalive;
+    keepalive.push_back(g_pinned);
+    keepalive.push_back(g_dev);
+    g_pinned = at::Tensor();
+    g_dev = at::Tensor();
+    g_capacity = 0;
+  }
+  return {out_p, out_buf};
+}

@@ -66,3 +66,44 @@ def _adam_fake(params, grads, exp_avgs, exp_avg_sqs, steps, lr, beta1, beta2,
             [torch.empty_like(t) for t in exp_avgs],
             [torch.empty_like(t) for t in exp_avg_sqs],
             [torch.empty_like(s) for s in steps])
+
+
+lib.define("fused_sgd_step(Tensor[] params, Tensor[] grads, Tensor[] bufs, "
+           "float lr, float momentum, float dampening, float weight_decay, "
+           "bool nesterov) -> (Tensor[], Tensor[])")
+
+
+def _sgd_aten(params, grads, bufs, lr, momentum, dampening, weight_decay,
+              nesterov):
+    new_p, new_b = [], []
+    for i, (p, g) in enumerate(zip(params, grads)):
+        if weight_decay != 0:
+            g = g + weight_decay * p
+        if momentum != 0:
+            nb = momentum * bufs[i] + (1 - dampening) * g
+            new_b.append(nb)
+            g = g + momentum * nb if nesterov else nb
+        new_p.append(p - lr * g)
+    return new_p, new_b
+
+
+def _sgd_cuda(params, grads, bufs, lr, momentum, dampening, weight_decay,
+              nesterov):
+    ext = load_extension()
+    if ext is not None and all(p.dtype == torch.float32 for p in params):
+        return ext.fused_sgd_step(list(params), list(grads), list(bufs),
+                                  lr, momentum, dampening, weight_decay,
+                                  nesterov)
+    return _sgd_aten(params, grads, bufs, lr, momentum, dampening,
+                     weight_decay, nesterov)
+
+
+lib.impl("fused_sgd_step", _sgd_aten, "CPU")
+lib.impl("fused_sgd_step", _sgd_cuda, "CUDA")
+
+
+@torch.library.register_fake("easydist_amd::fused_sgd_step")
+def _sgd_fake(params, grads, bufs, lr, momentum, dampening, weight_decay,
+              nesterov):
+    return ([torch.empty_like(p) for p in params],
+            [torch.empty_like(b) for b in bufs] if momentum != 0 else [])

@@ -133,3 +133,44 @@ def _sharded_fuse_body(world_size):
 def test_fused_adam_sharded_states_ws2():
     from easydist_amd.utils.testing import spawn
     spawn(_sharded_fuse_body, args=(2,), world_size=2, port=29631)
+
+
+def _run_sgd(make_opt, caplog):
+    init_single_process()
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([0], ["spmd0"])
+    torch.manual_seed(0)
+    model = _Net()
+    model_ref = copy.deepcopy(model)
+    opt = make_opt(model.parameters())
+    opt_ref = make_opt(model_ref.parameters())
+    from easydist_amd import easydist_compile as edc
+    compiled = edc(_step, cuda_graph=False)
+    torch.manual_seed(7)
+    with caplog.at_level(
+            logging.INFO,
+            logger="easydist_amd.compiler.passes.fuse_optimizer"):
+        for i in range(3):
+            x = torch.randn(8, 16)
+            y = torch.randn(8, 16)
+            loss = compiled(model, opt, x, y)
+            ref = _step(model_ref, opt_ref, x, y)
+            assert abs(float(loss) - float(ref)) < 1e-5, \
+                (i, float(loss), float(ref))
+    assert any("SGD chains" in r.message for r in caplog.records), \
+        [r.message for r in caplog.records]
+    live = dict(compiled.named_parameters())
+    for n, pr in model_ref.named_parameters():
+        assert torch.allclose(live[n], pr, rtol=1e-5, atol=1e-6), \
+            (n, (live[n] - pr).abs().max())
+
+
+def test_fused_sgd_momentum(caplog):
+    _run_sgd(lambda ps: torch.optim.SGD(ps, lr=1e-2, momentum=0.9,
+                                        fused=True), caplog)
+
+
+def test_fused_sgd_momentum_wd_nesterov(caplog):
+    _run_sgd(lambda ps: torch.optim.SGD(ps, lr=1e-2, momentum=0.9,
+                                        weight_decay=0.01, nesterov=True,
+                                        fused=True), caplog)
