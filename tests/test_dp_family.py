@@ -126,3 +126,45 @@ def _gpt_zero_body(world_size, mode):
 def test_gpt_zero_ws2(mode):
     spawn(_gpt_zero_body, args=(2, mode), world_size=2,
           port=29547 + (mode == "zero3"))
+
+
+def _ddp_sgd_body(ws):
+    """ddp mode with the fused SGD optimizer (momentum): grads average
+    across ranks, trajectory matches single-process full-batch SGD."""
+    import copy
+
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(ws)), ["spmd0"])
+    torch.manual_seed(0)
+    model = MLP()
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.SGD(model.parameters(), lr=1e-2, momentum=0.9,
+                          fused=True)
+    opt_ref = torch.optim.SGD(model_ref.parameters(), lr=1e-2,
+                              momentum=0.9, fused=True)
+    compiled = easydist_compile(train_step, parallel_mode="ddp",
+                                cuda_graph=False)
+    torch.manual_seed(7)
+    for i in range(3):
+        # dp contract: every rank passes the GLOBAL batch; the compiled
+        # wrapper chunks it per rank and all-reduce(avg)s grads + loss
+        x = torch.randn(8, 16)
+        y = torch.randn(8, 16)
+        dist.broadcast(x, src=0)
+        dist.broadcast(y, src=0)
+        loss = compiled(model, opt, x, y)
+        ref = train_step(model_ref, opt_ref, x, y)
+        assert abs(float(loss) - float(ref)) < 1e-5, \
+            (i, float(loss), float(ref))
+
+
+@pytest.mark.world2
+def test_ddp_sgd_momentum_ws2():
+    spawn(_ddp_sgd_body, args=(2,), world_size=2, port=29642)
