@@ -26,3 +26,21 @@ def test_fused_adam_weight_decay_gpu():
         for a, b in zip(out[k], ref[k]):
             assert torch.allclose(a.float(), b.float(), rtol=1e-5,
                                   atol=1e-6), (k, (a - b).abs().max())
+
+
+def test_fused_sgd_gpu():
+    from easydist_amd.ops import optim
+    torch.manual_seed(0)
+    dev = "cuda"
+    ps = [torch.randn(1000, device=dev), torch.randn(257, device=dev)]
+    gs = [torch.randn_like(p) for p in ps]
+    bs = [torch.randn_like(p) for p in ps]
+    out = torch.ops.easydist_amd.fused_sgd_step(
+        ps, gs, bs, 1e-2, 0.9, 0.0, 0.01, True)
+    ref = optim._sgd_aten([p.clone() for p in ps], gs,
+                          [b.clone() for b in bs],
+                          1e-2, 0.9, 0.0, 0.01, True)
+    for k in range(2):
+        for a, b in zip(out[k], ref[k]):
+            assert torch.allclose(a, b, rtol=1e-5, atol=1e-6), \
+                (k, (a - b).abs().max())
