@@ -82,9 +82,24 @@ torch.library.register_autograd("easydist_amd::ce_fwd", _ce_backward,
                                 setup_context=_ce_setup)
 
 
-def cross_entropy(logits, targets):
+def cross_entropy(logits, targets, ignore_index=None):
     """Mean cross-entropy via the SUM kernel: the division by the global
     token count is a traced scalar, so a row-sharded graph keeps the sum
-    PARTIAL until the output reshard's single all_reduce."""
-    loss_sum, _ = torch.ops.easydist_amd.ce_fwd(logits, targets)
-    return loss_sum / logits.shape[0]
+    PARTIAL until the output reshard's single all_reduce.
+
+    ``ignore_index`` (e.g. -100 padding labels) routes through a masked
+    formulation: ignored rows are remapped to class 0 for the kernel,
+    their NLL contribution zeroed, and the mean divides by the VALID
+    count (matching ``F.cross_entropy``'s semantics). Sum and count stay
+    PARTIAL under row sharding — still one scalar all-reduce each.
+    """
+    if ignore_index is None:
+        loss_sum, _ = torch.ops.easydist_amd.ce_fwd(logits, targets)
+        return loss_sum / logits.shape[0]
+    valid = targets != ignore_index
+    safe_t = torch.where(valid, targets, torch.zeros_like(targets))
+    lf = logits.float()
+    lse = torch.logsumexp(lf, dim=-1)
+    nll = lse - lf.gather(1, safe_t.unsqueeze(1)).squeeze(1)
+    loss_sum = (nll * valid.to(nll.dtype)).sum()
+    return loss_sum / valid.to(nll.dtype).sum().clamp_min(1)

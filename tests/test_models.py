@@ -144,3 +144,17 @@ def test_vit_compiled_golden():
         ref = vit_train_step(model_ref, opt_ref, x, y)
         assert abs(float(loss) - float(ref)) < 1e-4, \
             (i, float(loss), float(ref))
+
+
+def test_fused_ce_ignore_index():
+    from easydist_amd.ops.ce import cross_entropy
+    torch.manual_seed(0)
+    lg = torch.randn(32, 11)
+    tg = torch.randint(0, 11, (32,))
+    tg[::5] = -100
+    ref = torch.nn.functional.cross_entropy(lg, tg, ignore_index=-100)
+    got = cross_entropy(lg, tg, ignore_index=-100)
+    assert abs(float(ref) - float(got)) < 1e-6
+    # all-ignored: defined (0), no div-by-zero
+    tg_all = torch.full((32,), -100)
+    assert float(cross_entropy(lg, tg_all, ignore_index=-100)) == 0.0
