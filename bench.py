@@ -20,6 +20,24 @@ import time
 import torch
 
 
+def _comm_summary(compiled):
+    """Count the collectives the solver's strategy put in ONE step —
+    the honest description of what 'auto' chose at this world size."""
+    try:
+        rts = list(getattr(compiled, "compiled", {}).values())
+        if not rts or not hasattr(rts[0], "gm"):
+            return None
+        from collections import Counter
+        cnt = Counter()
+        for n in rts[0].gm.graph.nodes:
+            t = getattr(n.target, "__name__", "")
+            if t.startswith("rt_") and t.endswith("_start"):
+                cnt[t[3:-6]] += 1
+        return dict(cnt) or None
+    except Exception:
+        return None
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -190,6 +208,7 @@ def main():
                 "global_batch": global_batch,
                 "seq_len": args.seq,
                 "parallelism": f"{args.parallel}(dp{n_gpus})",
+                "comm_per_step": _comm_summary(compiled),
                 "strategy_search_s": search_s,
                 "compile_warmup_s": round(compile_and_warmup_s, 2),
                 "loss": float(loss) if loss is not None else None,
