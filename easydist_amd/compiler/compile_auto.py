@@ -71,8 +71,9 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     gm, out_pl_env, search_time, solve_time = shard_graph(
         gm, mesh, io_map, ret_names, device, n_state=n_state)
 
-    # ---- 5b. comm overlap: widen every start/wait window -----------------
-    from .passes.comm_optimize import comm_optimize
+    # ---- 5b. comm dedup + overlap ----------------------------------------
+    from .passes.comm_optimize import comm_cse, comm_optimize
+    comm_cse(gm)
     comm_optimize(gm)
 
     # ---- 5b2. lower hot aten ops to the gfx950 kernels -------------------

@@ -44,3 +44,32 @@ def test_sink_and_raise():
     assert idx[wait.name] == idx[consumer.name] - 1, names
     # the x1/x2/x3 compute now lives inside the start..wait window
     assert idx[start.name] < idx["relu"] or idx[start.name] < idx["relu_2"]
+
+
+def test_comm_cse_dedupes_identical_collectives():
+    import torch.fx as fx
+
+    from easydist_amd.compiler.passes.comm_optimize import comm_cse
+    from easydist_amd.runtime import comm_runtime as crt
+
+    g = fx.Graph()
+    x = g.placeholder("x")
+    s1 = g.call_function(crt.rt_all_gather_start, (x, 0, 0))
+    w1 = g.call_function(crt.rt_wait, (s1,))
+    a = g.call_function(torch.relu, (w1,))
+    s2 = g.call_function(crt.rt_all_gather_start, (x, 0, 0))  # duplicate
+    w2 = g.call_function(crt.rt_wait, (s2,))
+    b = g.call_function(torch.tanh, (w2,))
+    # different args: NOT a duplicate
+    s3 = g.call_function(crt.rt_all_gather_start, (x, 1, 0))
+    w3 = g.call_function(crt.rt_wait, (s3,))
+    g.output((a, b, w3))
+    gm = fx.GraphModule(torch.nn.Module(), g)
+    removed = comm_cse(gm)
+    assert removed == 1
+    starts = [n for n in gm.graph.nodes
+              if n.target is crt.rt_all_gather_start]
+    assert len(starts) == 2
+    # tanh now consumes the FIRST wait's result
+    tanh = next(n for n in gm.graph.nodes if n.target is torch.tanh)
+    assert tanh.args[0].args[0] is starts[0]
