@@ -131,3 +131,45 @@ def _gpt_ws4_body(world_size):
 @pytest.mark.world4
 def test_gpt_auto_ws4_midsize():
     spawn(_gpt_ws4_body, args=(4,), world_size=4, port=29638)
+
+
+def _gpt_ws8_body(world_size):
+    """ws8 (the driver's largest bench N) on a 2-layer GPT-2-small
+    geometry: golden vs vanilla."""
+    import copy
+    from dataclasses import replace
+
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+    from easydist_amd.models.gpt import GPT, GPT2_SMALL, gpt_train_step
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+    torch.manual_seed(0)
+    cfg = replace(GPT2_SMALL, n_layer=2, block_size=128)
+    model = GPT(cfg)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=True)
+    opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-4,
+                               fused=True)
+    compiled = easydist_compile(gpt_train_step, cuda_graph=False)
+    torch.manual_seed(5)
+    for i in range(2):
+        idx = torch.randint(0, cfg.vocab_size, (8, 128))
+        tg = torch.randint(0, cfg.vocab_size, (8, 128))
+        dist.broadcast(idx, src=0)
+        dist.broadcast(tg, src=0)
+        loss = compiled(model, opt, idx, tg)
+        ref = gpt_train_step(model_ref, opt_ref, idx, tg)
+        assert abs(float(loss) - float(ref)) < 2e-2, \
+            (i, float(loss), float(ref))
+
+
+@pytest.mark.world8
+def test_gpt_auto_ws8():
+    spawn(_gpt_ws8_body, args=(8,), world_size=8, port=29648,
+          timeout=900.0)
