@@ -247,11 +247,74 @@ class AutoFlowSolver1D:
                 new_beam.append((cost, na, npl))
             beam = new_beam
         best = beam[0]
-        logger.info("beam search: %d clusters, width %d, cost %.3e",
-                    len(self.clusters), width, best[0])
-        # follow-edge costs applied post-hoc are already included via
-        # output_constraints unary terms; good enough for the fallback.
-        return {c.name: c.strategies[best[1][c.name]] for c in self.clusters}
+        assign = {c.name: best[1][c.name] for c in self.clusters}
+        assign, final_cost = self._refine(assign)
+        logger.info("beam search: %d clusters, width %d, cost %.3e "
+                    "(refined %.3e)", len(self.clusters), width, best[0],
+                    final_cost)
+        return {c.name: c.strategies[assign[c.name]]
+                for c in self.clusters}
+
+    def _refine(self, assign):
+        """Deterministic coordinate descent over the beam assignment:
+        re-pick each cluster's argmin strategy holding its neighbors
+        fixed (sweeps in topo order until a fixpoint). Repairs the beam's
+        greedy myopia; monotonically non-increasing objective, identical
+        on every rank."""
+        from collections import defaultdict
+        cl_by_name = {c.name: c for c in self.clusters}
+        touching = defaultdict(list)   # cluster name -> (cp, cc, v, nbytes)
+        for e in self._edges():
+            touching[e[0].name].append(e)
+            touching[e[1].name].append(e)
+        follow = defaultdict(list)     # cluster name -> (cp, cc, m, side)
+        for cp, cc, m in self._follow_edges():
+            follow[cp.name].append((cp, cc, m, 0))
+            follow[cc.name].append((cp, cc, m, 1))
+        unary = {c.name: self._unary_costs(c) for c in self.clusters}
+        var_bytes = {name: v.nbytes for name, v in self.graph.vars.items()}
+
+        def local_cost(c, s):
+            cost = float(unary[c.name][s])
+            for cp, cc, v, nb in touching[c.name]:
+                if cp.name == c.name:
+                    su, sv = c.strategies[s], \
+                        cc.strategies[assign[cc.name]]
+                else:
+                    su, sv = cp.strategies[assign[cp.name]], \
+                        c.strategies[s]
+                pv = sv.in_placements.get(v)
+                if pv is not None:
+                    cost += reshard_cost(su.out_placements.get(v, R), pv,
+                                         nb, self.mesh_size)
+            for cp, cc, m, side in follow[c.name]:
+                i = s if side == 0 else assign[cp.name]
+                j = assign[cc.name] if side == 0 else s
+                cost += float(m[i, j])
+            return cost
+
+        for _sweep in range(3):
+            changed = False
+            for c in self.clusters:
+                cur = assign[c.name]
+                costs = [local_cost(c, s) for s in range(len(c.strategies))]
+                b = min(range(len(costs)), key=lambda k: (costs[k], k))
+                if costs[b] + 1e-15 < costs[cur]:
+                    assign[c.name] = b
+                    changed = True
+            if not changed:
+                break
+        total = sum(float(unary[c.name][assign[c.name]])
+                    for c in self.clusters)
+        for cp, cc, v, nb in self._edges():
+            su = cp.strategies[assign[cp.name]]
+            pv = cc.strategies[assign[cc.name]].in_placements.get(v)
+            if pv is not None:
+                total += reshard_cost(su.out_placements.get(v, R), pv,
+                                      var_bytes.get(v, 0), self.mesh_size)
+        for cp, cc, m in self._follow_edges():
+            total += float(m[assign[cp.name], assign[cc.name]])
+        return assign, total
 
 
 def solve_mesh_dim(graph: MetaGraph, mesh_size: int, already_sharded=None,
