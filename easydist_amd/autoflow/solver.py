@@ -196,17 +196,22 @@ class AutoFlowSolver1D:
                             "mip_rel_gap": 1e-4})
         if res.x is None:
             raise RuntimeError(f"milp: {res.message}")
-        choice: Dict[str, ClusterStrategy] = {}
+        assign: Dict[str, int] = {}
         for c in self.clusters:
             best_s, best_v = 0, -1.0
             for s in range(len(c.strategies)):
                 xv = res.x[x_index[(c.name, s)]]
                 if xv > best_v:
                     best_v, best_s = xv, s
-            choice[c.name] = c.strategies[best_s]
-        logger.info("AutoFlow MILP: %d clusters, %d vars, %.2fs, obj=%.3e",
-                    len(self.clusters), nvar, time.time() - t0, res.fun)
-        return choice
+            assign[c.name] = best_s
+        # local refinement: a no-op on converged solves, but repairs a
+        # time-limited incumbent (monotone, deterministic)
+        assign, refined = self._refine(assign)
+        logger.info("AutoFlow MILP: %d clusters, %d vars, %.2fs, obj=%.3e "
+                    "(refined %.3e)", len(self.clusters), nvar,
+                    time.time() - t0, res.fun, refined)
+        return {c.name: c.strategies[assign[c.name]]
+                for c in self.clusters}
 
     # --------------------------------------------------------- beam search ---
     def beam_search(self) -> Dict[str, ClusterStrategy]:
