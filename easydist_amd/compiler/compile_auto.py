@@ -101,8 +101,9 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
                 gm.graph.lint()
                 gm.recompile()
 
-    from ..utils.dumps import dump_graph
+    from ..utils.dumps import dump_graph, dump_graph_dot
     dump_graph(gm, "auto_sharded")
+    dump_graph_dot(gm, "auto_sharded")
 
     # strip the pytree codegen: the runtime calls the graph with the flat
     # placeholder list and receives the flat output list

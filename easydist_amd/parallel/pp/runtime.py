@@ -97,7 +97,10 @@ class StageExecutor:
     def run_fw(self, m: int, data_chunks) -> None:
         args = [self._lookup(n, m, data_chunks) for n in self.sg.fw_inputs]
         try:
-            outs = self.sg.fw_gm(*args)
+            # torch.profiler visibility (reference: compile_pipeline.py:388)
+            with torch.profiler.record_function(
+                    f"pp_stage{self.sg.stage_idx}_fw_mb{m}"):
+                outs = self.sg.fw_gm(*args)
         except Exception:
             import torch as _t
             shapes = {n: (tuple(a.shape) if isinstance(a, _t.Tensor)
@@ -115,7 +118,9 @@ class StageExecutor:
 
     def run_bw(self, m: int, data_chunks) -> None:
         args = [self._lookup(n, m, data_chunks) for n in self.sg.bw_inputs]
-        outs = self.sg.bw_gm(*args)
+        with torch.profiler.record_function(
+                f"pp_stage{self.sg.stage_idx}_bw_mb{m}"):
+            outs = self.sg.bw_gm(*args)
         for name, val in zip(self.sg.bw_outputs, outs):
             self.stash[m][name] = val
             if name in self.sg.grad_names:
@@ -131,7 +136,9 @@ class StageExecutor:
         for name, g in self.grad_acc.items():
             env[name] = g.div_(nchunks) if scale_grads else g
         args = [env[n] for n in self.sg.step_inputs]
-        outs = self.sg.step_gm(*args)  # in-place (copy_) OR functional
+        with torch.profiler.record_function(
+                f"pp_stage{self.sg.stage_idx}_step"):
+            outs = self.sg.step_gm(*args)  # in-place (copy_) OR functional
         if self.sg.writeback:
             by_name = dict(zip(self.sg.step_outputs, outs))
             for ph, src in self.sg.writeback.items():
