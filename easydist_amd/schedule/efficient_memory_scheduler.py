@@ -107,3 +107,34 @@ def plan_memory(gm: fx.GraphModule, mem_info: GraphMemInfo):
                 "(%.0f%% saved)", len(entries), peak / 2**20, naive / 2**20,
                 stats["savings"] * 100)
     return entries, peak, stats
+
+
+def plot_plan(buffers, addresses, peak, path):
+    """Address-vs-time rectangle plot of a packed plan (reference
+    observability: efficient_memory_scheduler.py:385-405 matplotlib
+    memory plots). Each buffer is a rectangle [start,end] x
+    [offset, offset+size]."""
+    import matplotlib
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+    from matplotlib.patches import Rectangle
+
+    fig, ax = plt.subplots(figsize=(10, 6))
+    for b in buffers:
+        off = addresses.get((b.node_name, b.alloc_idx))
+        if off is None:
+            continue
+        ax.add_patch(Rectangle((b.start, off / 2**20),
+                               max(b.end - b.start, 0.5),
+                               b.size / 2**20,
+                               alpha=0.6, edgecolor="black",
+                               linewidth=0.3))
+    ax.set_xlim(0, max((b.end for b in buffers), default=1) + 1)
+    ax.set_ylim(0, peak / 2**20 * 1.05)
+    ax.set_xlabel("op index (lifetime)")
+    ax.set_ylabel("arena offset (MiB)")
+    ax.set_title(f"static memory plan — peak {peak / 2**20:.1f} MiB, "
+                 f"{len(buffers)} buffers")
+    fig.tight_layout()
+    fig.savefig(path, dpi=110)
+    plt.close(fig)

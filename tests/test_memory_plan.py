@@ -68,3 +68,16 @@ def test_ilp_pack_at_least_as_good_as_skyline():
             if _overlap_time(x, y):
                 assert not _overlap_addr(addr_i[(x.node_name, 0)], x,
                                          addr_i[(y.node_name, 0)], y)
+
+
+def test_plot_plan(tmp_path):
+    from easydist_amd.schedule.efficient_memory_scheduler import (
+        pack_buffers, plot_plan)
+    from easydist_amd.schedule.lifetime import Buffer
+    bufs = [Buffer(node_name=f"b{i}", alloc_idx=0,
+                   size=(i % 5 + 1) * 1024, start=i, end=i + 4,
+                   is_temp=False) for i in range(20)]
+    addr, peak = pack_buffers(bufs)
+    out = tmp_path / "plan.png"
+    plot_plan(bufs, addr, peak, str(out))
+    assert out.stat().st_size > 1000
