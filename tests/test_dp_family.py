@@ -168,3 +168,43 @@ def _ddp_sgd_body(ws):
 @pytest.mark.world2
 def test_ddp_sgd_momentum_ws2():
     spawn(_ddp_sgd_body, args=(2,), world_size=2, port=29642)
+
+
+def _zero2_adamw_body(ws):
+    """zero2 with AdamW: FLAT-sharded moments must follow the decoupled-
+    decay math; golden vs single-process full batch."""
+    import copy
+
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(ws)), ["spmd0"])
+    torch.manual_seed(0)
+    model = MLP()
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-2, weight_decay=0.05,
+                            fused=True)
+    opt_ref = torch.optim.AdamW(model_ref.parameters(), lr=1e-2,
+                                weight_decay=0.05, fused=True)
+    compiled = easydist_compile(train_step, parallel_mode="zero2",
+                                cuda_graph=False)
+    torch.manual_seed(7)
+    for i in range(3):
+        x = torch.randn(8, 16)
+        y = torch.randn(8, 16)
+        dist.broadcast(x, src=0)
+        dist.broadcast(y, src=0)
+        loss = compiled(model, opt, x, y)
+        ref = train_step(model_ref, opt_ref, x, y)
+        assert abs(float(loss) - float(ref)) < 1e-5, \
+            (i, float(loss), float(ref))
+
+
+@pytest.mark.world2
+def test_zero2_adamw_ws2():
+    spawn(_zero2_adamw_body, args=(2,), world_size=2, port=29656)

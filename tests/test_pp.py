@@ -162,3 +162,47 @@ def test_pp_nchunks_edge_cases():
         _golden_loop(MLP4,
                      lambda m: torch.optim.Adam(m.parameters(), lr=1e-2),
                      dict(split_points={"fc2"}, nchunks=nchunks), steps=2)
+
+
+def test_pp_tied_weights_error_is_loud():
+    """Tied weights spanning a split (BERT-style shared embedding/decoder)
+    are not supported by pipeline mode: the compile must fail with an
+    explanatory error instead of silently training wrong."""
+    import pytest as _pytest
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+    from easydist_amd.utils.testing import init_single_process
+
+    init_single_process()
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([0], ["pp"])
+
+    class Tied(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb = nn.Embedding(16, 8)
+            self.mid = nn.Linear(8, 8)
+            self.head = nn.Linear(8, 16, bias=False)
+            self.head.weight = self.emb.weight     # tied across the split
+
+        def forward(self, idx):
+            return self.head(torch.relu(self.mid(self.emb(idx))))
+
+    model = Tied()
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+
+    def step(model, opt, idx, tg):
+        loss = torch.nn.functional.cross_entropy(model(idx), tg)
+        loss.backward()
+        opt.step()
+        opt.zero_grad(True)
+        return loss
+
+    compiled = easydist_compile(step, parallel_mode="pp",
+                                cuda_graph=False, split_points={"mid"},
+                                nchunks=1)
+    idx = torch.randint(0, 16, (4,))
+    tg = torch.randint(0, 16, (4,))
+    with _pytest.raises(RuntimeError, match="multiple pipeline stages"):
+        compiled(model, opt, idx, tg)
