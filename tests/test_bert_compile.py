@@ -85,3 +85,8 @@ def test_bert_train_mode_dropout():
     losses = [float(compiled(model, opt, ids, ids)) for _ in range(6)]
     assert all(torch.isfinite(torch.tensor(losses))), losses
     assert losses[-1] < losses[0], losses
+
+
+@pytest.mark.world4
+def test_bert_compile_ws4():
+    spawn(_golden_body, args=(4,), world_size=4, port=29661)

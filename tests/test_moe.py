@@ -170,3 +170,8 @@ def test_moe_auto_ws1():
 @pytest.mark.world2
 def test_moe_auto_ws2():
     spawn(_moe_auto_body, args=(2,), world_size=2, port=29565)
+
+
+@pytest.mark.world4
+def test_moe_auto_ws4():
+    spawn(_moe_auto_body, args=(4,), world_size=4, port=29662)
