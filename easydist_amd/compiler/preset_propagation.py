@@ -878,3 +878,81 @@ def _maxpool_bwd_rule(input_shapes, args, kwargs):
             combs[sid] = _gather(d)
             sid += 1
     return ann, combs
+
+
+# ----------------------------------------------------------- softmax --------
+@register_preset(aten._softmax.default, aten._log_softmax.default,
+                 aten.softmax.int, aten.log_softmax.int)
+def _softmax_rule(input_shapes, args, kwargs):
+    # (x, dim, ...): every dim except the normalized one shards
+    shape = input_shapes[0]
+    r = len(shape)
+    sdim = args[1] % r if shape else 0
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(r):
+        if d == sdim or shape[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten._softmax_backward_data.default,
+                 aten._log_softmax_backward_data.default)
+def _softmax_bwd_rule(input_shapes, args, kwargs):
+    # (gout, out, dim, dtype)
+    shape = input_shapes[0]
+    r = len(shape)
+    sdim = args[2] % r if shape else 0
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(r):
+        if d == sdim or shape[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        ann[1][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.tril.default, aten.triu.default)
+def _tri_rule(input_shapes, args, kwargs):
+    # triangular masks act on the LAST TWO dims; batch dims shard
+    shape = input_shapes[0]
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(max(0, len(shape) - 2)):
+        if shape[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs
+
+
+@register_preset(aten.constant_pad_nd.default)
+def _pad_rule(input_shapes, args, kwargs):
+    # pad spec covers trailing dims (reversed pairs); unpadded dims shard
+    shape = input_shapes[0]
+    r = len(shape)
+    pad = list(args[1])
+    padded = set()
+    for i in range(len(pad) // 2):
+        if pad[2 * i] or pad[2 * i + 1]:
+            padded.add(r - 1 - i)
+    ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+    combs = {}
+    sid = 1
+    for d in range(r):
+        if d in padded or shape[d] <= 1:
+            continue
+        ann[0][d] = ShardDim.get_shard_dim(sid)
+        combs[sid] = _gather(d)
+        sid += 1
+    return ann, combs

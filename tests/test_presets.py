@@ -195,3 +195,24 @@ def test_embedding():
     w = T(16, 8, dtype=torch.float64)
     idx = torch.randint(0, 16, (4, 6))
     _run_case(aten.embedding.default, (w, idx), min_rules=3)
+
+
+def test_softmax_family():
+    torch.manual_seed(0)
+    _run_case(aten._softmax.default, (T(4, 6, 8, dtype=torch.float64), -1,
+                                      False), min_rules=2)
+    _run_case(aten._log_softmax.default, (T(4, 8, dtype=torch.float64), 1,
+                                          False), min_rules=1)
+    x = T(4, 6, 8, dtype=torch.float64)
+    out = aten._softmax.default(x, -1, False)
+    g = torch.randn_like(out)
+    _run_case(aten._softmax_backward_data.default,
+              (g, out, -1, torch.float64), min_rules=2)
+
+
+def test_tri_and_pad():
+    torch.manual_seed(0)
+    _run_case(aten.tril.default, (T(4, 8, 8, dtype=torch.float64),),
+              min_rules=1)
+    _run_case(aten.constant_pad_nd.default,
+              (T(4, 6, 8, dtype=torch.float64), [1, 1], 0.0), min_rules=2)
