@@ -158,3 +158,44 @@ def test_fused_ce_ignore_index():
     # all-ignored: defined (0), no div-by-zero
     tg_all = torch.full((32,), -100)
     assert float(cross_entropy(lg, tg_all, ignore_index=-100)) == 0.0
+
+
+def _vit_ws_body(world_size):
+    """ViT through auto-SPMD at world>1 (conv patch embed + attention +
+    class-token cat): golden vs vanilla."""
+    import copy
+
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, \
+        set_device_mesh
+    from easydist_amd.models.vit import ViT, ViTConfig, vit_train_step
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+    torch.manual_seed(0)
+    cfg = ViTConfig(image_size=32, patch_size=8, n_layer=2, n_head=2,
+                    n_embd=32, n_classes=10)
+    model = ViT(cfg)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    model_ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=True)
+    opt_ref = torch.optim.Adam(model_ref.parameters(), lr=1e-3, fused=True)
+    compiled = easydist_compile(vit_train_step, cuda_graph=False)
+    torch.manual_seed(3)
+    for i in range(2):
+        x = torch.randn(8, 3, 32, 32)
+        y = torch.randint(0, 10, (8,))
+        dist.broadcast(x, src=0)
+        dist.broadcast(y, src=0)
+        loss = compiled(model, opt, x, y)
+        ref = vit_train_step(model_ref, opt_ref, x, y)
+        assert abs(float(loss) - float(ref)) < 5e-3, \
+            (i, float(loss), float(ref))
+
+
+@pytest.mark.world2
+def test_vit_auto_ws2():
+    from easydist_amd.utils.testing import spawn
+    spawn(_vit_ws_body, args=(2,), world_size=2, port=29666)
