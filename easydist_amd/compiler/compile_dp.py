@@ -249,6 +249,25 @@ def _transform_zero(graph: fx.Graph, gm, opt_node: fx.Node, flat_outs,
         new_grads.append(rs)
     fused_args[1] = new_grads
 
+    # user-returned grads: without this, the step's grad outputs are full
+    # unreduced per-rank partials (rank-divergent, unlike the ddp path's
+    # averaged grads). Rebuild the averaged full grad from the
+    # reduce-scattered shard with one all-gather per RETURNED grad only.
+    rs_of = dict(zip(grad_nodes, new_grads))
+    regathered: Dict[fx.Node, fx.Node] = {}
+    for i in grad_slice:
+        g = flat_outs[i]
+        if not (isinstance(g, fx.Node) and g in rs_of):
+            continue
+        if g not in regathered:
+            val = g.meta.get("val") if hasattr(g, "meta") else None
+            if not isinstance(val, torch.Tensor):
+                continue
+            with graph.inserting_after(rs_of[g]):
+                regathered[g] = graph.call_function(
+                    dp_gather_flat, (rs_of[g], tuple(val.shape)))
+        flat_outs[i] = regathered[g]
+
     new_params = []
     gathered: Dict[fx.Node, fx.Node] = {}
     for p in param_nodes:

@@ -35,6 +35,14 @@ def sink_waits_raise_starts(gm: fx.GraphModule) -> int:
     graph = gm.graph
     moved = 0
 
+    # One order snapshot per phase (O(n)), not one per move (O(n^2) on the
+    # multi-thousand-node whole-step traces this runs on). Moving a start
+    # never reorders two NON-moved nodes relative to each other, so stale
+    # indices are only a correctness risk when a start's producer is
+    # itself a start moved earlier in this phase — rebuild just then.
+    # Adjacency ("already in place") is read off the node linked list.
+    moved_this_phase: set = set()
+
     # raise starts: insert each start right after its last-placed producer
     order = {n: i for i, n in enumerate(graph.nodes)}
     for n in list(graph.nodes):
@@ -43,28 +51,30 @@ def sink_waits_raise_starts(gm: fx.GraphModule) -> int:
         producers = [a for a in n.all_input_nodes]
         if not producers:
             continue
+        if moved_this_phase.intersection(producers):
+            order = {x: i for i, x in enumerate(graph.nodes)}
+            moved_this_phase.clear()
         anchor = max(producers, key=lambda p: order[p])
-        # already directly after its producer?
-        if order[n] == order[anchor] + 1:
+        if anchor.next is n:      # already directly after its producer
             continue
         anchor.append(n)          # move n to directly after anchor
         moved += 1
-        order = {x: i for i, x in enumerate(graph.nodes)}
+        moved_this_phase.add(n)
 
-    # sink waits: place each wait right before its first consumer
+    # sink waits: place each wait right before its first consumer. Waits
+    # are never another wait's input, so one snapshot suffices.
     order = {n: i for i, n in enumerate(graph.nodes)}
     for n in list(graph.nodes):
         if not _is_wait(n):
             continue
-        users = sorted(n.users, key=lambda u: order[u])
+        users = [u for u in n.users]
         if not users:
             continue
-        first = users[0]
-        if order[n] == order[first] - 1:
+        first = min(users, key=lambda u: order[u])
+        if n.next is first:       # already directly before first consumer
             continue
         first.prepend(n)
         moved += 1
-        order = {x: i for i, x in enumerate(graph.nodes)}
 
     if moved:
         graph.lint()

@@ -57,6 +57,25 @@ def _strip_l2(g, p_ph, wd):
     return None
 
 
+def _is_scaled_square(n, g) -> bool:
+    """Match s·g·g in any association: mul(mul(g, s), g), mul(g, mul(g, s))
+    or mul(mul(g, g), s)."""
+    if not (isinstance(n, fx.Node) and n.op == "call_function"
+            and n.target is aten.mul.Tensor):
+        return False
+    a, b = n.args
+    for x, y in ((a, b), (b, a)):
+        if y is g:
+            t, s = _tensor_scalar(x)
+            if t is g and s is not None:
+                return True
+        if isinstance(y, (int, float)) and isinstance(x, fx.Node) \
+                and x.op == "call_function" and x.target is aten.mul.Tensor \
+                and x.args[0] is g and x.args[1] is g:
+            return True
+    return False
+
+
 def _match_param_chain(p_ph, ea_ph, eas_ph, step_ph, p_new, ea_new,
                        eas_new, step_new, wd=0.0, decay_scale=None):
     """Verify the decomposed Adam/AdamW shape; return the grad node.
@@ -84,13 +103,29 @@ def _match_param_chain(p_ph, ea_ph, eas_ph, step_ph, p_new, ea_new,
         g = t0
     else:
         return None
+    # the eas chain squares the (possibly L2-decayed) grad node; keep it
+    # before unwrapping the decay for the return value
+    g_dec = g
     if wd != 0.0:
         g = _strip_l2(g, p_ph, wd)
         if g is None:
             return None
-    # eas_new = add(mul(eas_ph, b2), mul(mul(g, 1-b2), g))
+    # eas_new = add(mul(eas_ph, b2), mul(mul(g, 1-b2), g)) — verify BOTH
+    # operands, same as the ea chain: one term scales eas_ph, the other
+    # is the scaled square of the same grad node.
     if not (isinstance(eas_new, fx.Node)
             and eas_new.target is aten.add.Tensor):
+        return None
+    q0, q1 = eas_new.args
+    qt0, _ = _tensor_scalar(q0)
+    qt1, _ = _tensor_scalar(q1)
+    if qt0 is eas_ph:
+        g2 = q1
+    elif qt1 is eas_ph:
+        g2 = q0
+    else:
+        return None
+    if not _is_scaled_square(g2, g_dec):
         return None
     # p_new = sub(base, ...); base = p_ph (Adam) or mul(p_ph, 1-lr*wd)
     # (AdamW decomp always emits the mul, even at wd=0 -> scalar 1.0)

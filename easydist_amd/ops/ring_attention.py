@@ -8,23 +8,28 @@ requires one. MI355X-first design:
   each ring step overlaps one neighbor KV-block hop (7 xGMI p2p links
   make neighbor exchange ~free relative to the block attention GEMMs)
   with the local block-attention compute;
-* per-block attention reuses the gfx950 flash kernel (or the fp32 math
-  path on CPU); partial results merge by log-sum-exp accumulation, so
-  the result is EXACT attention, not an approximation;
+* per-block attention goes through the ``easydist_amd::flash_attention``
+  custom op — the gfx950 flash kernel on GPU (it returns the LSE the
+  merge needs), the fp32 math path on CPU; partial results merge by
+  log-sum-exp accumulation in fp32, so the result is EXACT attention,
+  not an approximation;
+* ring buffers travel in the SOURCE dtype (bf16 K/V hop = half the xGMI
+  traffic of an fp32 hop); only the dK/dV accumulators ride fp32;
 * causal masking with a sequence-ordered shard layout: KV blocks from
   earlier ranks attend fully, the own block causally, later blocks are
   skipped (their hop still happens to keep the ring in lockstep);
 * backward is the standard two-ring recompute: ring 1 accumulates dQ
-  locally, dK/dV accumulate into a rotating buffer that arrives back at
-  its owner after W hops.
+  locally, dK/dV accumulate into a rotating fp32 buffer that arrives
+  back at its owner after W hops.
 """
 from __future__ import annotations
 
-import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.distributed as dist
+
+from . import attention as _attention  # noqa: F401  (registers the ops)
 
 
 def _ring_exchange(t: torch.Tensor, group) -> torch.Tensor:
@@ -44,29 +49,23 @@ def _ring_exchange(t: torch.Tensor, group) -> torch.Tensor:
     return recv
 
 
+def _exchange_pair(a: torch.Tensor, b: torch.Tensor, group
+                   ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """One ring hop of two same-dtype tensors as a single message."""
+    st = _ring_exchange(torch.stack([a, b]), group)
+    return st[0], st[1]
+
+
 def _block_attn(q, k, v, causal_mode: str):
-    """One block attention returning (out, lse). causal_mode:
-    'full' | 'causal' | 'skip'."""
-    if causal_mode == "skip":
-        out = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
-        lse = torch.full(q.shape[:-1], float("-inf"), dtype=torch.float32,
-                         device=q.device)
-        return out, lse
-    scale = 1.0 / math.sqrt(q.shape[-1])
-    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
-    if causal_mode == "causal":
-        S, T = s.shape[-2], s.shape[-1]
-        mask = torch.ones(S, T, dtype=torch.bool, device=s.device).tril()
-        s = s.masked_fill(~mask, float("-inf"))
-    lse = torch.logsumexp(s, dim=-1)
-    p = torch.exp(s - lse.unsqueeze(-1))
-    p = torch.nan_to_num(p, nan=0.0)        # rows fully masked
-    out = torch.matmul(p, v.float())
-    return out, lse
+    """One block attention returning (out_f32, lse_f32). causal_mode:
+    'full' | 'causal' (callers handle 'skip')."""
+    out, lse = torch.ops.easydist_amd.flash_attention(
+        q, k, v, causal_mode == "causal")
+    return out.float(), lse.float()
 
 
 def _merge(out, lse, out_b, lse_b):
-    """Log-sum-exp merge of two partial attention results."""
+    """Log-sum-exp merge of two partial attention results (fp32)."""
     new_lse = torch.logaddexp(lse, lse_b)
     a = torch.exp(lse - new_lse).unsqueeze(-1)
     b = torch.exp(lse_b - new_lse).unsqueeze(-1)
@@ -87,14 +86,13 @@ def _ring_forward(q, k, v, group, causal):
     w = dist.get_world_size(group)
     rank = dist.get_rank(group)
     out, lse = _block_attn(q, k, v, "causal" if causal else "full")
-    kv = torch.stack([k.float(), v.float()])
+    kb, vb = k, v
     for step in range(1, w):
-        kv = _ring_exchange(kv, group)
+        kb, vb = _exchange_pair(kb, vb, group)   # source dtype on the wire
         src = (rank - step) % w
         mode = _mode(src, rank, causal)
         if mode != "skip":
-            out_b, lse_b = _block_attn(q, kv[0].to(q.dtype),
-                                       kv[1].to(q.dtype), mode)
+            out_b, lse_b = _block_attn(q, kb, vb, mode)
             out, lse = _merge(out, lse, out_b, lse_b)
         # skipped blocks still traveled: ring stays in lockstep
     return out.to(q.dtype), lse
@@ -117,43 +115,32 @@ class _RingAttention(torch.autograd.Function):
         group, causal = ctx.group, ctx.causal
         w = dist.get_world_size(group)
         rank = dist.get_rank(group)
-        scale = 1.0 / math.sqrt(q.shape[-1])
-        gradf = grad.float()
-        qf = q.float()
-        # delta = rowsum(dO * O) — constant across KV blocks
-        delta = (gradf * out.float()).sum(-1, keepdim=True)
-
-        dq = torch.zeros_like(qf)
-        # rotating buffer: [k, v, dk, dv] — dk/dv accumulate as the block
-        # passes by and arrive home after the final hop
-        buf = torch.stack([k.float(), v.float(),
-                           torch.zeros_like(k, dtype=torch.float32),
-                           torch.zeros_like(v, dtype=torch.float32)])
+        # flash_attention_bwd's math (P from the GLOBAL lse, delta from
+        # the GLOBAL out) is exactly one ring block's contribution, so
+        # the per-block backward dispatches to the same op family as the
+        # forward (HIP kernel / batched hipBLASLt on GPU).
+        dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        kb, vb = k, v                       # rotating K/V, source dtype
+        dkv = torch.zeros((2,) + k.shape, dtype=torch.float32,
+                          device=k.device)  # rotating dK/dV, fp32
+        grad = grad.contiguous()
         with torch.no_grad():
             for step in range(w):
                 if step > 0:
-                    buf = _ring_exchange(buf, group)
+                    kb, vb = _exchange_pair(kb, vb, group)
+                    dkv = _ring_exchange(dkv, group)
                 src = (rank - step) % w
                 mode = _mode(src, rank, causal)
                 if mode != "skip":
-                    kf, vf = buf[0], buf[1]
-                    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
-                    if mode == "causal":
-                        S, T = s.shape[-2], s.shape[-1]
-                        m = torch.ones(S, T, dtype=torch.bool,
-                                       device=s.device).tril()
-                        s = s.masked_fill(~m, float("-inf"))
-                    p = torch.exp(s - lse.float().unsqueeze(-1))
-                    p = torch.nan_to_num(p, nan=0.0)
-                    dv_b = torch.matmul(p.transpose(-1, -2), gradf)
-                    dp = torch.matmul(gradf, vf.transpose(-1, -2))
-                    ds = p * (dp - delta) * scale
-                    dq += torch.matmul(ds, kf)
-                    buf[2] += torch.matmul(ds.transpose(-1, -2), qf)
-                    buf[3] += dv_b
+                    dq_b, dk_b, dv_b = torch.ops.easydist_amd.\
+                        flash_attention_bwd(grad, q, kb, vb, out, lse,
+                                            mode == "causal")
+                    dq += dq_b.float()
+                    dkv[0] += dk_b.float()
+                    dkv[1] += dv_b.float()
             # one final hop returns each block (with its grads) home
-            buf = _ring_exchange(buf, group)
-        return (dq.to(q.dtype), buf[2].to(k.dtype), buf[3].to(v.dtype),
+            dkv = _ring_exchange(dkv, group)
+        return (dq.to(q.dtype), dkv[0].to(k.dtype), dkv[1].to(v.dtype),
                 None, None)
 
 
