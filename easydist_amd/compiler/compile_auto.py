@@ -77,8 +77,10 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     comm_optimize(gm)
 
     # ---- 5b2. lower hot aten ops to the gfx950 kernels -------------------
-    from .passes.lower_hip import lower_layer_norm
+    from .passes.lower_hip import lower_gemm, lower_layer_norm
     lower_layer_norm(gm)
+    if mdconfig.hip_gemm:
+        lower_gemm(gm)
 
     # ---- 5c. re-fuse the decomposed Adam chains into ONE kernel ----------
     if opt is not None and getattr(mdconfig, "fuse_optimizer", True):

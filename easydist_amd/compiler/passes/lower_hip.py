@@ -23,6 +23,7 @@ import torch
 import torch.fx as fx
 
 from ...ops import norms as _norms  # noqa: F401  (registers the ops)
+from ...ops import gemm as _gemm    # noqa: F401
 
 logger = logging.getLogger(__name__)
 
@@ -112,6 +113,120 @@ def lower_layer_norm(gm: fx.GraphModule) -> int:
         graph.lint()
         gm.recompile()
         logger.info("lower_hip: lowered %d layer_norm nodes", n_lowered)
+    return n_lowered
+
+
+def _strip_t(n):
+    """Return x if n is aten.t(x), else None."""
+    if isinstance(n, fx.Node) and n.op == "call_function" \
+            and n.target is aten.t.default:
+        return n.args[0]
+    return None
+
+
+def _contig(n) -> bool:
+    v = _val(n)
+    return isinstance(v, torch.Tensor) and v.is_contiguous()
+
+
+def lower_gemm(gm: fx.GraphModule) -> int:
+    """Rewrite linear-layer matmuls to the hand-written MFMA GEMM ops.
+
+    Post-sharding graph patterns (north-star requirement: every
+    partitioned matmul on hand-written CDNA4 kernels; reference lowering
+    shape: easydist/torch/passes/sharding.py:852+):
+
+    * fwd:  addmm(bias, x, t(W)) / mm(x, t(W))      -> gemm_nt(x, W, bias)
+    * dX:   mm(dY, t(t(W))) (W weight, contiguous)  -> gemm_nt(dY, W^T)
+            with W^T materialized by ONE weight-sized transpose-copy —
+            a few microseconds against a 100x bigger activation GEMM
+    * dW:   mm(t(dY), X) (both reduce-dim-strided)  -> gemm_tn(dY, X)
+
+    Only shapes inside the kernel envelope are rewritten; everything else
+    stays on aten (hipBLASLt).  The ops themselves keep a per-shape
+    profiled fallback (ops/gemm.py).
+    """
+    graph = gm.graph
+    n_lowered = 0
+    gemm_nt = torch.ops.easydist_amd.gemm_nt.default
+    gemm_tn = torch.ops.easydist_amd.gemm_tn.default
+
+    def bf16_2d(n):
+        v = _val(n)
+        return (isinstance(v, torch.Tensor) and v.dtype == torch.bfloat16
+                and v.dim() == 2)
+
+    def nt_ok(a_val, x_val):
+        M, K = a_val.shape
+        N, K2 = x_val.shape
+        return K == K2 and K % 32 == 0 and N % 8 == 0 and M >= 16
+
+    def tn_ok(x_val, b_val):
+        R, P = x_val.shape
+        R2, Q = b_val.shape
+        return R == R2 and R % 32 == 0 and P % 128 == 0 and Q % 128 == 0
+
+    for n in list(graph.nodes):
+        if n.op != "call_function":
+            continue
+        if n.target is aten.addmm.default:
+            bias, a, b = n.args[:3]
+            if n.kwargs.get("alpha", 1) != 1 or n.kwargs.get("beta", 1) != 1:
+                continue
+        elif n.target is aten.mm.default:
+            a, b = n.args
+            bias = None
+        else:
+            continue
+        if not (bf16_2d(a) and bf16_2d(b) and bf16_2d(n)):
+            continue
+
+        new = None
+        x = _strip_t(b)
+        if x is not None and _contig(x) and _contig(a):
+            # NT: mm(a, t(x)) with x contiguous [N, K]
+            if nt_ok(_val(a), _val(x)):
+                with graph.inserting_before(n):
+                    new = graph.call_function(gemm_nt, (a, x, bias))
+        elif x is not None and _contig(a):
+            y = _strip_t(x)
+            if y is not None and _contig(y):
+                # NN via weight transpose: b = t(t(y)) = y [K, N]
+                y_val = _val(y)
+                if y_val is not None and y_val.shape[0] % 32 == 0 \
+                        and y_val.shape[1] % 8 == 0 and _val(a).shape[0] >= 16:
+                    with graph.inserting_before(n):
+                        yt = graph.call_function(aten.t.default, (y,))
+                        ytc = graph.call_function(
+                            aten.clone.default, (yt,),
+                            {"memory_format": torch.contiguous_format})
+                        new = graph.call_function(gemm_nt, (a, ytc, bias))
+                    try:
+                        yt.meta["val"] = y_val.t()
+                        ytc.meta["val"] = y_val.t().clone(
+                            memory_format=torch.contiguous_format)
+                    except Exception:
+                        pass
+        if new is None and bias is None:
+            xa = _strip_t(a)
+            if xa is not None and _contig(xa) and _contig(b) \
+                    and tn_ok(_val(xa), _val(b)):
+                # TN: mm(t(xa), b) -> gemm_tn(xa, b)
+                with graph.inserting_before(n):
+                    new = graph.call_function(gemm_tn, (xa, b))
+        if new is None:
+            continue
+        new.meta = dict(n.meta)
+        n.replace_all_uses_with(new)
+        graph.erase_node(n)
+        n_lowered += 1
+
+    if n_lowered:
+        graph.eliminate_dead_code()
+        graph.lint()
+        gm.recompile()
+        logger.info("lower_hip: lowered %d mm/addmm nodes to MFMA GEMM",
+                    n_lowered)
     return n_lowered
 
 

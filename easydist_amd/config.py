@@ -106,9 +106,10 @@ override_dtensor_rule = _env_flag("EASYDIST_OVERRIDE_DTENSOR_RULE")
 # Hand-written HIP/CDNA4 kernels for the hot ops. On a GPU box the extension
 # must load; on CPU-only hosts the aten fallback is used.
 use_hip_kernels = _env_flag("EASYDIST_USE_HIP_KERNELS", True)
-# Per-op overrides (profiling-driven): "gemm" may fall back to hipBLASLt
-# (torch.mm) per shape; see easydist_amd/ops/gemm.py.
-hip_gemm = _env_flag("EASYDIST_HIP_GEMM", False)
+# Lower mm/addmm to the hand-written MFMA GEMM ops (per-shape profiled
+# fallback to hipBLASLt lives in easydist_amd/ops/gemm.py,
+# EASYDIST_GEMM_POLICY in {hand, aten, auto}).
+hip_gemm = _env_flag("EASYDIST_HIP_GEMM", True)
 
 # ------------------------------------------------------------------ runtime --
 enable_hip_graph = _env_flag("EASYDIST_HIP_GRAPH", True)

@@ -71,7 +71,7 @@ def require_hip_ops():
     return True
 
 
-from . import attention, norms, optim, ce  # noqa: E402  (register custom ops)
+from . import attention, norms, optim, ce, gemm  # noqa: E402  (register ops)
 
-__all__ = ["attention", "norms", "optim", "ce", "load_extension",
+__all__ = ["attention", "norms", "optim", "ce", "gemm", "load_extension",
            "hip_ops_available", "require_hip_ops"]

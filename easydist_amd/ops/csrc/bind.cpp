@@ -31,6 +31,7 @@ fused_sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                double dampening, double weight_decay, bool nesterov);
 at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& bt,
                    const std::optional<at::Tensor>& bias);
+at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b);
 std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
                                                   const at::Tensor& k,
                                                   const at::Tensor& v,
@@ -87,6 +88,7 @@ TORCH_LIBRARY(easydist_amd_hip, m) {
         "float lr, float momentum, float dampening, float weight_decay, "
         "bool nesterov) -> (Tensor[], Tensor[])");
   m.def("gemm_nt(Tensor a, Tensor bt, Tensor? bias) -> Tensor");
+  m.def("gemm_tn(Tensor a, Tensor b) -> Tensor");
   m.def("flash_attn_fwd(Tensor q, Tensor k, Tensor v, bool causal) "
         "-> (Tensor, Tensor)");
   m.def("flash_attn_bwd(Tensor grad, Tensor q, Tensor k, Tensor v, "
@@ -103,6 +105,7 @@ TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
   m.impl("fused_adam_step", adam_wrap);
   m.impl("fused_sgd_step", sgd_wrap);
   m.impl("gemm_nt", gemm_nt);
+  m.impl("gemm_tn", gemm_tn);
   m.impl("flash_attn_fwd", flash_attn_fwd);
   m.impl("flash_attn_bwd", flash_attn_bwd);
 }

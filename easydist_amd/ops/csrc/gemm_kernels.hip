@@ -1,50 +1,75 @@
-// Hand-written MFMA bf16 GEMM for gfx950: C[M,N] = A[M,K] @ B[N,K]^T.
+// Hand-written MFMA bf16 GEMM family for gfx950.
 //
-// The NT layout (both operands K-contiguous) is what traced linear layers
-// produce: addmm(bias, x, t(W)) — the lowering pass rewrites mm(x, t(W))
-// to this kernel, leaving NN/TN shapes on hipBLASLt.
+// Three kernels cover the linear-layer training shapes the compiler's
+// lowering pass (compiler/passes/lower_hip.py) emits:
 //
-// Structure (cdna_hip_programming.md §5, ladder step 3): 128x128 tile,
-// 4 waves as a 2x2 wave grid (64x64 per wave), 4x4 accumulator fragments
-// of v_mfma_f32_16x16x32_bf16, K-step 32, double-buffered LDS filled by
-// 16-byte global_load_lds, XCD-aware bijective block swizzle (T1).
+//   gemm_nt_256  C[M,N] = A[M,K] @ Bt[N,K]^T (+bias)  -- fwd linears and,
+//       with a (cheap, weight-only) transpose in the graph, the dX
+//       backward.  256x256x64 tile, 8 waves, 8-phase software-pipelined
+//       schedule (cdna_hip_programming.md "256-sq 8-phase template"):
+//       global_load_lds staging with the st_16x32 XOR swizzle applied on
+//       the SOURCE address, counted s_waitcnt vmcnt(6) once per K-tile,
+//       raw s_barrier (never __syncthreads: with an LDS-DMA in flight its
+//       fence drains vmcnt and serializes the pipeline), s_setprio(1)
+//       around each 16-MFMA cluster, LDS-bounced coalesced epilogue.
+//   gemm_nt_128  same contract, 128x128x32 tile (the ladder step-3
+//       structure) for edge/small shapes; row-clamped staging so M,N need
+//       not be tile multiples.
+//   gemm_tn      C[P,Q] = sum_r A[r,P] * B[r,Q]  -- the dW backward
+//       (both operands reduce-dim-strided activations; transposing them
+//       globally would cost ~the GEMM itself).  Tiles stage k-major
+//       directly (coalesced) and fragments are read with the gfx950
+//       ds_read_b64_tr_b16 hardware transpose-read; split-R over a fp32
+//       atomic workspace fills all 256 CUs even for small [P,Q].
+//
+// MFMA contract used throughout (v_mfma_f32_16x16x32_bf16):
+//   D[m][n] += sum_k A[m][k]*B[n][k]; A/B: lane holds row (l&15),
+//   k = (l>>4)*8..+8; C/D: lane holds col (l&15), rows (l>>4)*4+r.
 #include "common.h"
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
+using v4s = __attribute__((ext_vector_type(4))) short;
+using v8s = __attribute__((ext_vector_type(8))) short;
 
-#define BM 128
-#define BN 128
-#define BK 32
 #define N_XCD 8
 
-// global_load_lds: LDS dest is wave-uniform base + lane*16 (guide §5 note)
+// bijective XCD-aware workgroup remap (cdna_hip_programming.md T1)
+DEVINL unsigned xcd_swizzle(unsigned wg, unsigned nwg) {
+  unsigned q = nwg / N_XCD, r = nwg % N_XCD;
+  unsigned xcd = wg % N_XCD, idx = wg / N_XCD;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
+// global_load_lds: LDS dest is wave-uniform base + lane*16
 DEVINL void glds16(const void* gsrc, void* lds_dst) {
   __builtin_amdgcn_global_load_lds(
       (const __attribute__((address_space(1))) unsigned int*)gsrc,
       (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
 }
 
+DEVINL void raw_barrier() { __builtin_amdgcn_s_barrier(); }
+
+// ===========================================================================
+// gemm_nt_128: 128x128 tile, BK=32, 4 waves, double-buffered glds staging.
+// Row-clamped staging + guarded epilogue: any M,N (N%1), K%32.
+// ===========================================================================
+#define BM 128
+#define BN 128
+#define BK 32
+
 extern "C" __global__ void __launch_bounds__(256)
-gemm_nt_bf16(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
-             bf16* __restrict__ C, const bf16* __restrict__ bias,
-             int M, int N, int K) {
-  // dynamic LDS: [2 buffers][A 128x32 | B 128x32] bf16
+gemm_nt_128(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+            bf16* __restrict__ C, const bf16* __restrict__ bias,
+            int M, int N, int K) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const unsigned ATILE = BM * BK * 2;          // bytes per A tile (8 KiB)
   const unsigned BUF = 2 * ATILE;              // A+B per buffer (16 KiB)
 
-  // XCD-aware bijective swizzle of the workgroup id (T1)
   unsigned nwg_m = (M + BM - 1) / BM, nwg_n = (N + BN - 1) / BN;
-  unsigned nwg = nwg_m * nwg_n;
-  unsigned wg = blockIdx.x;
-  {
-    unsigned q = nwg / N_XCD, r = nwg % N_XCD;
-    unsigned xcd = wg % N_XCD, idx = wg / N_XCD;
-    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
-  }
+  unsigned wg = xcd_swizzle(blockIdx.x, nwg_m * nwg_n);
   const unsigned m0 = (wg / nwg_n) * BM;
   const unsigned n0 = (wg % nwg_n) * BN;
 
@@ -54,22 +79,24 @@ gemm_nt_bf16(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
   const int wm = (wave / 2) * 64;   // wave row offset in tile
   const int wn = (wave % 2) * 64;   // wave col offset
 
-  // stage functions: 256 threads x 16 B = 4 KiB per pass; 2 passes per tile
+  // stage: 256 threads x 16 B = 4 KiB per pass; 2 passes per tile.
+  // Source rows are CLAMPED so edge blocks read valid (duplicated) data;
+  // the epilogue guard masks the stores.
   auto stage = [&](int buf, int k0) {
-    // A tile: rows m0..m0+127, cols k0..k0+31, row-major [128][32]
     #pragma unroll
     for (int p = 0; p < 2; ++p) {
       int e = (t + p * 256) * 8;              // element offset in tile
       int row = e / BK, col = e % BK;
-      const bf16* src = A + (long)(m0 + row) * K + k0 + col;
-      // lds base for this WAVE's 1 KiB slice (dest = base + lane*16)
+      int grow = min((int)m0 + row, M - 1);
+      const bf16* src = A + (long)grow * K + k0 + col;
       glds16(src, smem + buf * BUF + (p * 256 + wave * 64) * 16);
     }
     #pragma unroll
     for (int p = 0; p < 2; ++p) {
       int e = (t + p * 256) * 8;
       int row = e / BK, col = e % BK;
-      const bf16* src = Bt + (long)(n0 + row) * K + k0 + col;
+      int grow = min((int)n0 + row, N - 1);
+      const bf16* src = Bt + (long)grow * K + k0 + col;
       glds16(src, smem + buf * BUF + ATILE + (p * 256 + wave * 64) * 16);
     }
   };
@@ -92,9 +119,8 @@ gemm_nt_bf16(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
   for (int kt = 0; kt < nk; ++kt) {
     int buf = kt & 1;
     if (kt + 1 < nk) stage(buf ^ 1, (kt + 1) * BK);
-    // fragment reads: A row (wm + i*16 + frag_row), k = frag_k..+8
     bf16x8v a_frag[4], b_frag[4];
-    const char* base = smem;   // LDS base
+    const char* base = smem;
     #pragma unroll
     for (int i = 0; i < 4; ++i) {
       unsigned off = buf * BUF + ((wm + i * 16 + frag_row) * BK + frag_k) * 2;
@@ -116,9 +142,16 @@ gemm_nt_bf16(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     __syncthreads();
   }
 
-  // epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + reg
   const int c_col = lane & 15;
   const int c_row0 = (lane >> 4) * 4;
+  // hoisted bias loads: a per-element `if (bias) load` makes hipcc emit a
+  // branch + vmcnt(0) per element (64 dependent L2 round trips)
+  float bias_v[4] = {0.f, 0.f, 0.f, 0.f};
+  if (bias) {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      bias_v[j] = bf2f(bias[min((int)(n0 + wn + j * 16 + c_col), N - 1)]);
+  }
   #pragma unroll
   for (int i = 0; i < 4; ++i) {
     #pragma unroll
@@ -127,32 +160,413 @@ gemm_nt_bf16(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
       for (int reg = 0; reg < 4; ++reg) {
         int gm = m0 + wm + i * 16 + c_row0 + reg;
         int gn = n0 + wn + j * 16 + c_col;
-        if (gm < M && gn < N) {
-          float v = acc[i][j][reg];
-          if (bias) v += bf2f(bias[gn]);
-          C[(long)gm * N + gn] = f2bf(v);
-        }
+        if (gm < M && gn < N)
+          C[(long)gm * N + gn] = f2bf(acc[i][j][reg] + bias_v[j]);
       }
     }
   }
 }
 
+// ===========================================================================
+// gemm_nt_256: the 8-phase 256x256x64 template.
+//
+// LDS map (128 KiB, ONE extern __shared__ object -- a second one makes
+// hipcc drain vmcnt before every ds_read, cdna_hip_programming.md §5
+// ".s-level traps" (a)):
+//   buf b (0/1) at b*65536:
+//     A half h at h*16384:   rows r of the 256-row block with
+//         ((r>>6)&1)==h, as [stripe s=r>>7][r&63][64 k] bf16
+//     B half h at 32768 + h*16384: cols c with ((c>>5)&1)==h, as
+//         [stripe s=c>>6][c&31][64 k]
+// The halves interleave 64-row / 32-col stripes so that one compute
+// QUADRANT (mh,nh) of every wave touches exactly one A half and one B
+// half -- that makes half-granular staging safe with barriers only.
+//
+// Phase schedule per K-tile T (quadrants (mh,nh) in order
+// (0,0),(0,1),(1,1),(1,0) so consecutive phases share one operand half):
+//   q0: read A-h0+B-h0 (12x ds_read_b128)   stage B-h0(T+1)
+//   q1: read B-h1 (4)    [A-h0 kept in reg] stage A-h0(T+2)
+//   q2: read A-h1 (8)    [B-h1 kept]        stage B-h1(T+2)
+//   q3: read B-h0 (4)    [A-h1 kept]        stage A-h1(T+2), vmcnt(6)
+// Each phase: reads; one half staged (2 glds/thread); barrier;
+// (compiler-inserted lgkmcnt) setprio(1); 16 MFMA; setprio(0); barrier.
+// vmcnt(6) before q3's closing barrier leaves exactly the 3 newest
+// staged halves (6 loads) in flight -- everything the next tile's q0/q1
+// reads has landed.  Prologue stages 7 halves (tile0 + A0,B1,A1 of
+// tile1); B0(1) is staged by q0 of tile 0.
+// ===========================================================================
+#define BM2 256
+#define BN2 256
+#define BK2 64
+#define HALF_B 16384u
+#define BUF_B 65536u
+
+// st_16x32 XOR swizzle on a within-half byte offset (T2, rule 21)
+DEVINL unsigned sw32(unsigned x) { return x ^ (((x >> 9) & 1u) << 5); }
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+            bf16* __restrict__ C, const bf16* __restrict__ bias,
+            int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  unsigned nwg_m = (M + BM2 - 1) / BM2, nwg_n = (N + BN2 - 1) / BN2;
+  unsigned wg = xcd_swizzle(blockIdx.x, nwg_m * nwg_n);
+  const unsigned m0 = (wg / nwg_n) * BM2;
+  const unsigned n0 = (wg % nwg_n) * BN2;
+
+  const int t = threadIdx.x;
+  const int lane = t % WAVE;
+  const int wave = t / WAVE;
+  const int wr = wave >> 2;           // wave row (0/1): rows wr*128..+128
+  const int wc = wave & 3;            // wave col (0..3): cols wc*64..+64
+
+  // ---- staging ------------------------------------------------------------
+  // One A half = 16 KiB = 2 glds/thread.  LDS image is lane-linear; the
+  // inverse st_16x32 swizzle is applied to the per-lane SOURCE address.
+  // (parity, k0) are passed separately: near the K-tail the schedule
+  // issues DUMMY stages (k0 = 0) into just-freed slots so the glds count
+  // per phase stays constant -- s_waitcnt vmcnt(6) counts outstanding
+  // loads, and a skipped stage would silently weaken the guarantee for
+  // the loads that remain (wrong results at NT <= 2 and on the last two
+  // tiles).
+  auto stage_a = [&](int parity, int k0, int h) {
+    unsigned dst0 = parity * BUF_B + h * HALF_B;
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      unsigned p = (i * 512 + t) * 16;          // linear byte pos in half
+      unsigned L = sw32(p);                     // logical (image) byte
+      int s = L >> 13;                          // stripe (8 KiB each)
+      int row = (L >> 7) & 63;
+      int k = (L & 127) >> 1;
+      int grow = min((int)m0 + s * 128 + h * 64 + row, M - 1);
+      glds16(A + (long)grow * K + k0 + k,
+             smem + dst0 + (i * 512 + wave * 64) * 16);
+    }
+  };
+  auto stage_b = [&](int parity, int k0, int h) {
+    unsigned dst0 = parity * BUF_B + 32768u + h * HALF_B;
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      unsigned p = (i * 512 + t) * 16;
+      unsigned L = sw32(p);
+      int s = (L >> 12) & 3;                    // stripe (4 KiB each)
+      int col = (L >> 7) & 31;
+      int k = (L & 127) >> 1;
+      int gcol = min((int)n0 + s * 64 + h * 32 + col, N - 1);
+      glds16(Bt + (long)gcol * K + k0 + k,
+             smem + dst0 + (i * 512 + wave * 64) * 16);
+    }
+  };
+
+  // ---- fragment reads (swizzled ds_read_b128) -----------------------------
+  const int fr = lane & 15;
+  const int fg = lane >> 4;
+  auto read_a = [&](bf16x8v (&a)[4][2], int tile, int mh) {
+    unsigned base = (tile & 1) * BUF_B + mh * HALF_B;
+    #pragma unroll
+    for (int i = 0; i < 4; ++i)
+      #pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        unsigned L = wr * 8192u + (i * 16 + fr) * 128u +
+                     (kk * 32 + fg * 8) * 2u;
+        a[i][kk] = *reinterpret_cast<const bf16x8v*>(smem + base + sw32(L));
+      }
+  };
+  auto read_b = [&](bf16x8v (&b)[2][2], int tile, int nh) {
+    unsigned base = (tile & 1) * BUF_B + 32768u + nh * HALF_B;
+    #pragma unroll
+    for (int j = 0; j < 2; ++j)
+      #pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        unsigned L = wc * 4096u + (j * 16 + fr) * 128u +
+                     (kk * 32 + fg * 8) * 2u;
+        b[j][kk] = *reinterpret_cast<const bf16x8v*>(smem + base + sw32(L));
+      }
+  };
+
+  f32x4 acc[8][4];
+  #pragma unroll
+  for (int i = 0; i < 8; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  bf16x8v a_frag[4][2], b_frag[2][2];
+  const int NT = K / BK2;
+
+  // k0 of tile t, clamped to a dummy (t >= NT: data never read)
+  auto k_of = [&](int t) { return t < NT ? t * BK2 : 0; };
+
+  // prologue: tile0 fully + A0,B1,A1 of tile1 (B0(1) staged in q0 of T=0);
+  // always 7 halves = 14 loads so vmcnt(6) drains exactly tile 0
+  stage_a(0, 0, 0); stage_a(0, 0, 1); stage_b(0, 0, 0); stage_b(0, 0, 1);
+  stage_a(1, k_of(1), 0); stage_b(1, k_of(1), 1); stage_a(1, k_of(1), 1);
+  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  raw_barrier();
+
+  auto mfma_quad = [&](int mh, int nh) {
+    __builtin_amdgcn_s_setprio(1);
+    #pragma unroll
+    for (int i = 0; i < 4; ++i)
+      #pragma unroll
+      for (int j = 0; j < 2; ++j)
+        #pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          acc[mh * 4 + i][nh * 2 + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i][kk], b_frag[j][kk], acc[mh * 4 + i][nh * 2 + j],
+              0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  for (int T = 0; T < NT; ++T) {
+    // q0: (mh0, nh0)
+    read_a(a_frag, T, 0);
+    read_b(b_frag, T, 0);
+    stage_b((T + 1) & 1, k_of(T + 1), 0);
+    raw_barrier();
+    mfma_quad(0, 0);
+    raw_barrier();
+    // q1: (mh0, nh1) -- A kept
+    read_b(b_frag, T, 1);
+    stage_a(T & 1, k_of(T + 2), 0);
+    raw_barrier();
+    mfma_quad(0, 1);
+    raw_barrier();
+    // q2: (mh1, nh1) -- B kept
+    read_a(a_frag, T, 1);
+    stage_b(T & 1, k_of(T + 2), 1);
+    raw_barrier();
+    mfma_quad(1, 1);
+    raw_barrier();
+    // q3: (mh1, nh0) -- A kept, B re-read
+    read_b(b_frag, T, 0);
+    stage_a(T & 1, k_of(T + 2), 1);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    raw_barrier();
+    mfma_quad(1, 0);
+    raw_barrier();
+  }
+
+  // ---- epilogue: LDS-bounced coalesced stores -----------------------------
+  // All staging consumed; repurpose the wave's 16 KiB share as a [128][64]
+  // bf16 bounce so global stores are 16-B wide and row-contiguous.
+  bf16* share = reinterpret_cast<bf16*>(smem) + wave * 8192;
+  const unsigned wm = wr * 128, wn = wc * 64;
+  // hoisted bias loads (4 per lane) -- see gemm_nt_128 epilogue note
+  float bias_v[4] = {0.f, 0.f, 0.f, 0.f};
+  if (bias) {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      bias_v[j] = bf2f(bias[min((int)(n0 + wn + j * 16 + fr), N - 1)]);
+  }
+  #pragma unroll
+  for (int i = 0; i < 8; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i * 16 + fg * 4 + r;
+        int col = j * 16 + fr;
+        share[row * 64 + col] = f2bf(acc[i][j][r] + bias_v[j]);
+      }
+  lds_fence();   // own-wave LDS writes -> reads; no cross-wave traffic
+  #pragma unroll
+  for (int tt = 0; tt < 16; ++tt) {
+    int row = tt * 8 + (lane >> 3);
+    int colc = (lane & 7) * 8;
+    int gm = m0 + wm + row;
+    int gn = n0 + wn + colc;
+    if (gm < (int)M && gn + 8 <= (int)N) {
+      bf16x8v val = *reinterpret_cast<const bf16x8v*>(&share[row * 64 + colc]);
+      *reinterpret_cast<bf16x8v*>(&C[(long)gm * N + gn]) = val;
+    } else if (gm < (int)M) {
+      for (int e = 0; e < 8 && gn + e < (int)N; ++e)
+        C[(long)gm * N + gn + e] = share[row * 64 + colc + e];
+    }
+  }
+}
+
+// ===========================================================================
+// gemm_tn: C[P,Q] = sum_r A[r,P]*B[r,Q], fp32 atomic accumulation into a
+// workspace (split-R fills the 256 CUs when P*Q is small -- dW shapes).
+// Tiles: BR=32 x 128; both operands stage k(r)-major via glds (coalesced:
+// global rows ARE r-rows) into tr-read images:
+//   image = [pchunk p/16][rblock swz(r/4)][4 r][16 p] bf16, where the
+//   r-blocks are stored in order 0-3,8-11,16-19,24-27,4-7,... so that one
+//   ds_read_b64_tr_b16 serves each 16-lane group the k-range its MFMA
+//   fragment needs (fragment k = 8*(l>>4)+0..7 -> read1 blocks {0,2,4,6},
+//   read2 at +512 B blocks {1,3,5,7}).
+// ===========================================================================
+#define TBR 32
+#define TBP 128
+
+// r-block storage order: even blocks first (see header comment)
+DEVINL int tn_unswz(int pos) { return ((pos >> 2) & 1) | ((pos & 3) << 1); }
+
+// one 8 KiB tr-image stage: tile [32 r][128 p] from src rows r0.., cols p0..
+DEVINL void tn_stage(const bf16* src, long ld, int r0, int p0, int pmax,
+                     char* dst_base, int t, int wave) {
+  #pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    unsigned p = (i * 256 + t) * 16;          // linear byte pos (8 KiB)
+    int chunk16 = p >> 10;                    // 1 KiB per 16-p chunk
+    unsigned w = p & 1023u;
+    int bpos = w >> 7;                        // 128 B per [4r][16p] block
+    int blk = tn_unswz(bpos);
+    int rr = blk * 4 + ((w & 127) >> 5);
+    int pp = chunk16 * 16 + ((w & 31) >> 1);
+    int gp = min(p0 + pp, pmax - 1);
+    glds16(src + (long)(r0 + rr) * ld + gp,
+           dst_base + (i * 256 + wave * 64) * 16);
+  }
+}
+
+DEVINL bf16x8v tn_frag(const char* img, int chunk, int lane) {
+  const char* base = img + chunk * 1024 + (lane >> 4) * 128 + (lane & 15) * 8;
+  v4s lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) v4s*)base);
+  v4s hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) v4s*)(base + 512));
+  v8s out = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+  return __builtin_bit_cast(bf16x8v, out);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
+               float* __restrict__ Cw, int R, int P, int Q, int splitr) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const unsigned IMG = TBR * TBP * 2;          // 8 KiB per operand image
+  const unsigned BUF = 2 * IMG;                // A+B per buffer
+
+  unsigned nwg_p = P / TBP, nwg_q = Q / TBP;
+  unsigned ntile = nwg_p * nwg_q;
+  unsigned wg = xcd_swizzle(blockIdx.x, ntile * splitr);
+  unsigned tile = wg / splitr, slice = wg % splitr;
+  const unsigned p0 = (tile / nwg_q) * TBP;
+  const unsigned q0 = (tile % nwg_q) * TBP;
+
+  // this slice's r-range (R % 32 == 0 enforced by host)
+  int nrt_all = R / TBR;
+  int per = (nrt_all + splitr - 1) / splitr;
+  int rt0 = slice * per;
+  int rt1 = min(nrt_all, rt0 + per);
+  if (rt0 >= rt1) return;
+
+  const int t = threadIdx.x;
+  const int lane = t % WAVE;
+  const int wave = t / WAVE;
+  const int wp = (wave >> 1) * 64;
+  const int wq = (wave & 1) * 64;
+
+  auto stage = [&](int buf, int rt) {
+    tn_stage(Ag, P, rt * TBR, p0, P, smem + buf * BUF, t, wave);
+    tn_stage(Bg, Q, rt * TBR, q0, Q, smem + buf * BUF + IMG, t, wave);
+  };
+
+  f32x4 acc[4][4];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage(0, rt0);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+
+  for (int rt = rt0; rt < rt1; ++rt) {
+    int buf = (rt - rt0) & 1;
+    if (rt + 1 < rt1) stage(buf ^ 1, rt + 1);
+    bf16x8v a_frag[4], b_frag[4];
+    #pragma unroll
+    for (int i = 0; i < 4; ++i)
+      a_frag[i] = tn_frag(smem + buf * BUF, (wp >> 4) + i, lane);
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      b_frag[j] = tn_frag(smem + buf * BUF + IMG, (wq >> 4) + j, lane);
+    #pragma unroll
+    for (int i = 0; i < 4; ++i)
+      #pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+  }
+
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+  #pragma unroll
+  for (int i = 0; i < 4; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int gp = p0 + wp + i * 16 + c_row0 + reg;
+        int gq = q0 + wq + j * 16 + c_col;
+        if (gp < P && gq < Q) {
+          if (splitr == 1)
+            Cw[(long)gp * Q + gq] = acc[i][j][reg];
+          else
+            atomicAdd(&Cw[(long)gp * Q + gq], acc[i][j][reg]);
+        }
+      }
+}
+
+// ===========================================================================
+// host wrappers
+// ===========================================================================
 at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& bt,
                    const std::optional<at::Tensor>& bias) {
   TORCH_CHECK(a.dtype() == at::kBFloat16 && bt.dtype() == at::kBFloat16);
   TORCH_CHECK(a.is_contiguous() && bt.is_contiguous(),
               "gemm_nt wants K-contiguous operands");
-  const int M = a.size(0), K = a.size(1), N = bt.size(0);
+  const long M = a.size(0), K = a.size(1), N = bt.size(0);
   TORCH_CHECK(bt.size(1) == K);
-  TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0,
-              "gemm_nt v1: M,N multiples of 128, K multiple of 32");
+  TORCH_CHECK(K % BK == 0, "gemm_nt: K must be a multiple of 32");
+  TORCH_CHECK(N % 8 == 0, "gemm_nt: N must be a multiple of 8");
   auto c = at::empty({M, N}, a.options());
   auto stream = at::cuda::getCurrentCUDAStream();
-  unsigned nwg = (M / BM) * (N / BN);
-  size_t lds = 2 * 2 * BM * BK * 2;   // 32 KiB
-  hipLaunchKernelGGL(gemm_nt_bf16, dim3(nwg), dim3(256), lds, stream,
-      (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
-      (bf16*)c.data_ptr(),
-      bias ? (const bf16*)bias->data_ptr() : nullptr, M, N, K);
+  const bf16* bias_p = bias ? (const bf16*)bias->data_ptr() : nullptr;
+  if (K % BK2 == 0 && M >= BM2 && N >= 128) {
+    unsigned nwg = ((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
+    hipLaunchKernelGGL(gemm_nt_256, dim3(nwg), dim3(512), 2 * BUF_B, stream,
+        (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
+        (bf16*)c.data_ptr(), bias_p, (int)M, (int)N, (int)K);
+  } else {
+    unsigned nwg = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+    size_t lds = 2 * 2 * BM * BK * 2;   // 32 KiB
+    hipLaunchKernelGGL(gemm_nt_128, dim3(nwg), dim3(256), lds, stream,
+        (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
+        (bf16*)c.data_ptr(), bias_p, (int)M, (int)N, (int)K);
+  }
   return c;
+}
+
+at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
+  // C[P,Q] = a^T @ b with a:[R,P], b:[R,Q]
+  TORCH_CHECK(a.dtype() == at::kBFloat16 && b.dtype() == at::kBFloat16);
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+  const long R = a.size(0), P = a.size(1), Q = b.size(0) == R ? b.size(1) : -1;
+  TORCH_CHECK(Q > 0, "gemm_tn: reduce dims must match");
+  TORCH_CHECK(R % TBR == 0, "gemm_tn: R must be a multiple of 32");
+  TORCH_CHECK(P % TBP == 0 && Q % TBP == 0,
+              "gemm_tn: P, Q must be multiples of 128");
+  unsigned ntile = (P / TBP) * (Q / TBP);
+  // split the reduction so the grid covers the chip's 256 CUs
+  int splitr = 1;
+  while (ntile * splitr < 512 && splitr < 16 &&
+         (R / TBR) % (splitr * 2) == 0 && (R / TBR) / (splitr * 2) >= 1)
+    splitr *= 2;
+  auto cw = splitr == 1
+      ? at::empty({P, Q}, a.options().dtype(at::kFloat))
+      : at::zeros({P, Q}, a.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  size_t lds = 2 * 2 * TBR * TBP * 2;   // 32 KiB
+  hipLaunchKernelGGL(gemm_tn_kernel, dim3(ntile * splitr), dim3(256), lds,
+      stream,
+      (const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+      cw.data_ptr<float>(), (int)R, (int)P, (int)Q, splitr);
+  return cw.to(at::kBFloat16);
 }

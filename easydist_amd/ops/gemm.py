@@ -1,0 +1,163 @@
+"""Hand-written MFMA GEMM as custom ops, with per-shape profiled dispatch.
+
+The lowering pass (compiler/passes/lower_hip.py:lower_gemm) rewrites the
+sharded graph's linear-layer matmuls to these ops:
+
+* ``easydist_amd::gemm_nt(a, bt, bias)`` — C = a @ bt.T (+bias), the
+  NT shape every traced ``nn.Linear`` forward produces (both operands
+  K-contiguous), and — with a cheap weight-side transpose inserted by the
+  pass — the dX backward too.  CUDA: the 256x256 8-phase MFMA kernel
+  (csrc/gemm_kernels.hip), 128x128 for edge shapes.
+* ``easydist_amd::gemm_tn(a, b)`` — C = a.T @ b with BOTH operands
+  reduce-dim-strided (the dW backward; transposing the activations
+  globally would cost about the GEMM itself).  CUDA: split-R tr-read
+  MFMA kernel.
+
+Dispatch policy (EASYDIST_GEMM_POLICY): ``hand`` always uses the HIP
+kernel when the shape is supported; ``aten`` never does (hipBLASLt);
+``auto`` (default) microbenchmarks both on first sight of a shape and
+caches the winner, preferring the hand kernel within a 3% tie margin
+(reference VERDICT item 1: per-shape profiled fallback).  The autotuner
+refuses to run while a hipGraph capture is active (it synchronizes) —
+shapes must be warmed up eagerly first, which the compile warmup does.
+"""
+from __future__ import annotations
+
+import logging
+import os
+from typing import Optional
+
+import torch
+
+from . import load_extension
+from .. import config as mdconfig
+
+logger = logging.getLogger(__name__)
+
+lib = torch.library.Library("easydist_amd", "FRAGMENT")
+lib.define("gemm_nt(Tensor a, Tensor bt, Tensor? bias) -> Tensor")
+lib.define("gemm_tn(Tensor a, Tensor b) -> Tensor")
+
+
+def _policy() -> str:
+    return os.environ.get("EASYDIST_GEMM_POLICY", "auto")
+
+
+# ---------------------------------------------------------------- CPU -------
+def _nt_cpu(a, bt, bias):
+    out = a @ bt.t()
+    if bias is not None:
+        out = out + bias
+    return out
+
+
+def _tn_cpu(a, b):
+    return a.t() @ b
+
+
+# ---------------------------------------------------------------- CUDA ------
+def _nt_supported(a, bt):
+    return (a.dtype == torch.bfloat16 and bt.dtype == torch.bfloat16
+            and a.dim() == 2 and bt.dim() == 2
+            and a.shape[1] == bt.shape[1]
+            and a.shape[1] % 32 == 0 and bt.shape[0] % 8 == 0
+            and a.shape[0] >= 16)
+
+
+def _tn_supported(a, b):
+    return (a.dtype == torch.bfloat16 and b.dtype == torch.bfloat16
+            and a.dim() == 2 and b.dim() == 2 and a.shape[0] == b.shape[0]
+            and a.shape[0] % 32 == 0
+            and a.shape[1] % 128 == 0 and b.shape[1] % 128 == 0)
+
+
+_tune_cache = {}
+
+
+def _time_fn(fn, iters=3):
+    start = torch.cuda.Event(enable_timing=True)
+    end = torch.cuda.Event(enable_timing=True)
+    fn()  # warm
+    start.record()
+    for _ in range(iters):
+        fn()
+    end.record()
+    end.synchronize()
+    return start.elapsed_time(end) / iters
+
+
+def _choose(key, hand_fn, aten_fn):
+    """Per-shape profiled dispatch. Returns True to use the hand kernel."""
+    pol = _policy()
+    if pol == "hand":
+        return True
+    if pol == "aten":
+        return False
+    if key in _tune_cache:
+        return _tune_cache[key]
+    if torch.cuda.is_current_stream_capturing():
+        # cannot time inside capture; the warmup should have seeded the
+        # cache — default to the hand kernel rather than silently
+        # switching the captured graph's kernels
+        return True
+    try:
+        t_hand = _time_fn(hand_fn)
+        t_aten = _time_fn(aten_fn)
+        use = t_hand <= t_aten * 1.03
+        logger.info("gemm autotune %s: hand %.3fms aten %.3fms -> %s",
+                    key, t_hand, t_aten, "hand" if use else "aten")
+    except Exception as e:  # pragma: no cover
+        logger.warning("gemm autotune failed for %s: %s", key, e)
+        use = False
+    _tune_cache[key] = use
+    return use
+
+
+def _nt_cuda(a, bt, bias):
+    ext = load_extension()
+    if ext is None or not _nt_supported(a, bt) \
+            or not mdconfig.use_hip_kernels:
+        if ext is None and mdconfig.use_hip_kernels:
+            from . import require_hip_ops
+            require_hip_ops()
+        return _nt_cpu(a, bt, bias)
+    a = a.contiguous()
+    bt = bt.contiguous()
+    b_c = bias.contiguous() if bias is not None else None
+    key = ("nt", a.shape[0], a.shape[1], bt.shape[0], bias is not None)
+    if _choose(key, lambda: ext.gemm_nt(a, bt, b_c),
+               lambda: _nt_cpu(a, bt, bias)):
+        return ext.gemm_nt(a, bt, b_c)
+    return _nt_cpu(a, bt, bias)
+
+
+def _tn_cuda(a, b):
+    ext = load_extension()
+    if ext is None or not _tn_supported(a, b) \
+            or not mdconfig.use_hip_kernels:
+        if ext is None and mdconfig.use_hip_kernels:
+            from . import require_hip_ops
+            require_hip_ops()
+        return _tn_cpu(a, b)
+    a = a.contiguous()
+    b = b.contiguous()
+    key = ("tn", a.shape[0], a.shape[1], b.shape[1])
+    if _choose(key, lambda: ext.gemm_tn(a, b), lambda: _tn_cpu(a, b)):
+        return ext.gemm_tn(a, b)
+    return _tn_cpu(a, b)
+
+
+lib.impl("gemm_nt", _nt_cpu, "CPU")
+lib.impl("gemm_nt", _nt_cuda, "CUDA")
+lib.impl("gemm_tn", _tn_cpu, "CPU")
+lib.impl("gemm_tn", _tn_cuda, "CUDA")
+
+
+@torch.library.register_fake("easydist_amd::gemm_nt")
+def _nt_fake(a, bt, bias):
+    return a.new_empty((a.shape[0], bt.shape[0]))
+
+
+@torch.library.register_fake("easydist_amd::gemm_tn")
+def _tn_fake(a, b):
+    return a.new_empty((a.shape[1], b.shape[1]))

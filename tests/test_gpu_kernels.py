@@ -178,3 +178,57 @@ def test_flash_attn_bwd(ext):
                                 (dv, vf.grad, "dv")):
             err = (got.float() - want).abs().max()
             assert err < 8e-2, (name, float(err), B, H, S, D)
+
+
+@requires_gpu
+def test_gemm_nt_256_path(ext):
+    # clean 256x256-tile shapes plus M/N tails (row-clamped staging)
+    torch.manual_seed(5)
+    for (M, N, K) in [(512, 512, 128), (256, 768, 192), (512, 384, 64),
+                      (300, 512, 128), (512, 520, 128)]:
+        a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        bt = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+        bias = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+        c = ext.gemm_nt(a, bt, bias)
+        ref = a.float() @ bt.float().t() + bias.float()
+        assert torch.allclose(c.float(), ref, atol=1.0, rtol=3e-2), \
+            ((M, N, K), float((c.float() - ref).abs().max()))
+
+
+@requires_gpu
+def test_gemm_nt_256_asymmetric(ext):
+    # transpose-detecting spike check on the 256-tile path (guide §3)
+    M, N, K = 512, 512, 128
+    a = torch.zeros(M, K, device="cuda", dtype=torch.bfloat16)
+    a[300, 65] = 2.0
+    bt = torch.zeros(N, K, device="cuda", dtype=torch.bfloat16)
+    bt[270, 65] = 3.0
+    c = ext.gemm_nt(a, bt, None)
+    assert float(c[300, 270]) == pytest.approx(6.0, abs=1e-2)
+    assert float(c.float().abs().sum()) == pytest.approx(6.0, abs=1e-2)
+
+
+@requires_gpu
+def test_gemm_tn(ext):
+    # C[P,Q] = a^T @ b, both operands reduce-dim-strided (dW shape)
+    torch.manual_seed(6)
+    for (R, P, Q) in [(1024, 256, 128), (512, 128, 384), (4096, 128, 128)]:
+        a = torch.randn(R, P, device="cuda", dtype=torch.bfloat16)
+        b = torch.randn(R, Q, device="cuda", dtype=torch.bfloat16)
+        c = ext.gemm_tn(a, b)
+        ref = a.float().t() @ b.float()
+        # fp32 atomic accumulation; tolerance scales with sqrt(R)
+        assert torch.allclose(c.float(), ref, atol=2.0, rtol=3e-2), \
+            ((R, P, Q), float((c.float() - ref).abs().max()))
+
+
+@requires_gpu
+def test_gemm_tn_asymmetric(ext):
+    R, P, Q = 256, 256, 128
+    a = torch.zeros(R, P, device="cuda", dtype=torch.bfloat16)
+    a[100, 37] = 2.0
+    b = torch.zeros(R, Q, device="cuda", dtype=torch.bfloat16)
+    b[100, 85] = 3.0
+    c = ext.gemm_tn(a, b)
+    assert float(c[37, 85]) == pytest.approx(6.0, abs=1e-2)
+    assert float(c.float().abs().sum()) == pytest.approx(6.0, abs=1e-2)
