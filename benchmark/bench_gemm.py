@@ -12,6 +12,10 @@ from __future__ import annotations
 
 import argparse
 import json
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
@@ -70,12 +74,22 @@ def main():
     p.add_argument("--tokens", type=int, default=65536)  # b64 x s1024
     p.add_argument("--hidden", type=int, default=768)
     p.add_argument("--vocab", type=int, default=50304)
+    p.add_argument("--only", default=None,
+                   help="single shape 'nt,M,N,K' or 'tn,R,P,Q' (for PMC runs)")
     args = p.parse_args()
     assert torch.cuda.is_available()
     ext = ops.load_extension()
     assert ext is not None, "build the extension first"
 
     M, H, V = args.tokens, args.hidden, args.vocab
+    if args.only:
+        parts = args.only.split(",")
+        if parts[0] == "nt":
+            r = bench_nt(ext, int(parts[1]), int(parts[2]), int(parts[3]))
+        else:
+            r = bench_tn(ext, int(parts[1]), int(parts[2]), int(parts[3]))
+        print(json.dumps(r))
+        return
     rows = []
     # square probes (vs guide ladder numbers)
     for s in (4096, 8192):

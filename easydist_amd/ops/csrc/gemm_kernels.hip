@@ -292,7 +292,7 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     for (int j = 0; j < 4; ++j)
       acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  bf16x8v a_frag[4][2], b_frag[2][2];
+  bf16x8v a_frag[4][2], b_frag[2][2], b0_frag[2][2];
   const int NT = K / BK2;
 
   // k0 of tile t, clamped to a dummy (t >= NT: data never read)
@@ -305,7 +305,7 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
   asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   raw_barrier();
 
-  auto mfma_quad = [&](int mh, int nh) {
+  auto mfma_quad = [&](int mh, int nh, bf16x8v (&bf)[2][2]) {
     __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -314,7 +314,7 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
         #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
           acc[mh * 4 + i][nh * 2 + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[i][kk], b_frag[j][kk], acc[mh * 4 + i][nh * 2 + j],
+              a_frag[i][kk], bf[j][kk], acc[mh * 4 + i][nh * 2 + j],
               0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
   };
@@ -322,29 +322,28 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
   for (int T = 0; T < NT; ++T) {
     // q0: (mh0, nh0)
     read_a(a_frag, T, 0);
-    read_b(b_frag, T, 0);
+    read_b(b0_frag, T, 0);
     stage_b((T + 1) & 1, k_of(T + 1), 0);
     raw_barrier();
-    mfma_quad(0, 0);
+    mfma_quad(0, 0, b0_frag);
     raw_barrier();
     // q1: (mh0, nh1) -- A kept
     read_b(b_frag, T, 1);
     stage_a(T & 1, k_of(T + 2), 0);
     raw_barrier();
-    mfma_quad(0, 1);
+    mfma_quad(0, 1, b_frag);
     raw_barrier();
     // q2: (mh1, nh1) -- B kept
     read_a(a_frag, T, 1);
     stage_b(T & 1, k_of(T + 2), 1);
     raw_barrier();
-    mfma_quad(1, 1);
+    mfma_quad(1, 1, b_frag);
     raw_barrier();
-    // q3: (mh1, nh0) -- A kept, B re-read
-    read_b(b_frag, T, 0);
+    // q3: (mh1, nh0) -- A kept, B-h0 kept from q0 (no re-read)
     stage_a(T & 1, k_of(T + 2), 1);
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     raw_barrier();
-    mfma_quad(1, 0);
+    mfma_quad(1, 0, b0_frag);
     raw_barrier();
   }
 
