@@ -201,10 +201,23 @@ gemm_nt_128(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
 #define HALF_B 16384u
 #define BUF_B 65536u
 
-// st_16x32 XOR swizzle on a within-half byte offset (T2, rule 21)
-DEVINL unsigned sw32(unsigned x) { return x ^ (((x >> 9) & 1u) << 5); }
+// XOR swizzle on a within-half byte offset (T2, rule 21).  The row bits
+// (>=7: 128-B rows) are folded into the 16-B-slot bits so that the 16
+// lanes of one ds_read_b128 service group land on distinct banks.  The
+// HW's lane grouping is not documented; variant 1 is conflict-free if
+// groups are {l%4==g} (row bits 9-10 -> slot bits 6-7), variant 2 if
+// they are contiguous {16g..16g+15} (row bits 8-10 -> slot bits 4-6);
+// variant 0 is the guide's 1-bit st_16x32 form.  A/B'd on hardware via
+// EASYDIST_NT256_SWZ; the win ships as the default.
+template <int SWZ>
+DEVINL unsigned sw32(unsigned x) {
+  if constexpr (SWZ == 0) return x ^ (((x >> 9) & 1u) << 5);
+  else if constexpr (SWZ == 1) return x ^ (((x >> 9) & 3u) << 6);
+  else return x ^ (((x >> 8) & 7u) << 4);
+}
 
-extern "C" __global__ void __launch_bounds__(512)
+template <int SWZ>
+__global__ void __launch_bounds__(512)
 gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
             bf16* __restrict__ C, const bf16* __restrict__ bias,
             int M, int N, int K) {
@@ -235,7 +248,7 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     #pragma unroll
     for (int i = 0; i < 2; ++i) {
       unsigned p = (i * 512 + t) * 16;          // linear byte pos in half
-      unsigned L = sw32(p);                     // logical (image) byte
+      unsigned L = sw32<SWZ>(p);                     // logical (image) byte
       int s = L >> 13;                          // stripe (8 KiB each)
       int row = (L >> 7) & 63;
       int k = (L & 127) >> 1;
@@ -249,7 +262,7 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     #pragma unroll
     for (int i = 0; i < 2; ++i) {
       unsigned p = (i * 512 + t) * 16;
-      unsigned L = sw32(p);
+      unsigned L = sw32<SWZ>(p);
       int s = (L >> 12) & 3;                    // stripe (4 KiB each)
       int col = (L >> 7) & 31;
       int k = (L & 127) >> 1;
@@ -270,7 +283,7 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
       for (int kk = 0; kk < 2; ++kk) {
         unsigned L = wr * 8192u + (i * 16 + fr) * 128u +
                      (kk * 32 + fg * 8) * 2u;
-        a[i][kk] = *reinterpret_cast<const bf16x8v*>(smem + base + sw32(L));
+        a[i][kk] = *reinterpret_cast<const bf16x8v*>(smem + base + sw32<SWZ>(L));
       }
   };
   auto read_b = [&](bf16x8v (&b)[2][2], int tile, int nh) {
@@ -281,7 +294,7 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
       for (int kk = 0; kk < 2; ++kk) {
         unsigned L = wc * 4096u + (j * 16 + fr) * 128u +
                      (kk * 32 + fg * 8) * 2u;
-        b[j][kk] = *reinterpret_cast<const bf16x8v*>(smem + base + sw32(L));
+        b[j][kk] = *reinterpret_cast<const bf16x8v*>(smem + base + sw32<SWZ>(L));
       }
   };
 
@@ -348,8 +361,14 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
   }
 
   // ---- epilogue: LDS-bounced coalesced stores -----------------------------
-  // All staging consumed; repurpose the wave's 16 KiB share as a [128][64]
-  // bf16 bounce so global stores are 16-B wide and row-contiguous.
+  // DRAIN the LDS-DMA queue first: the tail iterations issued dummy
+  // stages whose glds may still be in flight — without this they land on
+  // top of the bounce data below (timing-dependent corruption, seen as
+  // err~100 on the 50k-wide vocab GEMM).
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  raw_barrier();
+  // Repurpose the wave's 16 KiB share as a [128][64] bf16 bounce so
+  // global stores are 16-B wide and row-contiguous.
   bf16* share = reinterpret_cast<bf16*>(smem) + wave * 8192;
   const unsigned wm = wr * 128, wn = wc * 64;
   // hoisted bias loads (4 per lane) -- see gemm_nt_128 epilogue note
@@ -530,7 +549,13 @@ at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& bt,
   const bf16* bias_p = bias ? (const bf16*)bias->data_ptr() : nullptr;
   if (K % BK2 == 0 && M >= BM2 && N >= 128) {
     unsigned nwg = ((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
-    hipLaunchKernelGGL(gemm_nt_256, dim3(nwg), dim3(512), 2 * BUF_B, stream,
+    static int swz = []() {
+      const char* e = getenv("EASYDIST_NT256_SWZ");
+      return e ? atoi(e) : 1;
+    }();
+    auto kern = swz == 0 ? gemm_nt_256<0> : swz == 2 ? gemm_nt_256<2>
+                                          : gemm_nt_256<1>;
+    hipLaunchKernelGGL(kern, dim3(nwg), dim3(512), 2 * BUF_B, stream,
         (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
         (bf16*)c.data_ptr(), bias_p, (int)M, (int)N, (int)K);
   } else {
