@@ -332,30 +332,34 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     __builtin_amdgcn_s_setprio(0);
   };
 
+  // ONE barrier per phase (the closing one).  Correctness argument: a
+  // phase's ds_reads are consumed by its MFMAs, so they complete before
+  // the closing barrier in program order; a glds staged in phase p+1
+  // overwrites data whose last read was in a phase <= p, separated by at
+  // least that closing barrier.  The pre-MFMA barrier of the textbook
+  // template only phase-locks the waves -- dropping it lets a wave start
+  // its MFMAs as soon as ITS reads land (measured: the two-barrier form
+  // spent 8x hipBLASLt's cycles waiting on LDS).
   for (int T = 0; T < NT; ++T) {
     // q0: (mh0, nh0)
     read_a(a_frag, T, 0);
     read_b(b0_frag, T, 0);
     stage_b((T + 1) & 1, k_of(T + 1), 0);
-    raw_barrier();
     mfma_quad(0, 0, b0_frag);
     raw_barrier();
     // q1: (mh0, nh1) -- A kept
     read_b(b_frag, T, 1);
     stage_a(T & 1, k_of(T + 2), 0);
-    raw_barrier();
     mfma_quad(0, 1, b_frag);
     raw_barrier();
     // q2: (mh1, nh1) -- B kept
     read_a(a_frag, T, 1);
     stage_b(T & 1, k_of(T + 2), 1);
-    raw_barrier();
     mfma_quad(1, 1, b_frag);
     raw_barrier();
     // q3: (mh1, nh0) -- A kept, B-h0 kept from q0 (no re-read)
     stage_a(T & 1, k_of(T + 2), 1);
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-    raw_barrier();
     mfma_quad(1, 0, b0_frag);
     raw_barrier();
   }
@@ -489,13 +493,20 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
     for (int j = 0; j < 4; ++j)
       acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  // Pipelined loop: ONE raw barrier + counted vmcnt per R-tile so the
+  // next tile's 4 glds stay in flight under the MFMAs (a __syncthreads
+  // here would drain vmcnt and serialize — guide §5 glds table).
+  // Per iteration: barrier (all waves done reading buf^1 last iter);
+  // stage rt+1 into buf^1; vmcnt(4) (leaves exactly those 4 in flight,
+  // so rt's own stages from last iteration have landed); read + MFMA rt.
+  // The tail restages rt (dummy) to keep the glds count per iteration
+  // constant — vmcnt(N) counts outstanding loads.
   stage(0, rt0);
-  __builtin_amdgcn_s_waitcnt(0);
-  __syncthreads();
-
   for (int rt = rt0; rt < rt1; ++rt) {
     int buf = (rt - rt0) & 1;
-    if (rt + 1 < rt1) stage(buf ^ 1, rt + 1);
+    raw_barrier();
+    stage(buf ^ 1, rt + 1 < rt1 ? rt + 1 : rt);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     bf16x8v a_frag[4], b_frag[4];
     #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -509,8 +520,6 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
       for (int j = 0; j < 4; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
-    __builtin_amdgcn_s_waitcnt(0);
-    __syncthreads();
   }
 
   const int c_col = lane & 15;
