@@ -341,14 +341,16 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
   // its MFMAs as soon as ITS reads land (measured: the two-barrier form
   // spent 8x hipBLASLt's cycles waiting on LDS).
   for (int T = 0; T < NT; ++T) {
-    // q0: (mh0, nh0)
+    // q0: (mh0, nh0); BOTH B halves read here (B-h1(T) was staged a full
+    // tile before B-h0(T), so the vmcnt that covers B-h0 covers it too) —
+    // q1/q3 then have no LDS reads at all.
     read_a(a_frag, T, 0);
     read_b(b0_frag, T, 0);
+    read_b(b_frag, T, 1);
     stage_b((T + 1) & 1, k_of(T + 1), 0);
     mfma_quad(0, 0, b0_frag);
     raw_barrier();
-    // q1: (mh0, nh1) -- A kept
-    read_b(b_frag, T, 1);
+    // q1: (mh0, nh1) -- A and B-h1 kept
     stage_a(T & 1, k_of(T + 2), 0);
     mfma_quad(0, 1, b_frag);
     raw_barrier();
@@ -357,7 +359,7 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     stage_b(T & 1, k_of(T + 2), 1);
     mfma_quad(1, 1, b_frag);
     raw_barrier();
-    // q3: (mh1, nh0) -- A kept, B-h0 kept from q0 (no re-read)
+    // q3: (mh1, nh0) -- A kept, B-h0 kept from q0
     stage_a(T & 1, k_of(T + 2), 1);
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     mfma_quad(1, 0, b0_frag);
@@ -493,20 +495,16 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
     for (int j = 0; j < 4; ++j)
       acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // Pipelined loop: ONE raw barrier + counted vmcnt per R-tile so the
-  // next tile's 4 glds stay in flight under the MFMAs (a __syncthreads
-  // here would drain vmcnt and serialize — guide §5 glds table).
-  // Per iteration: barrier (all waves done reading buf^1 last iter);
-  // stage rt+1 into buf^1; vmcnt(4) (leaves exactly those 4 in flight,
-  // so rt's own stages from last iteration have landed); read + MFMA rt.
-  // The tail restages rt (dummy) to keep the glds count per iteration
-  // constant — vmcnt(N) counts outstanding loads.
+  // Double-buffered loop: stage rt+1, compute rt, drain, barrier.  The
+  // counted-vmcnt raw-barrier variant was TRIED and measured 25% SLOWER
+  // across the dW shapes (the issue-then-drain placement lets the glds
+  // overlap the whole MFMA phase; the early vmcnt wait did not).
   stage(0, rt0);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
   for (int rt = rt0; rt < rt1; ++rt) {
     int buf = (rt - rt0) & 1;
-    raw_barrier();
-    stage(buf ^ 1, rt + 1 < rt1 ? rt + 1 : rt);
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    if (rt + 1 < rt1) stage(buf ^ 1, rt + 1);
     bf16x8v a_frag[4], b_frag[4];
     #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -520,6 +518,8 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
       for (int j = 0; j < 4; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
   }
 
   const int c_col = lane & 15;
