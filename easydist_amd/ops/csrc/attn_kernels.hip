@@ -23,16 +23,18 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 #define QBP (QB + 8)
 
 template <int D>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                  const bf16* __restrict__ V, bf16* __restrict__ O,
                  float* __restrict__ LSE, int B, int H, int S, bool causal,
                  float scale) {
-  // grid: (ceil(S/QBLK), B*H); QBLK = 128 q rows per workgroup — each of
-  // the 4 waves owns TWO 16-row fragments (RF=2), doubling the MFMA work
-  // per staged K/V tile (arithmetic intensity) at the same LDS footprint.
-  constexpr int RF = 2;
-  constexpr int QBLK = 4 * 16 * RF;
+  // grid: (ceil(S/QBLK), B*H); QBLK = 128 q rows per workgroup, 8 waves
+  // of ONE 16-row fragment each.  The round-1 4-wave/RF=2 form ran ONE
+  // wave per SIMD (110 KB LDS, 256 threads) — every softmax/LDS stall
+  // fully exposed, ~100 TF effective.  8 waves + the smaller strip pool
+  // give 3 blocks/CU at D=64 (6 waves/SIMD).
+  constexpr int RF = 1;
+  constexpr int QBLK = 8 * 16 * RF;
   const int qb0 = blockIdx.x * QBLK;
   const int bh = blockIdx.y;
   const long base = (long)bh * S * D;
@@ -51,6 +53,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   constexpr int TILE_VT = KBP * D;
   bf16* smem_b = reinterpret_cast<bf16*>(smem);
   bf16* p_lds = smem_b + 2 * (TILE_K + TILE_VT) + wave * 16 * KB;
+  constexpr int NTHR = 8 * WAVE;   // 512
 
   const int fr = lane & 15;        // fragment row/col index
   const int fg = lane >> 4;        // fragment k-group (8 contiguous)
@@ -95,13 +98,13 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
 
   const int kv_end = causal ? min(S, qb0 + QBLK) : S;
   const int n_tiles = (kv_end + KB - 1) / KB;
-  constexpr int PF = TILE_K / (256 * 8);     // 16B vectors per thread
+  constexpr int PF = TILE_K / (NTHR * 8);    // 16B vectors per thread
   bf16x8 kreg[PF], vreg[PF];
 
   auto load_tile = [&](int t) {
     #pragma unroll
     for (int pi = 0; pi < PF; ++pi) {
-      const int e = threadIdx.x * 8 + pi * 2048;
+      const int e = threadIdx.x * 8 + pi * (NTHR * 8);
       kreg[pi] = *reinterpret_cast<const bf16x8*>(&k[(long)t * TILE_K + e]);
       vreg[pi] = *reinterpret_cast<const bf16x8*>(&v[(long)t * TILE_K + e]);
     }
@@ -111,7 +114,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     bf16* vb = kb + TILE_K;
     #pragma unroll
     for (int pi = 0; pi < PF; ++pi) {
-      const int e = threadIdx.x * 8 + pi * 2048;
+      const int e = threadIdx.x * 8 + pi * (NTHR * 8);
       *reinterpret_cast<bf16x8*>(&kb[e]) = kreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
@@ -268,8 +271,8 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
   auto out = at::empty_like(q);
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
-  dim3 grid((S + 127) / 128, B * H), block(256);
-  size_t lds = (2 * ((size_t)KB * D + (size_t)KBP * D) + 4 * 16 * KB) * 2;
+  dim3 grid((S + 127) / 128, B * H), block(512);
+  size_t lds = (2 * ((size_t)KB * D + (size_t)KBP * D) + 8 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
   if (D == 64)
     hipLaunchKernelGGL(flash_fwd_kernel<64>, grid, block, lds, stream,
@@ -350,19 +353,36 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   for (int j = 0; j < D / 16; ++j) dq_acc[j] = {0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = causal ? min(S, qb0 + QB) : S;
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
-    __syncthreads();
-    for (int e = threadIdx.x * 8; e < KB * D; e += 256 * 8) {
-      bf16x8 kk = *reinterpret_cast<const bf16x8*>(&k[(long)kv0 * D + e]);
-      *reinterpret_cast<bf16x8*>(&k_lds[e]) = kk;
-      *reinterpret_cast<bf16x8*>(&v_lds[e]) =
-          *reinterpret_cast<const bf16x8*>(&v[(long)kv0 * D + e]);
+  // register-staged K/V prefetch: tile kv0+KB streams from HBM while the
+  // MFMA/softmax work on kv0 runs (the unprefetched form exposed the
+  // full HBM latency behind two barriers every tile)
+  constexpr int NV = KB * D / (256 * 8);
+  bf16x8 kreg[NV], vreg[NV];
+  auto kv_load = [&](int t0) {
+    #pragma unroll
+    for (int pi = 0; pi < NV; ++pi) {
+      const int e = threadIdx.x * 8 + pi * 2048;
+      kreg[pi] = *reinterpret_cast<const bf16x8*>(&k[(long)t0 * D + e]);
+      vreg[pi] = *reinterpret_cast<const bf16x8*>(&v[(long)t0 * D + e]);
+    }
+  };
+  auto kv_store = [&]() {
+    #pragma unroll
+    for (int pi = 0; pi < NV; ++pi) {
+      const int e = threadIdx.x * 8 + pi * 2048;
+      *reinterpret_cast<bf16x8*>(&k_lds[e]) = kreg[pi];
+      *reinterpret_cast<bf16x8*>(&v_lds[e]) = vreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i)
-        kt_lds[(col + i) * KBP + row] = kk.v[i];
+        kt_lds[(col + i) * KBP + row] = kreg[pi].v[i];
     }
+  };
+  kv_load(0);
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    kv_store();
     __syncthreads();
+    if (kv0 + KB < kv_end) kv_load(kv0 + KB);
 
     // S = Q K^T and dP = dO V^T (both mfma-native: B = rows)
     f32x4 s_acc[KB / 16], dp_acc[KB / 16];
@@ -433,6 +453,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                                                             0);
       }
     }
+    __syncthreads();   // all reads of this tile done before re-staging
   }
   #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -494,21 +515,36 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   }
 
   const int q_start = causal ? kb0 : 0;
-  for (int q0 = q_start; q0 < S; q0 += QB) {
-    __syncthreads();
-    for (int e = threadIdx.x * 8; e < QB * D; e += 256 * 8) {
-      bf16x8 qq = *reinterpret_cast<const bf16x8*>(&q[(long)q0 * D + e]);
-      bf16x8 dd = *reinterpret_cast<const bf16x8*>(&dO_[(long)q0 * D + e]);
-      *reinterpret_cast<bf16x8*>(&q_lds[e]) = qq;
-      *reinterpret_cast<bf16x8*>(&do_lds[e]) = dd;
+  // register-staged Q/dO prefetch (see dq kernel note)
+  constexpr int NV = QB * D / (256 * 8);
+  bf16x8 qreg[NV], dreg[NV];
+  auto q_load = [&](int t0) {
+    #pragma unroll
+    for (int pi = 0; pi < NV; ++pi) {
+      const int e = threadIdx.x * 8 + pi * 2048;
+      qreg[pi] = *reinterpret_cast<const bf16x8*>(&q[(long)t0 * D + e]);
+      dreg[pi] = *reinterpret_cast<const bf16x8*>(&dO_[(long)t0 * D + e]);
+    }
+  };
+  auto q_store = [&]() {
+    #pragma unroll
+    for (int pi = 0; pi < NV; ++pi) {
+      const int e = threadIdx.x * 8 + pi * 2048;
+      *reinterpret_cast<bf16x8*>(&q_lds[e]) = qreg[pi];
+      *reinterpret_cast<bf16x8*>(&do_lds[e]) = dreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        qt_lds[(col + i) * QBP + row] = qq.v[i];
-        dot_lds[(col + i) * QBP + row] = dd.v[i];
+        qt_lds[(col + i) * QBP + row] = qreg[pi].v[i];
+        dot_lds[(col + i) * QBP + row] = dreg[pi].v[i];
       }
     }
+  };
+  q_load(q_start);
+  for (int q0 = q_start; q0 < S; q0 += QB) {
+    q_store();
     __syncthreads();
+    if (q0 + QB < S) q_load(q0 + QB);
 
     // S^T = K Q^T and dP^T = V dO^T (B operands: row reads from LDS)
     f32x4 st_acc[QB / 16], dpt_acc[QB / 16];
@@ -609,6 +645,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                                                             0);
       }
     }
+    __syncthreads();   // all reads of this tile done before re-staging
   }
   #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -622,6 +659,31 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   }
 }
 
+// delta = rowsum(dO * O) in ONE bf16 pass (the aten composite spelled
+// dO.float() * O.float() then sum(-1): three full fp32 materializations
+// of activation-sized tensors per backward).
+template <int D>
+__global__ void __launch_bounds__(256)
+attn_delta_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ O,
+                  float* __restrict__ delta, long rows) {
+  constexpr int LPR = D / 8;            // lanes per row (16B each)
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  constexpr int RPW = WAVE / LPR;       // rows per wave
+  long row = (long)blockIdx.x * (4 * RPW) + wave * RPW + lane / LPR;
+  if (row >= rows) return;
+  const int c0 = (lane % LPR) * 8;
+  const bf16x8 d8 = *reinterpret_cast<const bf16x8*>(&dO[row * D + c0]);
+  const bf16x8 o8 = *reinterpret_cast<const bf16x8*>(&O[row * D + c0]);
+  float acc = 0.f;
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) acc += bf2f(d8.v[i]) * bf2f(o8.v[i]);
+  #pragma unroll
+  for (int off = LPR / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  if (lane % LPR == 0) delta[row] = acc;
+}
+
 std::tuple<at::Tensor, at::Tensor, at::Tensor>
 flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
                const at::Tensor& k, const at::Tensor& v,
@@ -633,8 +695,22 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
   TORCH_CHECK(D == 64 || D == 128, "flash_attn_bwd: D in {64,128}");
   TORCH_CHECK(S % QB == 0, "flash_attn_bwd: S multiple of 64");
   auto gradc = grad.contiguous();
-  // delta = rowsum(dO * O), fp32
-  auto delta = (gradc.to(at::kFloat) * out.to(at::kFloat)).sum(-1);
+  // delta = rowsum(dO * O), fp32 — one fused bf16 pass
+  auto delta = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  {
+    long rows = (long)B * H * S;
+    int rpw = (D == 64) ? 8 : 4;
+    long nblk = (rows + 4 * rpw - 1) / (4 * rpw);
+    auto stream0 = at::cuda::getCurrentCUDAStream();
+    if (D == 64)
+      hipLaunchKernelGGL(attn_delta_kernel<64>, dim3(nblk), dim3(256), 0,
+          stream0, (const bf16*)gradc.data_ptr(), (const bf16*)out.data_ptr(),
+          delta.data_ptr<float>(), rows);
+    else
+      hipLaunchKernelGGL(attn_delta_kernel<128>, dim3(nblk), dim3(256), 0,
+          stream0, (const bf16*)gradc.data_ptr(), (const bf16*)out.data_ptr(),
+          delta.data_ptr<float>(), rows);
+  }
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
