@@ -22,6 +22,14 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
                       // collapses onto 2 banks -> 8-way conflicts)
 #define QBP (QB + 8)
 
+// Bank-conflict XOR swizzle for LINEAR [rows][64 or 128] bf16 LDS tiles
+// (element-index form): a ds_read_b128 fragment read walks 16 rows at a
+// 128/256-B stride, collapsing onto 2 (or 1) 16-B bank slots — 8/16-way
+// conflicts (guide T2).  Folding row bits into the slot bits
+// (e ^= ((e>>7)&15)<<3) is an involution, keeps 8-element chunks whole,
+// and is injective across the 16 rows of a fragment for both row widths.
+DEVINL int lsw(int e) { return e ^ (((e >> 7) & 15) << 3); }
+
 template <int D>
 __global__ void __launch_bounds__(512)
 flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
@@ -115,7 +123,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     #pragma unroll
     for (int pi = 0; pi < PF; ++pi) {
       const int e = threadIdx.x * 8 + pi * (NTHR * 8);
-      *reinterpret_cast<bf16x8*>(&kb[e]) = kreg[pi];
+      *reinterpret_cast<bf16x8*>(&kb[lsw(e)]) = kreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i)
@@ -144,7 +152,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         #pragma unroll
         for (int ks = 0; ks < D / 32; ++ks) {
           bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
-              &k_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+              &k_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
           s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[rf][ks], kf,
                                                              s_acc[j], 0, 0,
                                                              0);
@@ -214,13 +222,13 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       for (int j = 0; j < KB / 16; ++j)
         #pragma unroll
         for (int r = 0; r < 4; ++r)
-          p_lds[(4 * fg + r) * KB + j * 16 + fr] = f2bf(s_acc[j][r]);
+          p_lds[lsw((4 * fg + r) * KB + j * 16 + fr)] = f2bf(s_acc[j][r]);
       lds_fence();
       bf16x8v pf[KB / 32];
       #pragma unroll
       for (int ks = 0; ks < KB / 32; ++ks)
         pf[ks] = *reinterpret_cast<const bf16x8v*>(
-            &p_lds[fr * KB + ks * 32 + fg * 8]);
+            &p_lds[lsw(fr * KB + ks * 32 + fg * 8)]);
       lds_fence();   // strip is reused by the next fragment
       #pragma unroll
       for (int j = 0; j < D / 16; ++j) {
@@ -370,8 +378,8 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     #pragma unroll
     for (int pi = 0; pi < NV; ++pi) {
       const int e = threadIdx.x * 8 + pi * 2048;
-      *reinterpret_cast<bf16x8*>(&k_lds[e]) = kreg[pi];
-      *reinterpret_cast<bf16x8*>(&v_lds[e]) = vreg[pi];
+      *reinterpret_cast<bf16x8*>(&k_lds[lsw(e)]) = kreg[pi];
+      *reinterpret_cast<bf16x8*>(&v_lds[lsw(e)]) = vreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i)
@@ -393,9 +401,9 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < D / 32; ++ks) {
         bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
-            &k_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+            &k_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
-            &v_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+            &v_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], kf,
                                                            s_acc[j], 0, 0, 0);
         dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[ks], vf,
@@ -434,13 +442,13 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     for (int j = 0; j < KB / 16; ++j)
       #pragma unroll
       for (int r = 0; r < 4; ++r)
-        s_lds[(4 * fg + r) * KB + j * 16 + fr] = f2bf(s_acc[j][r]);
+        s_lds[lsw((4 * fg + r) * KB + j * 16 + fr)] = f2bf(s_acc[j][r]);
     lds_fence();
     bf16x8v dsf[KB / 32];
     #pragma unroll
     for (int ks = 0; ks < KB / 32; ++ks)
       dsf[ks] = *reinterpret_cast<const bf16x8v*>(
-          &s_lds[fr * KB + ks * 32 + fg * 8]);
+          &s_lds[lsw(fr * KB + ks * 32 + fg * 8)]);
     // dq += dS @ K: B-operand from K^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
@@ -530,8 +538,8 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     #pragma unroll
     for (int pi = 0; pi < NV; ++pi) {
       const int e = threadIdx.x * 8 + pi * 2048;
-      *reinterpret_cast<bf16x8*>(&q_lds[e]) = qreg[pi];
-      *reinterpret_cast<bf16x8*>(&do_lds[e]) = dreg[pi];
+      *reinterpret_cast<bf16x8*>(&q_lds[lsw(e)]) = qreg[pi];
+      *reinterpret_cast<bf16x8*>(&do_lds[lsw(e)]) = dreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i) {
@@ -555,9 +563,9 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < D / 32; ++ks) {
         bf16x8v qfb = *reinterpret_cast<const bf16x8v*>(
-            &q_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+            &q_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         bf16x8v dob = *reinterpret_cast<const bf16x8v*>(
-            &do_lds[(j * 16 + fr) * D + ks * 32 + fg * 8]);
+            &do_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[ks], qfb,
                                                             st_acc[j], 0, 0,
                                                             0);
@@ -582,7 +590,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
           if (!(causal && krow > qcol) && qcol < S)
             p = __expf(st_acc[j][r] * scale - l);
           st_acc[j][r] = p;
-          s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(p);
+          s_lds[lsw((4 * fg + r) * QB + j * 16 + fr)] = f2bf(p);
         }
       }
     } else {
@@ -593,7 +601,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
         for (int r = 0; r < 4; ++r) {
           float p = __expf(st_acc[j][r] * scale - l);
           st_acc[j][r] = p;
-          s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(p);
+          s_lds[lsw((4 * fg + r) * QB + j * 16 + fr)] = f2bf(p);
         }
       }
     }
@@ -602,7 +610,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     #pragma unroll
     for (int ks = 0; ks < QB / 32; ++ks)
       ptf[ks] = *reinterpret_cast<const bf16x8v*>(
-          &s_lds[fr * QB + ks * 32 + fg * 8]);
+          &s_lds[lsw(fr * QB + ks * 32 + fg * 8)]);
     // dV += P^T @ dO: B-operand from dO^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
@@ -624,7 +632,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float ds = st_acc[j][r] * (dpt_acc[j][r] - dlt) * scale;
-        s_lds[(4 * fg + r) * QB + j * 16 + fr] = f2bf(ds);
+        s_lds[lsw((4 * fg + r) * QB + j * 16 + fr)] = f2bf(ds);
       }
     }
     lds_fence();
@@ -632,7 +640,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     #pragma unroll
     for (int ks = 0; ks < QB / 32; ++ks)
       dstf[ks] = *reinterpret_cast<const bf16x8v*>(
-          &s_lds[fr * QB + ks * 32 + fg * 8]);
+          &s_lds[lsw(fr * QB + ks * 32 + fg * 8)]);
     // dK += dS^T @ Q: B-operand from Q^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
