@@ -55,10 +55,10 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   const int lane = threadIdx.x % WAVE;
   const int wave = threadIdx.x / WAVE;
 
-  // LDS: DOUBLE-BUFFERED K [KB][D] + V^T [D][KBP] tile pairs + P strips.
+  // LDS: DOUBLE-BUFFERED K [KB][D] + V^T [D][KB] (lsw) pairs + P strips.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr int TILE_K = KB * D;
-  constexpr int TILE_VT = KBP * D;
+  constexpr int TILE_VT = KB * D;   // [D][KB] linear + lsw swizzle
   bf16* smem_b = reinterpret_cast<bf16*>(smem);
   bf16* p_lds = smem_b + 2 * (TILE_K + TILE_VT) + wave * 16 * KB;
   constexpr int NTHR = 8 * WAVE;   // 512
@@ -127,7 +127,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i)
-        vb[(col + i) * KBP + row] = vreg[pi].v[i];
+        vb[lsw((col + i) * KB + row)] = vreg[pi].v[i];
     }
   };
 
@@ -235,7 +235,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         #pragma unroll
         for (int ks = 0; ks < KB / 32; ++ks) {
           bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
-              &vt_lds[(j * 16 + fr) * KBP + ks * 32 + fg * 8]);
+              &vt_lds[lsw((j * 16 + fr) * KB + ks * 32 + fg * 8)]);
           o_acc[rf][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               pf[ks], vf, o_acc[rf][j], 0, 0, 0);
         }
@@ -280,7 +280,7 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid((S + 127) / 128, B * H), block(512);
-  size_t lds = (2 * ((size_t)KB * D + (size_t)KBP * D) + 8 * 16 * KB) * 2;
+  size_t lds = (2 * ((size_t)KB * D + (size_t)KB * D) + 8 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
   if (D == 64)
     hipLaunchKernelGGL(flash_fwd_kernel<64>, grid, block, lds, stream,
@@ -335,8 +335,8 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* k_lds = reinterpret_cast<bf16*>(smem);
   bf16* v_lds = k_lds + KB * D;
-  bf16* kt_lds = v_lds + KB * D;                   // K^T [D][KBP]
-  bf16* s_lds = kt_lds + KBP * D + wave * 16 * KB;  // wave-private strip
+  bf16* kt_lds = v_lds + KB * D;                   // K^T [D][KB] + lsw
+  bf16* s_lds = kt_lds + KB * D + wave * 16 * KB;   // wave-private strip
 
   // A-operand fragments for this wave's 16 q rows
   bf16x8v qf[D / 32], dof[D / 32];
@@ -383,7 +383,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i)
-        kt_lds[(col + i) * KBP + row] = kreg[pi].v[i];
+        kt_lds[lsw((col + i) * KB + row)] = kreg[pi].v[i];
     }
   };
   kv_load(0);
@@ -455,7 +455,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < KB / 32; ++ks) {
         bf16x8v kcol = *reinterpret_cast<const bf16x8v*>(
-            &kt_lds[(j * 16 + fr) * KBP + ks * 32 + fg * 8]);
+            &kt_lds[lsw((j * 16 + fr) * KB + ks * 32 + fg * 8)]);
         dq_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf[ks], kcol,
                                                             dq_acc[j], 0, 0,
                                                             0);
@@ -502,9 +502,9 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* q_lds = reinterpret_cast<bf16*>(smem);
   bf16* do_lds = q_lds + QB * D;
-  bf16* qt_lds = do_lds + QB * D;                  // Q^T [D][QBP]
-  bf16* dot_lds = qt_lds + QBP * D;                // dO^T [D][QBP]
-  bf16* s_lds = dot_lds + QBP * D + wave * 16 * QB;
+  bf16* qt_lds = do_lds + QB * D;                  // Q^T [D][QB] + lsw
+  bf16* dot_lds = qt_lds + QB * D;                 // dO^T [D][QB] + lsw
+  bf16* s_lds = dot_lds + QB * D + wave * 16 * QB;
 
   bf16x8v kf[D / 32], vf[D / 32];
   #pragma unroll
@@ -543,8 +543,8 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        qt_lds[(col + i) * QBP + row] = qreg[pi].v[i];
-        dot_lds[(col + i) * QBP + row] = dreg[pi].v[i];
+        qt_lds[lsw((col + i) * QB + row)] = qreg[pi].v[i];
+        dot_lds[lsw((col + i) * QB + row)] = dreg[pi].v[i];
       }
     }
   };
@@ -617,7 +617,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < QB / 32; ++ks) {
         bf16x8v docol = *reinterpret_cast<const bf16x8v*>(
-            &dot_lds[(j * 16 + fr) * QBP + ks * 32 + fg * 8]);
+            &dot_lds[lsw((j * 16 + fr) * QB + ks * 32 + fg * 8)]);
         dv_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf[ks], docol,
                                                             dv_acc[j], 0, 0,
                                                             0);
@@ -647,7 +647,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < QB / 32; ++ks) {
         bf16x8v qcolf = *reinterpret_cast<const bf16x8v*>(
-            &qt_lds[(j * 16 + fr) * QBP + ks * 32 + fg * 8]);
+            &qt_lds[lsw((j * 16 + fr) * QB + ks * 32 + fg * 8)]);
         dk_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf[ks], qcolf,
                                                             dk_acc[j], 0, 0,
                                                             0);
@@ -724,8 +724,8 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
   auto dv = at::empty_like(v);
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid(S / QB, B * H), block(256);
-  size_t lds = (2 * KB * (size_t)D + (size_t)KBP * D + 4 * 16 * KB) * 2;
-  size_t lds_kv = (2 * QB * (size_t)D + 2 * (size_t)QBP * D
+  size_t lds = (2 * KB * (size_t)D + (size_t)KB * D + 4 * 16 * KB) * 2;
+  size_t lds_kv = (2 * QB * (size_t)D + 2 * (size_t)QB * D
                    + 4 * 16 * QB) * 2;
   float scale = 1.f / sqrtf((float)D);
   if (D == 64) {
