@@ -65,7 +65,7 @@ def _kernel_ok(q, k, v):
     arbitrarily sharded shapes — anything outside goes to the math path."""
     return (q.dtype == torch.bfloat16 and q.dim() == 4
             and q.shape == k.shape and q.shape == v.shape
-            and q.shape[-1] in (64, 128) and q.shape[2] % 64 == 0)
+            and q.shape[-1] in (64, 128) and q.shape[2] % 128 == 0)
 
 
 def _fwd_cuda(q, k, v, causal):

@@ -309,13 +309,15 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
 // ---------------------------------------------------------------------------
 
 template <int D>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                     const bf16* __restrict__ K, const bf16* __restrict__ V,
                     const float* __restrict__ LSE,
                     const float* __restrict__ DELTA, bf16* __restrict__ DQ,
                     int B, int H, int S, bool causal, float scale) {
-  const int qb0 = blockIdx.x * QB;
+  // 8 waves x 16 q rows: two 64-row halves share each staged K/V tile
+  // (2x arithmetic intensity vs the 4-wave form) at 6+ waves/SIMD.
+  const int qb0 = blockIdx.x * (2 * QB);
   const int bh = blockIdx.y;
   const long base = (long)bh * S * D;
   const bf16* dO_ = dO + base;
@@ -360,16 +362,16 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   #pragma unroll
   for (int j = 0; j < D / 16; ++j) dq_acc[j] = {0.f, 0.f, 0.f, 0.f};
 
-  const int kv_end = causal ? min(S, qb0 + QB) : S;
+  const int kv_end = causal ? min(S, qb0 + 2 * QB) : S;
   // register-staged K/V prefetch: tile kv0+KB streams from HBM while the
   // MFMA/softmax work on kv0 runs (the unprefetched form exposed the
   // full HBM latency behind two barriers every tile)
-  constexpr int NV = KB * D / (256 * 8);
+  constexpr int NV = KB * D / (512 * 8);
   bf16x8 kreg[NV], vreg[NV];
   auto kv_load = [&](int t0) {
     #pragma unroll
     for (int pi = 0; pi < NV; ++pi) {
-      const int e = threadIdx.x * 8 + pi * 2048;
+      const int e = threadIdx.x * 8 + pi * 4096;
       kreg[pi] = *reinterpret_cast<const bf16x8*>(&k[(long)t0 * D + e]);
       vreg[pi] = *reinterpret_cast<const bf16x8*>(&v[(long)t0 * D + e]);
     }
@@ -377,7 +379,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   auto kv_store = [&]() {
     #pragma unroll
     for (int pi = 0; pi < NV; ++pi) {
-      const int e = threadIdx.x * 8 + pi * 2048;
+      const int e = threadIdx.x * 8 + pi * 4096;
       *reinterpret_cast<bf16x8*>(&k_lds[lsw(e)]) = kreg[pi];
       *reinterpret_cast<bf16x8*>(&v_lds[lsw(e)]) = vreg[pi];
       const int row = e / D, col = e % D;
@@ -474,14 +476,15 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
 }
 
 template <int D>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                      const bf16* __restrict__ K, const bf16* __restrict__ V,
                      const float* __restrict__ LSE,
                      const float* __restrict__ DELTA, bf16* __restrict__ DK,
                      bf16* __restrict__ DV, int B, int H, int S, bool causal,
                      float scale) {
-  const int kb0 = blockIdx.x * KB;
+  // 8 waves x 16 key rows share each staged Q/dO tile (see dq note)
+  const int kb0 = blockIdx.x * (2 * KB);
   const int bh = blockIdx.y;
   const long base = (long)bh * S * D;
   const bf16* dO_ = dO + base;
@@ -524,12 +527,12 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
 
   const int q_start = causal ? kb0 : 0;
   // register-staged Q/dO prefetch (see dq kernel note)
-  constexpr int NV = QB * D / (256 * 8);
+  constexpr int NV = QB * D / (512 * 8);
   bf16x8 qreg[NV], dreg[NV];
   auto q_load = [&](int t0) {
     #pragma unroll
     for (int pi = 0; pi < NV; ++pi) {
-      const int e = threadIdx.x * 8 + pi * 2048;
+      const int e = threadIdx.x * 8 + pi * 4096;
       qreg[pi] = *reinterpret_cast<const bf16x8*>(&q[(long)t0 * D + e]);
       dreg[pi] = *reinterpret_cast<const bf16x8*>(&dO_[(long)t0 * D + e]);
     }
@@ -537,7 +540,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   auto q_store = [&]() {
     #pragma unroll
     for (int pi = 0; pi < NV; ++pi) {
-      const int e = threadIdx.x * 8 + pi * 2048;
+      const int e = threadIdx.x * 8 + pi * 4096;
       *reinterpret_cast<bf16x8*>(&q_lds[lsw(e)]) = qreg[pi];
       *reinterpret_cast<bf16x8*>(&do_lds[lsw(e)]) = dreg[pi];
       const int row = e / D, col = e % D;
@@ -701,7 +704,7 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
   TORCH_CHECK(q.sizes() == k.sizes() && q.sizes() == v.sizes());
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(D == 64 || D == 128, "flash_attn_bwd: D in {64,128}");
-  TORCH_CHECK(S % QB == 0, "flash_attn_bwd: S multiple of 64");
+  TORCH_CHECK(S % (2 * QB) == 0, "flash_attn_bwd: S multiple of 128");
   auto gradc = grad.contiguous();
   // delta = rowsum(dO * O), fp32 — one fused bf16 pass
   auto delta = at::empty({B, H, S}, q.options().dtype(at::kFloat));
@@ -723,10 +726,10 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
   auto stream = at::cuda::getCurrentCUDAStream();
-  dim3 grid(S / QB, B * H), block(256);
-  size_t lds = (2 * KB * (size_t)D + (size_t)KB * D + 4 * 16 * KB) * 2;
+  dim3 grid(S / (2 * QB), B * H), block(512);
+  size_t lds = (2 * KB * (size_t)D + (size_t)KB * D + 8 * 16 * KB) * 2;
   size_t lds_kv = (2 * QB * (size_t)D + 2 * (size_t)QB * D
-                   + 4 * 16 * QB) * 2;
+                   + 8 * 16 * QB) * 2;
   float scale = 1.f / sqrtf((float)D);
   if (D == 64) {
     hipLaunchKernelGGL(flash_bwd_dq_kernel<64>, grid, block, lds, stream,
