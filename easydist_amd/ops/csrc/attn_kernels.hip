@@ -142,10 +142,15 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     if (t + 1 < n_tiles)
       load_tile(t + 1);          // global loads overlap the MFMA loop
 
+    // a 128-row block spans 8 waves: tiles strictly above this wave's
+    // causal diagonal are all-masked — skip the compute (NOT the
+    // barriers: every wave still arrives at __syncthreads)
+    const bool wave_active = !(causal && kv0 > qr0[0] + 15);
     #pragma unroll
-    for (int rf = 0; rf < RF; ++rf) {
+    for (int rf = 0; rf < RF && wave_active; ++rf) {
       // ---- S = Q K^T ----------------------------------------------------
       f32x4 s_acc[KB / 16];
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int j = 0; j < KB / 16; ++j) {
         s_acc[j] = {0.f, 0.f, 0.f, 0.f};
@@ -158,6 +163,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                                                              0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
 
       // ---- online softmax ----------------------------------------------
       // interior tiles (fully below the causal diagonal, fully in-range)
@@ -230,6 +236,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         pf[ks] = *reinterpret_cast<const bf16x8v*>(
             &p_lds[lsw(fr * KB + ks * 32 + fg * 8)]);
       lds_fence();   // strip is reused by the next fragment
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int j = 0; j < D / 16; ++j) {
         #pragma unroll
@@ -240,6 +247,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
               pf[ks], vf, o_acc[rf][j], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     if (t + 1 < n_tiles)
@@ -393,6 +401,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     kv_store();
     __syncthreads();
     if (kv0 + KB < kv_end) kv_load(kv0 + KB);
+    if (causal && kv0 > qr0 + 15) { __syncthreads(); continue; }
 
     // S = Q K^T and dP = dO V^T (both mfma-native: B = rows)
     f32x4 s_acc[KB / 16], dp_acc[KB / 16];
@@ -556,6 +565,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     q_store();
     __syncthreads();
     if (q0 + QB < S) q_load(q0 + QB);
+    if (causal && q0 + QB - 1 < kr0) { __syncthreads(); continue; }
 
     // S^T = K Q^T and dP^T = V dO^T (B operands: row reads from LDS)
     f32x4 st_acc[QB / 16], dpt_acc[QB / 16];
