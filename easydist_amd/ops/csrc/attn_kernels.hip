@@ -31,7 +31,7 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 DEVINL int lsw(int e) { return e ^ (((e >> 7) & 15) << 3); }
 
 template <int D>
-__global__ void __launch_bounds__(512)
+__global__ void __launch_bounds__(512, (D == 64 ? 4 : 3))  // VGPR cap: 128/170
 flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                  const bf16* __restrict__ V, bf16* __restrict__ O,
                  float* __restrict__ LSE, int B, int H, int S, bool causal,
@@ -146,8 +146,9 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     // causal diagonal are all-masked — skip the compute (NOT the
     // barriers: every wave still arrives at __syncthreads)
     const bool wave_active = !(causal && kv0 > qr0[0] + 15);
+    if (wave_active) {
     #pragma unroll
-    for (int rf = 0; rf < RF && wave_active; ++rf) {
+    for (int rf = 0; rf < RF; ++rf) {
       // ---- S = Q K^T ----------------------------------------------------
       f32x4 s_acc[KB / 16];
       __builtin_amdgcn_s_setprio(1);
@@ -248,6 +249,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         }
       }
       __builtin_amdgcn_s_setprio(0);
+    }
     }
 
     if (t + 1 < n_tiles)
