@@ -31,7 +31,7 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 DEVINL int lsw(int e) { return e ^ (((e >> 7) & 15) << 3); }
 
 template <int D>
-__global__ void __launch_bounds__(512, (D == 64 ? 4 : 3))  // VGPR cap: 128/170
+__global__ void __launch_bounds__(512, (D == 64 ? 4 : 1))  // VGPR cap at D=64 only
 flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                  const bf16* __restrict__ V, bf16* __restrict__ O,
                  float* __restrict__ LSE, int B, int H, int S, bool causal,
