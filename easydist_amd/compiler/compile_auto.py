@@ -77,8 +77,11 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     comm_optimize(gm)
 
     # ---- 5b2. lower hot aten ops to the gfx950 kernels -------------------
-    from .passes.lower_hip import lower_gemm, lower_layer_norm
+    from .passes.lower_hip import (lower_cross_entropy, lower_gemm,
+                                   lower_layer_norm, lower_sdpa)
     lower_layer_norm(gm)
+    lower_sdpa(gm)
+    lower_cross_entropy(gm)
     if mdconfig.hip_gemm:
         lower_gemm(gm)
 

@@ -13,6 +13,7 @@ from . import load_extension
 
 lib = torch.library.Library("easydist_amd", "FRAGMENT")
 lib.define("ce_fwd(Tensor logits, Tensor targets) -> (Tensor, Tensor)")
+lib.define("ce_fwd_rows(Tensor logits, Tensor targets) -> (Tensor, Tensor)")
 lib.define("ce_bwd(Tensor grad, Tensor logits, Tensor targets, Tensor lse) "
            "-> Tensor")
 
@@ -24,6 +25,20 @@ def _ce_fwd_aten(logits, targets):
     # SUM (not mean): under row sharding the output is PARTIAL(sum), which
     # ShardCombine discovers and the solver reshards with one all_reduce
     return nll.sum(), lse
+
+
+def _ce_fwd_rows_aten(logits, targets):
+    lf = logits.float()
+    lse = torch.logsumexp(lf, dim=-1)
+    nll = lse - lf.gather(1, targets.unsqueeze(1)).squeeze(1)
+    return nll, lse
+
+
+def _ce_fwd_rows_cuda(logits, targets):
+    ext = load_extension()
+    if ext is not None and logits.dim() == 2:
+        return ext.ce_fwd_rows(logits.contiguous(), targets.contiguous())
+    return _ce_fwd_rows_aten(logits, targets)
 
 
 def _ce_fwd_cuda(logits, targets):
@@ -39,7 +54,8 @@ def _ce_bwd_aten(grad, logits, targets, lse):
     p.scatter_add_(1, targets.unsqueeze(1),
                    torch.full_like(targets.unsqueeze(1), -1.0,
                                    dtype=p.dtype))
-    return (p * grad).to(logits.dtype)
+    g = grad if grad.dim() == 0 else grad.reshape(-1, 1)   # per-row grads
+    return (p * g).to(logits.dtype)
 
 
 def _ce_bwd_cuda(grad, logits, targets, lse):
@@ -51,6 +67,8 @@ def _ce_bwd_cuda(grad, logits, targets, lse):
 
 lib.impl("ce_fwd", _ce_fwd_aten, "CPU")
 lib.impl("ce_fwd", _ce_fwd_cuda, "CUDA")
+lib.impl("ce_fwd_rows", _ce_fwd_rows_aten, "CPU")
+lib.impl("ce_fwd_rows", _ce_fwd_rows_cuda, "CUDA")
 lib.impl("ce_bwd", _ce_bwd_aten, "CPU")
 lib.impl("ce_bwd", _ce_bwd_cuda, "CUDA")
 
@@ -58,6 +76,12 @@ lib.impl("ce_bwd", _ce_bwd_cuda, "CUDA")
 @torch.library.register_fake("easydist_amd::ce_fwd")
 def _ce_fwd_fake(logits, targets):
     return (logits.new_empty((), dtype=torch.float32),
+            logits.new_empty((logits.shape[0],), dtype=torch.float32))
+
+
+@torch.library.register_fake("easydist_amd::ce_fwd_rows")
+def _ce_fwd_rows_fake(logits, targets):
+    return (logits.new_empty((logits.shape[0],), dtype=torch.float32),
             logits.new_empty((logits.shape[0],), dtype=torch.float32))
 
 

@@ -16,6 +16,8 @@ rms_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
              const at::Tensor& rstd, const std::optional<at::Tensor>& w);
 std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
                                           const at::Tensor& targets);
+std::tuple<at::Tensor, at::Tensor> ce_fwd_rows(const at::Tensor& logits,
+                                               const at::Tensor& targets);
 at::Tensor ce_bwd(const at::Tensor& grad, const at::Tensor& logits,
                   const at::Tensor& targets, const at::Tensor& lse);
 std::tuple<std::vector<at::Tensor>, std::vector<at::Tensor>,
@@ -78,6 +80,7 @@ TORCH_LIBRARY(easydist_amd_hip, m) {
   m.def("rms_norm_bwd(Tensor grad, Tensor x, Tensor rstd, Tensor? w) "
         "-> (Tensor, Tensor)");
   m.def("ce_fwd(Tensor logits, Tensor targets) -> (Tensor, Tensor)");
+  m.def("ce_fwd_rows(Tensor logits, Tensor targets) -> (Tensor, Tensor)");
   m.def("ce_bwd(Tensor grad, Tensor logits, Tensor targets, Tensor lse) "
         "-> Tensor");
   m.def("fused_adam_step(Tensor[] params, Tensor[] grads, Tensor[] exp_avgs, "
@@ -101,6 +104,7 @@ TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
   m.impl("rms_norm_fwd", rms_norm_fwd);
   m.impl("rms_norm_bwd", rms_norm_bwd);
   m.impl("ce_fwd", ce_fwd);
+  m.impl("ce_fwd_rows", ce_fwd_rows);
   m.impl("ce_bwd", ce_bwd);
   m.impl("fused_adam_step", adam_wrap);
   m.impl("fused_sgd_step", sgd_wrap);

@@ -77,7 +77,7 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
   }
 }
 
-template <typename T>
+template <typename T, bool PER_ROW>
 __global__ void ce_bwd_kernel(const float* __restrict__ grad_scalar,
                               const T* __restrict__ logits,
                               const long* __restrict__ targets,
@@ -89,7 +89,8 @@ __global__ void ce_bwd_kernel(const float* __restrict__ grad_scalar,
   if (row >= N) return;
   const T* lrow = logits + (long)row * V;
   T* drow = dlogits + (long)row * V;
-  const float scale = grad_scalar[0];   // SUM semantics: caller scales
+  // SUM semantics (scalar) or per-row grad (the lowered nll_loss path)
+  const float scale = PER_ROW ? grad_scalar[row] : grad_scalar[0];
   const float l = lse[row];
   const long tg = targets[row];
   const int V8 = (V / 8) * 8;
@@ -114,8 +115,8 @@ __global__ void ce_bwd_kernel(const float* __restrict__ grad_scalar,
   }
 }
 
-std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
-                                          const at::Tensor& targets) {
+std::tuple<at::Tensor, at::Tensor> ce_fwd_rows(const at::Tensor& logits,
+                                               const at::Tensor& targets) {
   TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
   TORCH_CHECK(targets.scalar_type() == at::kLong);
   TORCH_CHECK(targets.numel() == logits.size(0), "targets/rows mismatch");
@@ -134,6 +135,12 @@ std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
         logits.data_ptr<float>(), targets.data_ptr<long>(),
         nll.data_ptr<float>(), lse.data_ptr<float>(), N, V);
   }
+  return {nll, lse};
+}
+
+std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
+                                          const at::Tensor& targets) {
+  auto [nll, lse] = ce_fwd_rows(logits, targets);
   return {nll.sum(), lse};
 }
 
@@ -148,13 +155,19 @@ at::Tensor ce_bwd(const at::Tensor& grad, const at::Tensor& logits,
   auto stream = at::cuda::getCurrentCUDAStream();
   const int WPB = 4;
   dim3 block(WAVE * WPB), grid((N + WPB - 1) / WPB);
+  const bool per_row = gradf.numel() == N;
+  TORCH_CHECK(per_row || gradf.numel() == 1, "ce_bwd: grad scalar or [N]");
   if (logits.scalar_type() == at::kBFloat16) {
-    hipLaunchKernelGGL(ce_bwd_kernel<bf16>, grid, block, 0, stream,
+    auto kern = per_row ? ce_bwd_kernel<bf16, true>
+                        : ce_bwd_kernel<bf16, false>;
+    hipLaunchKernelGGL(kern, grid, block, 0, stream,
         gradf.data_ptr<float>(), (const bf16*)logits.data_ptr(),
         targets.data_ptr<long>(), lse.data_ptr<float>(),
         (bf16*)dlogits.data_ptr(), N, V);
   } else {
-    hipLaunchKernelGGL(ce_bwd_kernel<float>, grid, block, 0, stream,
+    auto kern = per_row ? ce_bwd_kernel<float, true>
+                        : ce_bwd_kernel<float, false>;
+    hipLaunchKernelGGL(kern, grid, block, 0, stream,
         gradf.data_ptr<float>(), logits.data_ptr<float>(),
         targets.data_ptr<long>(), lse.data_ptr<float>(),
         dlogits.data_ptr<float>(), N, V);

@@ -1,0 +1,87 @@
+"""Unmodified-model lowering: aten SDPA and log_softmax+nll chains are
+rewritten to the flash-attention and fused-CE kernels (VERDICT item 2;
+reference capability: unmodified PyTorch train step, README.md:14-36).
+
+Golden on CPU incl. ignore_index masking; the GPU kernel paths are the
+same custom ops covered by tests/test_gpu_kernels.py.
+"""
+import copy
+import os
+
+import pytest
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+def _init_pg():
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29552")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+
+
+class Toy(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.qkv = nn.Linear(64, 192)
+        self.out = nn.Linear(64, 64)
+        self.cls = nn.Linear(64, 128)
+
+    def forward(self, x):
+        B, T, C = x.shape
+        q, k, v = self.qkv(x).split(64, dim=2)
+        q = q.view(B, T, 4, 16).transpose(1, 2)
+        k = k.view(B, T, 4, 16).transpose(1, 2)
+        v = v.view(B, T, 4, 16).transpose(1, 2)
+        y = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        y = y.transpose(1, 2).reshape(B, T, C)
+        return self.cls(self.out(y))
+
+
+@pytest.mark.parametrize("with_ignore", [False, True])
+def test_sdpa_ce_lowering_golden(with_ignore):
+    _init_pg()
+    from easydist_amd import easydist_compile, easydist_setup, set_device_mesh
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh([0], ["spmd0"])
+
+    torch.manual_seed(0)
+    model = Toy()
+    ref = copy.deepcopy(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    opt_ref = torch.optim.SGD(ref.parameters(), lr=0.01)
+
+    def train_step(model, opt, x, y):
+        opt.zero_grad()
+        with torch.autocast("cpu", dtype=torch.bfloat16):
+            logits = model(x)
+            loss = F.cross_entropy(logits.view(-1, 128), y.view(-1))
+        loss.backward()
+        opt.step()
+        return loss
+
+    compiled = easydist_compile(train_step, parallel_mode="auto")
+    x = torch.randn(2, 32, 64)
+    y = torch.randint(0, 128, (2, 32))
+    if with_ignore:
+        y[0, :5] = -100
+    for _ in range(3):
+        loss = compiled(model, opt, x, y)
+        rl = train_step(ref, opt_ref, x, y)
+        assert abs(float(loss) - float(rl)) < 1e-3
+
+    gm = list(compiled.compiled.values())[0].gm
+    names = [getattr(n.target, "__name__", "") for n in gm.graph.nodes
+             if n.op == "call_function"]
+    assert names.count("flash_attention.default") == 1
+    assert names.count("flash_attention_bwd.default") == 1
+    assert names.count("ce_fwd_rows.default") == 1
+    assert names.count("ce_bwd.default") == 1
+    assert not any("scaled_dot" in s or "nll" in s or "log_softmax" in s
+                   for s in names)
+    for (n1, p1), (_, p2) in zip(model.named_parameters(),
+                                 ref.named_parameters()):
+        assert torch.allclose(p1, p2, rtol=1e-2, atol=2e-3), \
+            (n1, float((p1 - p2).abs().max()))
