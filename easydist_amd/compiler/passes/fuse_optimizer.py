@@ -221,6 +221,26 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
 
     graph = gm.graph
     out_node = next(n for n in graph.nodes if n.op == "output")
+
+    # grad-cast folding: the autocast trace casts every bf16 grad to fp32
+    # just for the optimizer (~1 activation-free but param-count-sized
+    # kernel per parameter).  The HIP kernel reads bf16 grads directly —
+    # unwrap the casts when EVERY matched grad unwraps uniformly.
+    def _uncast_bf16(g):
+        if isinstance(g, fx.Node) and g.op == "call_function"                 and g.target is aten._to_copy.default                 and g.kwargs.get("dtype") == torch.float32:
+            src = g.args[0]
+            v = src.meta.get("val") if hasattr(src, "meta") else None
+            if isinstance(v, torch.Tensor) and v.dtype == torch.bfloat16:
+                return src
+        return None
+
+    unwrapped = [_uncast_bf16(m[3]) for m in matched]
+    if all(u is not None for u in unwrapped):
+        matched = [(m[0], m[1], m[2], u, m[4], m[5], m[6])
+                   for m, u in zip(matched, unwrapped)]
+        logger.info("fuse_optimizer: feeding %d bf16 grads directly "
+                    "(folded fp32 casts)", len(matched))
+
     with graph.inserting_before(out_node):
         p_list = [m[2] for m in matched]
         if decay_scale is not None:

@@ -521,3 +521,126 @@ def lower_cross_entropy(gm: fx.GraphModule) -> int:
         logger.info("lower_hip: lowered %d cross-entropy chains to fused CE",
                     n_lowered)
     return n_lowered
+
+
+def _strip_view(n):
+    if isinstance(n, fx.Node) and n.op == "call_function" \
+            and n.target in (aten.view.default, aten._unsafe_view.default):
+        return n.args[0]
+    return None
+
+
+def lower_gelu_bwd_fuse(gm: fx.GraphModule) -> int:
+    """Fuse gelu_backward into the MFMA GEMM that produces its grad.
+
+    Pattern (c_proj dX of a transformer MLP):
+        g2 = gemm_nt(dY, Wt)            # the incoming gradient
+        g3 = view(g2, [B,T,N])
+        d  = gelu_backward(g3, x3)      # x3 = view(pre-act 2d)
+    ->  g2' = gemm_nt_act(dY, Wt, None, GELU_BWD, x2); d := view(g2')
+    The epilogue multiplies by dgelu(aux) during the (already coalesced)
+    C store — removes one full activation-sized read+write pass.
+    """
+    graph = gm.graph
+    nt_act = torch.ops.easydist_amd.gemm_nt_act.default
+    nt = torch.ops.easydist_amd.gemm_nt.default
+    n_fused = 0
+    for n in list(graph.nodes):
+        if n.op != "call_function" \
+                or n.target is not aten.gelu_backward.default:
+            continue
+        approx = n.kwargs.get("approximate",
+                              n.args[2] if len(n.args) > 2 else "none")
+        act = 2 if approx == "tanh" else 4
+        g, x = n.args[0], n.args[1]
+        # strip 3-D views down to the 2-D gemm / pre-act nodes
+        g2 = _strip_view(g) if _strip_view(g) is not None else g
+        x2 = _strip_view(x) if _strip_view(x) is not None else x
+        if not (isinstance(g2, fx.Node) and g2.op == "call_function"
+                and g2.target is nt):
+            continue
+        gv, xv = _val(g2), _val(x2)
+        if not (isinstance(xv, torch.Tensor) and xv.dim() == 2
+                and isinstance(gv, torch.Tensor)
+                and tuple(xv.shape) == tuple(gv.shape)
+                and xv.is_contiguous()):
+            continue
+        # the grad must not feed anything else (its value changes)
+        g_users = set(g2.users) | (set(g.users) if g is not g2 else set())
+        if g_users - {n, g}:
+            continue
+        a, bt, bias = g2.args
+        with graph.inserting_before(g2):
+            new = graph.call_function(nt_act, (a, bt, bias, act, x2))
+        new.meta = dict(g2.meta)
+        g2.replace_all_uses_with(new)
+        graph.erase_node(g2)
+        # gelu_backward output == the (now activated) gemm, viewed 3-D
+        if g is not g2:
+            n.replace_all_uses_with(g)
+        else:
+            n.replace_all_uses_with(new)
+        graph.erase_node(n)
+        n_fused += 1
+    if n_fused:
+        graph.eliminate_dead_code()
+        graph.lint()
+        gm.recompile()
+        logger.info("lower_hip: fused %d gelu_backward into GEMM epilogues",
+                    n_fused)
+    return n_fused
+
+
+def lower_bias_grad_fuse(gm: fx.GraphModule) -> int:
+    """Fuse the linear-bias gradient column sum into the dW TN GEMM.
+
+    Pattern: sum(dY2, [0], keepdim) alongside gemm_tn(dY2, X) — the TN
+    kernel already streams every element of dY through LDS; it
+    accumulates the column sum there instead of aten re-reading the
+    whole activation-sized gradient.
+    """
+    graph = gm.graph
+    tn = torch.ops.easydist_amd.gemm_tn.default
+    tn_asum = torch.ops.easydist_amd.gemm_tn_asum.default
+    n_fused = 0
+    for n in list(graph.nodes):
+        if n.op != "call_function" or n.target is not tn:
+            continue
+        dy, xb = n.args
+        sum_nodes = [u for u in dy.users
+                     if u.op == "call_function"
+                     and u.target is aten.sum.dim_IntList
+                     and list(u.args[1]) == [0]
+                     and u.kwargs.get("dtype") is None]
+        if not sum_nodes:
+            continue
+        with graph.inserting_before(n):
+            new = graph.call_function(tn_asum, (dy, xb))
+            dw = graph.call_function(operator.getitem, (new, 0))
+            asum = graph.call_function(operator.getitem, (new, 1))
+        dw.meta = dict(n.meta)
+        n.replace_all_uses_with(dw)
+        graph.erase_node(n)
+        for sn in sum_nodes:
+            keepdim = len(sn.args) > 2 and sn.args[2]
+            sv = _val(sn)
+            with graph.inserting_before(sn):
+                out = asum
+                if isinstance(sv, torch.Tensor) and sv.dtype != torch.float32:
+                    out = graph.call_function(aten._to_copy.default, (out,),
+                                              {"dtype": sv.dtype})
+                if keepdim:
+                    out = graph.call_function(
+                        aten.view.default,
+                        (out, [1, _val(dy).shape[1]]))
+            out.meta["val"] = sv
+            sn.replace_all_uses_with(out)
+            graph.erase_node(sn)
+        n_fused += 1
+    if n_fused:
+        graph.eliminate_dead_code()
+        graph.lint()
+        gm.recompile()
+        logger.info("lower_hip: fused %d bias-grad sums into TN GEMMs",
+                    n_fused)
+    return n_fused

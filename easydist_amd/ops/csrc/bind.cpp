@@ -34,6 +34,11 @@ fused_sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& bt,
                    const std::optional<at::Tensor>& bias);
 at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b);
+at::Tensor gemm_nt_act(const at::Tensor& a, const at::Tensor& bt,
+                       const std::optional<at::Tensor>& bias, int64_t act,
+                       const std::optional<at::Tensor>& aux);
+std::tuple<at::Tensor, at::Tensor> gemm_tn_asum(const at::Tensor& a,
+                                                const at::Tensor& b);
 std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
                                                   const at::Tensor& k,
                                                   const at::Tensor& v,
@@ -92,6 +97,9 @@ TORCH_LIBRARY(easydist_amd_hip, m) {
         "bool nesterov) -> (Tensor[], Tensor[])");
   m.def("gemm_nt(Tensor a, Tensor bt, Tensor? bias) -> Tensor");
   m.def("gemm_tn(Tensor a, Tensor b) -> Tensor");
+  m.def("gemm_nt_act(Tensor a, Tensor bt, Tensor? bias, int act, "
+        "Tensor? aux) -> Tensor");
+  m.def("gemm_tn_asum(Tensor a, Tensor b) -> (Tensor, Tensor)");
   m.def("flash_attn_fwd(Tensor q, Tensor k, Tensor v, bool causal) "
         "-> (Tensor, Tensor)");
   m.def("flash_attn_bwd(Tensor grad, Tensor q, Tensor k, Tensor v, "
@@ -110,6 +118,8 @@ TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
   m.impl("fused_sgd_step", sgd_wrap);
   m.impl("gemm_nt", gemm_nt);
   m.impl("gemm_tn", gemm_tn);
+  m.impl("gemm_nt_act", gemm_nt_act);
+  m.impl("gemm_tn_asum", gemm_tn_asum);
   m.impl("flash_attn_fwd", flash_attn_fwd);
   m.impl("flash_attn_bwd", flash_attn_bwd);
 }

@@ -52,6 +52,33 @@ DEVINL void glds16(const void* gsrc, void* lds_dst) {
 
 DEVINL void raw_barrier() { __builtin_amdgcn_s_barrier(); }
 
+// epilogue activation modes (fused into the C store pass):
+//   0 none | 1 gelu(tanh) | 2 gelu(tanh) backward: out = acc*dgelu(aux)
+//   3 gelu(erf) | 4 gelu(erf) backward
+#define GELU_C 0.7978845608028654f      // sqrt(2/pi)
+#define GELU_A 0.044715f
+DEVINL float act_apply(int act, float x, float aux) {
+  if (act == 1) {
+    float u = GELU_C * (x + GELU_A * x * x * x);
+    return 0.5f * x * (1.f + tanhf(u));
+  }
+  if (act == 2) {
+    float u = GELU_C * (aux + GELU_A * aux * aux * aux);
+    float t = tanhf(u);
+    float d = 0.5f * (1.f + t)
+              + 0.5f * aux * (1.f - t * t) * GELU_C
+                    * (1.f + 3.f * GELU_A * aux * aux);
+    return x * d;
+  }
+  if (act == 3) return 0.5f * x * (1.f + erff(x * 0.70710678118654752f));
+  if (act == 4) {
+    float cdf = 0.5f * (1.f + erff(aux * 0.70710678118654752f));
+    float pdf = 0.3989422804014327f * __expf(-0.5f * aux * aux);
+    return x * (cdf + aux * pdf);
+  }
+  return x;
+}
+
 // ===========================================================================
 // gemm_nt_128: 128x128 tile, BK=32, 4 waves, double-buffered glds staging.
 // Row-clamped staging + guarded epilogue: any M,N (N%1), K%32.
@@ -220,7 +247,7 @@ template <int SWZ>
 __global__ void __launch_bounds__(512)
 gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
             bf16* __restrict__ C, const bf16* __restrict__ bias,
-            int M, int N, int K) {
+            const bf16* __restrict__ aux, int act, int M, int N, int K) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
 
   unsigned nwg_m = (M + BM2 - 1) / BM2, nwg_n = (N + BN2 - 1) / BN2;
@@ -403,10 +430,23 @@ gemm_nt_256(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     int gn = n0 + wn + colc;
     if (gm < (int)M && gn + 8 <= (int)N) {
       bf16x8v val = *reinterpret_cast<const bf16x8v*>(&share[row * 64 + colc]);
+      if (act) {
+        bf16x8v av{};
+        if (act == 2 || act == 4)
+          av = *reinterpret_cast<const bf16x8v*>(&aux[(long)gm * N + gn]);
+        #pragma unroll
+        for (int e = 0; e < 8; ++e)
+          val[e] = (__bf16)act_apply(act, bf2f((bf16)val[e]),
+                                     bf2f((bf16)av[e]));
+      }
       *reinterpret_cast<bf16x8v*>(&C[(long)gm * N + gn]) = val;
     } else if (gm < (int)M) {
-      for (int e = 0; e < 8 && gn + e < (int)N; ++e)
-        C[(long)gm * N + gn + e] = share[row * 64 + colc + e];
+      for (int e = 0; e < 8 && gn + e < (int)N; ++e) {
+        float xv = bf2f(share[row * 64 + colc + e]);
+        float axv = (act == 2 || act == 4)
+                        ? bf2f(aux[(long)gm * N + gn + e]) : 0.f;
+        C[(long)gm * N + gn + e] = f2bf(act_apply(act, xv, axv));
+      }
     }
   }
 }
@@ -458,7 +498,8 @@ DEVINL bf16x8v tn_frag(const char* img, int chunk, int lane) {
 
 extern "C" __global__ void __launch_bounds__(256)
 gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
-               float* __restrict__ Cw, int R, int P, int Q, int splitr) {
+               float* __restrict__ Cw, float* __restrict__ Asum,
+               int R, int P, int Q, int splitr) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const unsigned IMG = TBR * TBP * 2;          // 8 KiB per operand image
   const unsigned BUF = 2 * IMG;                // A+B per buffer
@@ -489,6 +530,7 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
   };
 
   f32x4 acc[4][4];
+  float asum_acc[4] = {0.f, 0.f, 0.f, 0.f};
   #pragma unroll
   for (int i = 0; i < 4; ++i)
     #pragma unroll
@@ -509,6 +551,18 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
     #pragma unroll
     for (int i = 0; i < 4; ++i)
       a_frag[i] = tn_frag(smem + buf * BUF, (wp >> 4) + i, lane);
+    if (Asum && q0 == 0 && (wave & 1) == 0) {
+      // dBias fused into dW (only the q0==0 block column: A tiles repeat
+      // across q tiles): the A operand (dY) is already on chip —
+      // column-sum it here instead of a separate full re-read
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        float acc_s = 0.f;
+        #pragma unroll
+        for (int u = 0; u < 8; ++u) acc_s += bf2f((bf16)a_frag[i][u]);
+        asum_acc[i] += acc_s;
+      }
+    }
     #pragma unroll
     for (int j = 0; j < 4; ++j)
       b_frag[j] = tn_frag(smem + buf * BUF + IMG, (wq >> 4) + j, lane);
@@ -522,6 +576,19 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
     __syncthreads();
   }
 
+  if (Asum && q0 == 0 && (wave & 1) == 0) {
+    // fold the r-slice partials (lanes l and l+16/32/48 share a column)
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      float v = asum_acc[i];
+      v += __shfl_down(v, 32, WAVE);
+      v += __shfl_down(v, 16, WAVE);
+      if (lane < 16) {
+        int gp = p0 + wp + i * 16 + lane;
+        if (gp < P) atomicAdd(&Asum[gp], v);
+      }
+    }
+  }
   const int c_col = lane & 15;
   const int c_row0 = (lane >> 4) * 4;
   #pragma unroll
@@ -544,8 +611,9 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
 // ===========================================================================
 // host wrappers
 // ===========================================================================
-at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& bt,
-                   const std::optional<at::Tensor>& bias) {
+at::Tensor gemm_nt_act(const at::Tensor& a, const at::Tensor& bt,
+                       const std::optional<at::Tensor>& bias, int64_t act,
+                       const std::optional<at::Tensor>& aux) {
   TORCH_CHECK(a.dtype() == at::kBFloat16 && bt.dtype() == at::kBFloat16);
   TORCH_CHECK(a.is_contiguous() && bt.is_contiguous(),
               "gemm_nt wants K-contiguous operands");
@@ -560,22 +628,41 @@ at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& bt,
     unsigned nwg = ((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
     static int swz = []() {
       const char* e = getenv("EASYDIST_NT256_SWZ");
-      return e ? atoi(e) : 1;
+      return e ? atoi(e) : 2;
     }();
     auto kern = swz == 0 ? gemm_nt_256<0> : swz == 2 ? gemm_nt_256<2>
                                           : gemm_nt_256<1>;
+    const bf16* aux_p = aux ? (const bf16*)aux->data_ptr() : nullptr;
+    TORCH_CHECK(act == 0 || !(act == 2 || act == 4) || aux_p,
+                "gemm_nt: bwd activation needs aux");
     hipLaunchKernelGGL(kern, dim3(nwg), dim3(512), 2 * BUF_B, stream,
         (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
-        (bf16*)c.data_ptr(), bias_p, (int)M, (int)N, (int)K);
+        (bf16*)c.data_ptr(), bias_p, aux_p, (int)act, (int)M, (int)N,
+        (int)K);
   } else {
     unsigned nwg = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
     size_t lds = 2 * 2 * BM * BK * 2;   // 32 KiB
     hipLaunchKernelGGL(gemm_nt_128, dim3(nwg), dim3(256), lds, stream,
         (const bf16*)a.data_ptr(), (const bf16*)bt.data_ptr(),
         (bf16*)c.data_ptr(), bias_p, (int)M, (int)N, (int)K);
+    if (act) {
+      // edge-shape path: apply the activation as a separate aten pass
+      if (act == 1) c = at::gelu(c, "tanh");
+      else if (act == 3) c = at::gelu(c);
+      else if (act == 2) c = at::gelu_backward(c, aux.value(), "tanh");
+      else if (act == 4) c = at::gelu_backward(c, aux.value(), "none");
+    }
   }
   return c;
 }
+
+at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& bt,
+                   const std::optional<at::Tensor>& bias) {
+  return gemm_nt_act(a, bt, bias, 0, std::nullopt);
+}
+
+std::tuple<at::Tensor, at::Tensor> gemm_tn_asum(const at::Tensor& a,
+                                                const at::Tensor& b);
 
 at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
   // C[P,Q] = a^T @ b with a:[R,P], b:[R,Q]
@@ -600,6 +687,32 @@ at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
   hipLaunchKernelGGL(gemm_tn_kernel, dim3(ntile * splitr), dim3(256), lds,
       stream,
       (const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
-      cw.data_ptr<float>(), (int)R, (int)P, (int)Q, splitr);
+      cw.data_ptr<float>(), nullptr, (int)R, (int)P, (int)Q, splitr);
   return cw.to(at::kBFloat16);
+}
+
+std::tuple<at::Tensor, at::Tensor> gemm_tn_asum(const at::Tensor& a,
+                                                const at::Tensor& b) {
+  // like gemm_tn, additionally returning colsum(a) (dBias) in ONE pass
+  TORCH_CHECK(a.dtype() == at::kBFloat16 && b.dtype() == at::kBFloat16);
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+  const long R = a.size(0), P = a.size(1), Q = b.size(0) == R ? b.size(1) : -1;
+  TORCH_CHECK(Q > 0 && R % TBR == 0 && P % TBP == 0 && Q % TBP == 0);
+  unsigned ntile = (P / TBP) * (Q / TBP);
+  int splitr = 1;
+  while (ntile * splitr < 512 && splitr < 16 &&
+         (R / TBR) % (splitr * 2) == 0 && (R / TBR) / (splitr * 2) >= 1)
+    splitr *= 2;
+  auto cw = splitr == 1
+      ? at::empty({P, Q}, a.options().dtype(at::kFloat))
+      : at::zeros({P, Q}, a.options().dtype(at::kFloat));
+  auto asum = at::zeros({P}, a.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  size_t lds = 2 * 2 * TBR * TBP * 2;
+  hipLaunchKernelGGL(gemm_tn_kernel, dim3(ntile * splitr), dim3(256), lds,
+      stream,
+      (const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+      cw.data_ptr<float>(), asum.data_ptr<float>(), (int)R, (int)P, (int)Q,
+      splitr);
+  return {cw.to(at::kBFloat16), asum};
 }

@@ -14,7 +14,7 @@
 
 struct AdamChunk {
   const float* p;
-  const float* g;
+  const void* g;        // fp32 or bf16 (g_bf16 flag of the launch)
   const float* ea;
   const float* eas;
   float* out_p;
@@ -24,10 +24,22 @@ struct AdamChunk {
   int n;          // elements in this chunk
 };
 
+DEVINL void load_g4(const void* g, int i, int g_bf16, float* out) {
+  if (g_bf16) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(&((const bf16*)g)[i & ~7]);
+    const int o = i & 4;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) out[j] = bf2f(v.v[o + j]);
+  } else {
+    float4 v = *reinterpret_cast<const float4*>(&((const float*)g)[i]);
+    out[0] = v.x; out[1] = v.y; out[2] = v.z; out[3] = v.w;
+  }
+}
+
 #define CHUNK_ELEMS 65536
 
 __global__ void fused_adam_kernel(const AdamChunk* __restrict__ chunks,
-                                  int n_chunks,
+                                  int n_chunks, int g_bf16,
                                   const float* __restrict__ steps,  // [T]
                                   float lr, float beta1, float beta2,
                                   float weight_decay, float eps) {
@@ -41,10 +53,11 @@ __global__ void fused_adam_kernel(const AdamChunk* __restrict__ chunks,
   const float rsqrt_bc2 = rsqrtf(bc2);
   for (int i = threadIdx.x * 4; i + 3 < ch.n; i += blockDim.x * 4) {
     float4 p = *reinterpret_cast<const float4*>(&ch.p[i]);
-    float4 g = *reinterpret_cast<const float4*>(&ch.g[i]);
+    float gg[4];
+    load_g4(ch.g, i, g_bf16, gg);
     float4 ea = *reinterpret_cast<const float4*>(&ch.ea[i]);
     float4 eas = *reinterpret_cast<const float4*>(&ch.eas[i]);
-    float* pp = &p.x; float* gg = &g.x; float* ee = &ea.x; float* ss = &eas.x;
+    float* pp = &p.x; float* ee = &ea.x; float* ss = &eas.x;
     #pragma unroll
     for (int j = 0; j < 4; ++j) {
       float grad = gg[j];
@@ -63,7 +76,8 @@ __global__ void fused_adam_kernel(const AdamChunk* __restrict__ chunks,
   // scalar tail
   int tail_start = (ch.n / 4) * 4;
   for (int i = tail_start + threadIdx.x; i < ch.n; i += blockDim.x) {
-    float grad = ch.g[i];
+    float grad = g_bf16 ? bf2f(((const bf16*)ch.g)[i])
+                        : ((const float*)ch.g)[i];
     if (weight_decay != 0.f) grad += weight_decay * ch.p[i];
     float m = beta1 * ch.ea[i] + (1.f - beta1) * grad;
     float v = beta2 * ch.eas[i] + (1.f - beta2) * grad * grad;
@@ -91,8 +105,13 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
   const int T = params.size();
   std::vector<at::Tensor> out_p, out_ea, out_eas, out_steps;
   out_p.reserve(T); out_ea.reserve(T); out_eas.reserve(T);
+  const bool g_bf16 = !grads.empty()
+      && grads[0].scalar_type() == at::kBFloat16;
   for (int t = 0; t < T; ++t) {
     TORCH_CHECK(params[t].scalar_type() == at::kFloat, "fp32 params only");
+    TORCH_CHECK(grads[t].scalar_type()
+                    == (g_bf16 ? at::kBFloat16 : at::kFloat),
+                "grads must be uniformly fp32 or bf16");
     TORCH_CHECK(params[t].is_contiguous() && grads[t].is_contiguous());
     out_p.push_back(at::empty_like(params[t]));
     out_ea.push_back(at::empty_like(exp_avgs[t]));
@@ -113,7 +132,8 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
     for (long off = 0; off < n; off += CHUNK_ELEMS) {
       AdamChunk c;
       c.p = params[t].data_ptr<float>() + off;
-      c.g = grads[t].data_ptr<float>() + off;
+      c.g = g_bf16 ? (const void*)((const bf16*)grads[t].data_ptr() + off)
+                   : (const void*)(grads[t].data_ptr<float>() + off);
       c.ea = exp_avgs[t].data_ptr<float>() + off;
       c.eas = exp_avg_sqs[t].data_ptr<float>() + off;
       c.out_p = out_p[t].data_ptr<float>() + off;
@@ -165,8 +185,8 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
     C10_CUDA_CHECK(hipEventRecord(g_h2d_done, stream));
   hipLaunchKernelGGL(fused_adam_kernel, dim3(chunks.size()), dim3(256), 0,
       stream, (const AdamChunk*)g_dev.data_ptr(), (int)chunks.size(),
-      steps_flat.data_ptr<float>(), (float)lr, (float)beta1, (float)beta2,
-      (float)weight_decay, (float)eps);
+      (int)g_bf16, steps_flat.data_ptr<float>(), (float)lr, (float)beta1,
+      (float)beta2, (float)weight_decay, (float)eps);
   if (capturing) {
     // the recorded H2D re-reads THIS pinned buffer on every replay:
     // retire the pair so no later call can overwrite it
@@ -226,8 +246,13 @@ fused_sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
               "momentum needs one buffer per param");
   std::vector<at::Tensor> out_p, out_buf;
   out_p.reserve(T);
+  const bool g_bf16 = !grads.empty()
+      && grads[0].scalar_type() == at::kBFloat16;
   for (int t = 0; t < T; ++t) {
     TORCH_CHECK(params[t].scalar_type() == at::kFloat, "fp32 params only");
+    TORCH_CHECK(grads[t].scalar_type()
+                    == (g_bf16 ? at::kBFloat16 : at::kFloat),
+                "grads must be uniformly fp32 or bf16");
     TORCH_CHECK(params[t].is_contiguous() && grads[t].is_contiguous());
     out_p.push_back(at::empty_like(params[t]));
     if (has_m) out_buf.push_back(at::empty_like(bufs[t]));
@@ -238,7 +263,8 @@ fused_sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
     for (long off = 0; off < n; off += CHUNK_ELEMS) {
       SgdChunk c;
       c.p = params[t].data_ptr<float>() + off;
-      c.g = grads[t].data_ptr<float>() + off;
+      c.g = g_bf16 ? (const void*)((const bf16*)grads[t].data_ptr() + off)
+                   : (const void*)(grads[t].data_ptr<float>() + off);
       c.buf = has_m ? bufs[t].data_ptr<float>() + off : nullptr;
       c.out_p = out_p[t].data_ptr<float>() + off;
       c.out_buf = has_m ? out_buf[t].data_ptr<float>() + off : nullptr;

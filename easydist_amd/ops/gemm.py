@@ -161,3 +161,77 @@ def _nt_fake(a, bt, bias):
 @torch.library.register_fake("easydist_amd::gemm_tn")
 def _tn_fake(a, b):
     return a.new_empty((a.shape[1], b.shape[1]))
+
+
+# ---------------------------------------------------------- fused variants --
+lib.define("gemm_nt_act(Tensor a, Tensor bt, Tensor? bias, int act, "
+           "Tensor? aux) -> Tensor")
+lib.define("gemm_tn_asum(Tensor a, Tensor b) -> (Tensor, Tensor)")
+
+_GELU_TANH, _GELU_TANH_BWD, _GELU_ERF, _GELU_ERF_BWD = 1, 2, 3, 4
+
+
+def _apply_act(out, act, aux):
+    if act == _GELU_TANH:
+        return torch.nn.functional.gelu(out, approximate="tanh")
+    if act == _GELU_ERF:
+        return torch.nn.functional.gelu(out)
+    if act == _GELU_TANH_BWD:
+        return torch.ops.aten.gelu_backward(out, aux, approximate="tanh")
+    if act == _GELU_ERF_BWD:
+        return torch.ops.aten.gelu_backward(out, aux, approximate="none")
+    return out
+
+
+def _nt_act_cpu(a, bt, bias, act, aux):
+    return _apply_act(_nt_cpu(a, bt, bias), act, aux)
+
+
+def _nt_act_cuda(a, bt, bias, act, aux):
+    ext = load_extension()
+    if ext is None or not _nt_supported(a, bt) \
+            or not mdconfig.use_hip_kernels:
+        return _nt_act_cpu(a, bt, bias, act, aux)
+    a = a.contiguous()
+    bt = bt.contiguous()
+    b_c = bias.contiguous() if bias is not None else None
+    x_c = aux.contiguous() if aux is not None else None
+    key = ("nt", a.shape[0], a.shape[1], bt.shape[0], bias is not None)
+    if _choose(key, lambda: ext.gemm_nt(a, bt, b_c),
+               lambda: _nt_cpu(a, bt, bias)):
+        return ext.gemm_nt_act(a, bt, b_c, act, x_c)
+    return _nt_act_cpu(a, bt, bias, act, aux)
+
+
+def _tn_asum_cpu(a, b):
+    return a.t() @ b, a.float().sum(0)
+
+
+def _tn_asum_cuda(a, b):
+    ext = load_extension()
+    if ext is None or not _tn_supported(a, b) \
+            or not mdconfig.use_hip_kernels:
+        return _tn_asum_cpu(a, b)
+    a = a.contiguous()
+    b = b.contiguous()
+    key = ("tn", a.shape[0], a.shape[1], b.shape[1])
+    if _choose(key, lambda: ext.gemm_tn(a, b), lambda: _tn_cpu(a, b)):
+        return ext.gemm_tn_asum(a, b)
+    return _tn_asum_cpu(a, b)
+
+
+lib.impl("gemm_nt_act", _nt_act_cpu, "CPU")
+lib.impl("gemm_nt_act", _nt_act_cuda, "CUDA")
+lib.impl("gemm_tn_asum", _tn_asum_cpu, "CPU")
+lib.impl("gemm_tn_asum", _tn_asum_cuda, "CUDA")
+
+
+@torch.library.register_fake("easydist_amd::gemm_nt_act")
+def _nt_act_fake(a, bt, bias, act, aux):
+    return a.new_empty((a.shape[0], bt.shape[0]))
+
+
+@torch.library.register_fake("easydist_amd::gemm_tn_asum")
+def _tn_asum_fake(a, b):
+    return (a.new_empty((a.shape[1], b.shape[1])),
+            a.new_empty((a.shape[1],), dtype=torch.float32))

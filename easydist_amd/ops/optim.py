@@ -46,8 +46,10 @@ def _adam_cuda(params, grads, exp_avgs, exp_avg_sqs, steps, lr, beta1, beta2,
     ext = load_extension()
     # capture-safe: the kernel stages its chunk table through persistent
     # pinned+device buffers sized by the eager warmup call, so hipGraph
-    # capture records one async H2D + one kernel launch
-    if ext is not None and all(p.dtype == torch.float32 for p in params):
+    # capture records one async H2D + one kernel launch.  Grads may be
+    # uniformly bf16 (the fuse pass folds the autocast fp32 casts).
+    g_dtypes = {g.dtype for g in grads}
+    if ext is not None and all(p.dtype == torch.float32 for p in params)             and (g_dtypes <= {torch.float32} or g_dtypes <= {torch.bfloat16}):
         return ext.fused_adam_step(list(params), list(grads), list(exp_avgs),
                                    list(exp_avg_sqs), list(steps), lr, beta1,
                                    beta2, weight_decay, eps)
