@@ -209,7 +209,7 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 // other's capture-recorded buffers).
 struct SgdChunk {
   const float* p;
-  const float* g;
+  const void* g;      // fp32 or bf16 (g_bf16 flag of the launch)
   const float* buf;   // nullptr when momentum == 0
   float* out_p;
   float* out_buf;     // nullptr when momentum == 0
@@ -217,14 +217,15 @@ struct SgdChunk {
 };
 
 __global__ void fused_sgd_kernel(const SgdChunk* __restrict__ chunks,
-                                 int n_chunks, float lr, float momentum,
-                                 float dampening, float weight_decay,
-                                 int nesterov) {
+                                 int n_chunks, int g_bf16, float lr,
+                                 float momentum, float dampening,
+                                 float weight_decay, int nesterov) {
   const int c = blockIdx.x;
   if (c >= n_chunks) return;
   const SgdChunk ch = chunks[c];
   for (int i = threadIdx.x; i < ch.n; i += blockDim.x) {
-    float grad = ch.g[i];
+    float grad = g_bf16 ? bf2f(((const bf16*)ch.g)[i])
+                        : ((const float*)ch.g)[i];
     if (weight_decay != 0.f) grad += weight_decay * ch.p[i];
     float upd = grad;
     if (momentum != 0.f) {
@@ -302,8 +303,8 @@ fused_sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
     C10_CUDA_CHECK(hipEventRecord(g_h2d_done, stream));
   hipLaunchKernelGGL(fused_sgd_kernel, dim3(chunks.size()), dim3(256), 0,
       stream, (const SgdChunk*)g_dev.data_ptr(), (int)chunks.size(),
-      (float)lr, (float)momentum, (float)dampening, (float)weight_decay,
-      nesterov ? 1 : 0);
+      (int)g_bf16, (float)lr, (float)momentum, (float)dampening,
+      (float)weight_decay, nesterov ? 1 : 0);
   if (capturing) {
     static std::vector<at::Tensor> keepalive;
     keepalive.push_back(g_pinned);

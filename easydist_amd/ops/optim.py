@@ -92,7 +92,9 @@ def _sgd_aten(params, grads, bufs, lr, momentum, dampening, weight_decay,
 def _sgd_cuda(params, grads, bufs, lr, momentum, dampening, weight_decay,
               nesterov):
     ext = load_extension()
-    if ext is not None and all(p.dtype == torch.float32 for p in params):
+    g_dtypes = {g.dtype for g in grads}
+    if ext is not None and all(p.dtype == torch.float32 for p in params) \
+            and (g_dtypes <= {torch.float32} or g_dtypes <= {torch.bfloat16}):
         return ext.fused_sgd_step(list(params), list(grads), list(bufs),
                                   lr, momentum, dampening, weight_decay,
                                   nesterov)
