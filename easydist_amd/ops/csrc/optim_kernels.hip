@@ -169,13 +169,14 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
   }
   // the PREVIOUS call's async H2D must have consumed the pinned buffer
   // before the host overwrites it (race -> garbage chunk table -> NaNs).
-  // hipEventSynchronize on an event recorded OUTSIDE capture is legal
-  // during capture, so sync unconditionally: an in-flight eager H2D from
-  // the call immediately before capture must not read a torn table.
+  // hipEventSynchronize is REJECTED inside capture on ROCm
+  // (hipErrorStreamCaptureUnsupported) — the runtime device-synchronizes
+  // right before capture begins (compiled_func.run_graph), which retires
+  // any in-flight eager H2D, so the capture path may skip the wait.
   if (g_h2d_done == nullptr)
     C10_CUDA_CHECK(hipEventCreateWithFlags(&g_h2d_done,
                                            hipEventDisableTiming));
-  else
+  else if (!capturing)
     C10_CUDA_CHECK(hipEventSynchronize(g_h2d_done));
   memcpy(g_pinned.data_ptr(), chunks.data(), bytes);
   auto stream = at::cuda::getCurrentCUDAStream();

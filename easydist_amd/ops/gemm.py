@@ -74,10 +74,11 @@ def _tn_supported(a, b):
 _tune_cache = {}
 
 
-def _time_fn(fn, iters=3):
+def _time_fn(fn, iters=5):
     start = torch.cuda.Event(enable_timing=True)
     end = torch.cuda.Event(enable_timing=True)
-    fn()  # warm
+    fn(); fn()  # warm
+    torch.cuda.synchronize()
     start.record()
     for _ in range(iters):
         fn()

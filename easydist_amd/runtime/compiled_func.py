@@ -286,6 +286,10 @@ class EDCompiledFunc:
                 for pos, t in self.state.items():
                     t.copy_(snapshot[pos])
             del snapshot
+            # retire ALL in-flight eager work (e.g. the fused optimizer's
+            # async chunk-table H2D) — event syncs are illegal during
+            # capture, so the kernels rely on this barrier
+            torch.cuda.synchronize()
             self._graph = torch.cuda.CUDAGraph()
             self._graph_prepared = prepared
             with torch.cuda.graph(self._graph):
