@@ -205,7 +205,9 @@ def _nt_act_cuda(a, bt, bias, act, aux):
 
 
 def _tn_asum_cpu(a, b):
-    return a.t() @ b, a.float().sum(0)
+    # dtype=... keeps the reduction fused — a.float() would materialize a
+    # full fp32 copy of the activation-sized gradient
+    return a.t() @ b, a.sum(0, dtype=torch.float32)
 
 
 def _tn_asum_cuda(a, b):
