@@ -106,8 +106,9 @@ def main():
         if args.seq != cfg.block_size:
             from dataclasses import replace
             cfg = replace(cfg, block_size=args.seq)
-        ep_group = dist.group.WORLD if world_size > 1 else None
-        model = moem.MoEGPT(cfg, ep_group=ep_group).to(device)
+        # EP is chosen by the auto-SPMD solver (ops/moe_ops.py); the
+        # module-level ep_group path stays available for eager reference
+        model = moem.MoEGPT(cfg, ep_group=None).to(device)
         opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
 
         def train_step(model, opt, idx, targets):
@@ -139,21 +140,8 @@ def main():
                            generator=g).to(device)
         tokens_per_sample = args.seq
 
-    # MoE EP runs module-parallel (each rank computes its shard directly)
-    moe_ep = args.model.startswith("mixtral") and world_size > 1
-    if moe_ep:
-        class _Direct:
-            def __init__(self):
-                self.compiled = {}
-
-            def __call__(self, model, opt, i, t):
-                local_i = torch.chunk(i, world_size, 0)[rank]
-                local_t = torch.chunk(t, world_size, 0)[rank]
-                return train_step(model, opt, local_i, local_t)
-        compiled = _Direct()
-    else:
-        compiled = easydist_compile(train_step, parallel_mode=args.parallel,
-                                    cuda_graph=args.hipgraph)
+    compiled = easydist_compile(train_step, parallel_mode=args.parallel,
+                                cuda_graph=args.hipgraph)
 
     t_compile = time.time()
     for _ in range(max(args.warmup, 1)):

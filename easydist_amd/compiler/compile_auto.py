@@ -167,6 +167,7 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
                     for k, v in st.items()
                     if isinstance(v, torch.Tensor)])
     compiled.state_qualnames = {i: qn for i, qn in enumerate(qualnames)}
+    compiled.debug_pl_env = out_pl_env   # node name -> chosen placements
     compiled.meta = {
         "search_time": search_time, "solve_time": solve_time,
         "n_nodes": len(gm.graph.nodes), "out_spec": gm._out_spec

@@ -726,6 +726,52 @@ def _register_custom_ops():
         ann[1][0] = ShardDim.get_shard_dim(1)
         return ann, {1: [_reduce_add(), _gather(0)]}
 
+    # ---- MoE routing/combine (EP semantics -- see ops/moe_ops.py) ----
+    @register_preset(ed.moe_bins.default)
+    def _moe_bins_rule(input_shapes, args, kwargs):
+        # (tokens [N,C], topi [N,K], topv [N,K]) -> bins/gates/src/valid
+        # [E,cap,*]; token shard -> capacity-slice shard (parallel routing)
+        if len(input_shapes[0]) != 2 or input_shapes[0][0] <= 1:
+            return None, {}
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        for i in range(3):
+            ann[i][0] = ShardDim.get_shard_dim(1)
+        return ann, {1: [_gather(1), _gather(1), _gather(1), _gather(1)]}
+
+    @register_preset(ed.moe_bins_bwd.default)
+    def _moe_bins_bwd_rule(input_shapes, args, kwargs):
+        # (gbins [E,cap,C], ggates [E,cap], src, valid, tokens_ref [N,C])
+        # -> (gtokens [N,C], gtopv [N,K]); cap-slice shard -> token shard
+        if len(input_shapes[0]) != 3 or input_shapes[0][1] <= 1:
+            return None, {}
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        for i in range(4):
+            ann[i][1] = ShardDim.get_shard_dim(1)
+        ann[4][0] = ShardDim.get_shard_dim(1)
+        return ann, {1: [_gather(0), _gather(0)]}
+
+    @register_preset(ed.moe_combine.default)
+    def _moe_combine_rule(input_shapes, args, kwargs):
+        # (bins [E,cap,C], gates, src, valid, tokens_ref) -> out [N,C]
+        if len(input_shapes[0]) != 3 or input_shapes[0][1] <= 1:
+            return None, {}
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        for i in range(4):
+            ann[i][1] = ShardDim.get_shard_dim(1)
+        ann[4][0] = ShardDim.get_shard_dim(1)
+        return ann, {1: _gather(0)}
+
+    @register_preset(ed.moe_combine_bwd.default)
+    def _moe_combine_bwd_rule(input_shapes, args, kwargs):
+        # (gout [N,C], bins [E,cap,C], gates, src, valid) -> (gbins, ggates)
+        if len(input_shapes[1]) != 3 or input_shapes[1][1] <= 1:
+            return None, {}
+        ann = ShardAnnotation.init_from_input_shapes(input_shapes)
+        ann[0][0] = ShardDim.get_shard_dim(1)
+        for i in range(1, 5):
+            ann[i][1] = ShardDim.get_shard_dim(1)
+        return ann, {1: [_gather(1), _gather(1)]}
+
     @register_preset(ed.ce_bwd.default)
     def _ce_bwd_rule(input_shapes, args, kwargs):
         # (grad scalar, logits [N,C], targets [N], lse [N]) -> dlogits

@@ -77,7 +77,9 @@ beam_width = _env_int("EASYDIST_BEAM_WIDTH", 64)
 # limit (its timed-out incumbent is half-replicated and rank-divergent);
 # beam search finds the clean data/tensor-parallel assignment in seconds
 ilp_max_clusters = _env_int("EASYDIST_ILP_MAX_CLUSTERS", 1200)
-all_to_all_punish_factor = _env_float("EASYDIST_A2A_PUNISH", 1.5)
+# RCCL all_to_all_single on xGMI is simultaneous pairwise exchange; the
+# reference punished its all-gather+slice fallback 3x (sharding.py:155-163)
+all_to_all_punish_factor = _env_float("EASYDIST_A2A_PUNISH", 1.0)
 liveness_only_input = False
 
 # ----------------------------------------------------- MI355X cost model -----

@@ -51,8 +51,8 @@ MIXTRAL_BENCH_4L = MoEConfig(n_layer=4)
 
 def _capacity(tokens: int, n_experts: int, top_k: int,
               factor: float) -> int:
-    c = int(math.ceil(tokens * top_k / n_experts * factor))
-    return max(4, (c + 3) // 4 * 4)   # multiple of 4 for tidy kernels
+    from ..ops.moe_ops import capacity
+    return capacity(tokens, n_experts, top_k, factor)
 
 
 class ExpertFFN(nn.Module):
@@ -103,68 +103,37 @@ class MoELayer(nn.Module):
         topv, topi = probs.topk(K, dim=-1)             # [N, K]
         topv = (topv / topv.sum(-1, keepdim=True)).to(x.dtype)
 
-        # capacity per expert computed on LOCAL token count: payload shape
-        # [E, cap, C] is identical on every rank => equal-split all-to-all
-        cap = _capacity(N, E, K, self.cfg.capacity_factor)
-
-        # STATIC-SHAPE capacity routing: every (token, k) pair writes a
-        # slot; overflow pairs land in a per-expert TRASH slot (index
-        # `cap`) with gate 0. No data-dependent shapes -> whole-graph
-        # traceable and hipGraph-capturable.
-        flat_expert = topi.reshape(-1)                 # [N*K]
-        order = torch.argsort(flat_expert, stable=True)
-        sorted_e = flat_expert[order]
-        seg_start = torch.searchsorted(sorted_e, torch.arange(
-            E, device=x.device))
-        pos_in_seg = torch.arange(N * K, device=x.device) - \
-            seg_start[sorted_e]
-        keep = pos_in_seg < cap
-        slot = torch.where(keep, pos_in_seg,
-                           torch.full_like(pos_in_seg, cap))
-        tok_idx = order // K
-        k_idx = order % K
-        bins_x = tokens.new_zeros(E, cap + 1, C)
-        gates_x = topv.new_zeros(E, cap + 1)
-        src_x = torch.zeros(E, cap + 1, dtype=torch.long, device=x.device)
-        valid_x = torch.zeros(E, cap + 1, dtype=torch.bool,
-                              device=x.device)
-        bins_x[sorted_e, slot] = tokens[tok_idx]
-        gates_x[sorted_e, slot] = torch.where(
-            keep, topv[tok_idx, k_idx], torch.zeros_like(pos_in_seg,
-                                                         dtype=topv.dtype))
-        src_x[sorted_e, slot] = tok_idx
-        valid_x[sorted_e, slot] = keep
-        bins = bins_x[:, :cap].contiguous()
-        gates = gates_x[:, :cap]
-        src_index = src_x[:, :cap]
-        valid = valid_x[:, :cap]
+        # STATIC-SHAPE capacity routing through the compiler-visible
+        # custom ops (ops/moe_ops.py): the auto-SPMD solver sees their
+        # declared sharding algebra and can choose EP — bins resharded
+        # S(cap)->S(expert) with ONE rt_all_to_all over xGMI — against
+        # DP-experts, priced by its own cost model. capacity is computed
+        # inside the op from the LOCAL token count.
+        from ..ops import moe_ops  # noqa: F401  (registers the ops)
+        bins, gates, src_index, valid = torch.ops.easydist_amd.moe_bins(
+            tokens, topi, topv, E, self.cfg.capacity_factor)
+        cap = bins.shape[1]
 
         if self.ep_group is not None and self.ep_world > 1:
             from ..parallel import comm
             W, L = self.ep_world, self.n_local
-            # dispatch: chunk w of [E=W*L, cap, C] goes to expert-owner
-            # rank w — ONE equal-split all_to_all_single (pairwise xGMI)
+            # module-level EP (eager reference path; the compiled path
+            # reaches the same exchange through the solver)
             recv = comm.all_to_all_ep(bins, self.ep_group)
-            # recv[w*L+l] = rank w's token bin for my local expert l
             expert_in = recv.reshape(W, L, cap, C).transpose(0, 1) \
                 .reshape(L, W * cap, C)
             expert_out = self.experts(expert_in)
             send_back = expert_out.reshape(L, W, cap, C).transpose(0, 1) \
                 .reshape(W * L, cap, C).contiguous()
-            # combine: return every rank its tokens' expert outputs
             out_bins = comm.all_to_all_ep(send_back, self.ep_group)
             out_bins = out_bins.reshape(E, cap, C)
         else:
             out_bins = self.experts(bins).reshape(E, cap, C)
 
-        # combine: weighted scatter-add back to token positions (invalid
-        # slots add zeros to token 0 — static shapes, no boolean select)
-        out = tokens.new_zeros(N, C)
-        contrib = out_bins * gates.unsqueeze(-1) \
-            * valid.unsqueeze(-1).to(out_bins.dtype)
-        out.index_add_(0, src_index.reshape(-1),
-                       contrib.reshape(-1, C).to(out.dtype))
-        return out.reshape(B, T, C)
+        # combine: weighted scatter-add back to token positions
+        out = torch.ops.easydist_amd.moe_combine(
+            out_bins, gates, src_index, valid, tokens, K)
+        return out.to(x.dtype).reshape(B, T, C)
 
 
 class MoEBlock(nn.Module):

@@ -71,7 +71,7 @@ def require_hip_ops():
     return True
 
 
-from . import attention, norms, optim, ce, gemm  # noqa: E402  (register ops)
+from . import attention, norms, optim, ce, gemm, moe_ops  # noqa: E402
 
-__all__ = ["attention", "norms", "optim", "ce", "gemm", "load_extension",
-           "hip_ops_available", "require_hip_ops"]
+__all__ = ["attention", "norms", "optim", "ce", "gemm", "moe_ops",
+           "load_extension", "hip_ops_available", "require_hip_ops"]

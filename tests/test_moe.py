@@ -175,3 +175,104 @@ def test_moe_auto_ws2():
 @pytest.mark.world4
 def test_moe_auto_ws4():
     spawn(_moe_auto_body, args=(4,), world_size=4, port=29662)
+
+
+# ---------------------------------------------------------------------------
+# EP THROUGH THE COMPILER: the auto-SPMD solver chooses
+# the expert-parallel strategy for the opaque routing ops and the
+# sharding transform emits the S(cap)->S(expert) all-to-alls itself.
+# ---------------------------------------------------------------------------
+def _auto_ep_body(world_size):
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_compile, easydist_setup, set_device_mesh
+    from easydist_amd.models import moe as moem
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+    r = dist.get_rank()
+
+    torch.manual_seed(0)
+    cfg = moem.MoEConfig(vocab_size=64, n_layer=1, n_head=2, n_embd=64,
+                         block_size=32, n_experts=4, top_k=2, ffn_hidden=256,
+                         capacity_factor=8.0)   # no drops: exact comparison
+    model = moem.MoEGPT(cfg, ep_group=None)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    ref = copy.deepcopy(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, foreach=False)
+    opt_ref = torch.optim.Adam(ref.parameters(), lr=1e-3, foreach=False)
+
+    def train_step(model, opt, idx, targets):
+        return moem.moe_train_step(model, opt, idx, targets)
+
+    compiled = easydist_compile(train_step, parallel_mode="auto")
+
+    g = torch.Generator().manual_seed(7)
+    B = 2 * world_size
+    idx = torch.randint(0, cfg.vocab_size, (B, cfg.block_size), generator=g)
+    tg = torch.randint(0, cfg.vocab_size, (B, cfg.block_size), generator=g)
+
+    for step in range(2):
+        loss = compiled(model, opt, idx, tg)
+        ref_loss = train_step(ref, opt_ref, idx, tg)
+        assert abs(float(loss) - float(ref_loss)) < 5e-3, \
+            (step, float(loss), float(ref_loss))
+
+    gm = list(compiled.compiled.values())[0].gm
+    names = [getattr(n.target, "__name__", "") for n in gm.graph.nodes
+             if n.op == "call_function"]
+    # the routing runs through the compiler-visible ops; at this (small)
+    # scale the solver may legitimately prefer replicated routing — the
+    # EP all-to-all choice itself is asserted at realistic scale in
+    # test_moe_auto_ep_solver_ws2
+    assert names.count("moe_bins.default") == 1
+    assert names.count("moe_combine.default") == 1
+
+
+@pytest.mark.world2
+def test_moe_auto_ep_ws2():
+    spawn(_auto_ep_body, args=(2,), world_size=2, port=29563)
+
+
+def _auto_ep_solver_body(world_size):
+    """Solve-only (no execution): at Mixtral-like scale the solver must
+    choose expert parallelism — rt_all_to_all reshards around the expert
+    FFN — because replicating the expert bank (or all-reducing its
+    gradients) is costlier than exchanging the token bins over xGMI."""
+    import torch.distributed as dist
+
+    from easydist_amd import easydist_setup, set_device_mesh
+    from easydist_amd.compiler.compile_auto import shard_graph
+    from easydist_amd.compiler.passes.functionalize import canonicalize
+    from easydist_amd.compiler.tracing import ed_compile_func
+    from easydist_amd.models import moe as moem
+    from easydist_amd.parallel.device_mesh import get_device_mesh
+
+    easydist_setup(backend="torch", device="cpu")
+    set_device_mesh(list(range(world_size)), ["spmd0"])
+    torch.manual_seed(0)
+    cfg = moem.MoEConfig(vocab_size=4096, n_layer=1, n_head=8, n_embd=1024,
+                         block_size=512, n_experts=8, top_k=2,
+                         ffn_hidden=4096, capacity_factor=1.25)
+    model = moem.MoEGPT(cfg, ep_group=None)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, foreach=False)
+
+    def train_step(model, opt, idx, targets):
+        return moem.moe_train_step(model, opt, idx, targets)
+
+    idx = torch.randint(0, cfg.vocab_size, (8, 512))
+    tg = torch.randint(0, cfg.vocab_size, (8, 512))
+    params, buffers, named_states, gm = ed_compile_func(
+        train_step, "fake", (model, opt, idx, tg), {}, model, opt)
+    gm, io_map = canonicalize(gm)
+    gm2, env, _, _ = shard_graph(gm, get_device_mesh(), io_map, set(), "cpu")
+    names = [getattr(n.target, "__name__", "") for n in gm2.graph.nodes
+             if n.op == "call_function"]
+    n_a2a = sum(1 for s in names if s == "rt_all_to_all_start")
+    assert n_a2a >= 2, (n_a2a, [s for s in names if s.startswith("rt_")])
+
+
+@pytest.mark.world2
+def test_moe_auto_ep_solver_ws2():
+    spawn(_auto_ep_solver_body, args=(2,), world_size=2, port=29564)
