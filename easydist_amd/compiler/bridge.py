@@ -39,6 +39,21 @@ def _flops_of(node: fx.Node) -> float:
         v = a.meta.get("val") if isinstance(a, fx.Node) else None
         return tuple(v.shape) if isinstance(v, torch.Tensor) else None
     try:
+        ed = torch.ops.easydist_amd
+        if t in (ed.flash_attention.default,):
+            q = shape(0)
+            # QK^T + PV, causal halves the work
+            return 2.0 * q[0] * q[1] * q[2] * q[2] * q[3] * 2 * 0.5
+        if t in (ed.flash_attention_bwd.default,):
+            q = shape(1)
+            return 2.0 * q[0] * q[1] * q[2] * q[2] * q[3] * 5 * 0.5
+        if t in (aten._scaled_dot_product_flash_attention.default
+                 if hasattr(aten, "_scaled_dot_product_flash_attention")
+                 else aten.mm.default,
+                 aten._scaled_dot_product_flash_attention_for_cpu.default,):
+            q = shape(0)
+            if q and len(q) == 4:
+                return 2.0 * q[0] * q[1] * q[2] * q[2] * q[3] * 2 * 0.5
         if t in (aten.mm.default,):
             a, b = shape(0), shape(1)
             return 2.0 * a[0] * a[1] * b[1]

@@ -212,6 +212,33 @@ class ShardingTransform:
                 shape_arg = list(node.args[1])
                 node.update_arg(1, self._local_shape(shape_arg, out_pl))
 
+        # sequence-parallel rewrite: flash attention assigned S(2) inputs
+        # runs as ring attention over the xGMI neighbors of that mesh dim
+        # (SURVEY §5 long-context requirement: SP is a SOLVER strategy,
+        # not a model flag)
+        try:
+            import torch as _torch
+            _fa = _torch.ops.easydist_amd.flash_attention.default
+            _fab = _torch.ops.easydist_amd.flash_attention_bwd.default
+        except Exception:
+            _fa = _fab = None
+        if node.target in (_fa, _fab) and _fa is not None:
+            q_pl = self._required_in(node, 0)
+            sp_dims = [d for d, p in enumerate(q_pl)
+                       if p.is_shard() and p.dim == 2
+                       and self.mesh_shape[d] > 1]
+            if sp_dims:
+                from ...runtime import comm_runtime as _crt
+                md = sp_dims[0]
+                if node.target is _fa:
+                    q_, k_, v_, causal_ = node.args
+                    node.target = _crt.rt_ring_attention
+                    node.args = (q_, k_, v_, causal_, md)
+                else:
+                    g_, q_, k_, v_, o_, l_, causal_ = node.args
+                    node.target = _crt.rt_ring_attention_bwd
+                    node.args = (g_, q_, k_, v_, o_, l_, causal_, md)
+
     def _fix_outputs(self, graph: fx.Graph, out_node: fx.Node):
         """User-visible returns were constrained to REPLICATE by the solver;
         anything that still isn't (solver fallback) is resharded here so the

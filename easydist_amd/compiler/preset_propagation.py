@@ -695,6 +695,13 @@ def _register_custom_ops():
                 ann[i][d] = ShardDim.get_shard_dim(sid)
             combs[sid] = [_gather(d), _gather(d)]
             sid += 1
+        # sequence-parallel strategy: S(2) on q/k/v is realized by the
+        # ring-attention runtime (the sharding transform rewrites the op
+        # to rt_ring_attention) — long-context choice for the solver
+        if input_shapes[0][2] > 1:
+            for i in range(3):
+                ann[i][2] = ShardDim.get_shard_dim(sid)
+            combs[sid] = [_gather(2), _gather(2)]
         return ann, combs
 
     @register_preset(ed.flash_attention_bwd.default)
@@ -712,6 +719,11 @@ def _register_custom_ops():
                 ann[i][d] = ShardDim.get_shard_dim(sid)
             combs[sid] = [_gather(d)] * 3
             sid += 1
+        # sequence-parallel: rewritten to rt_ring_attention_bwd
+        if input_shapes[1][2] > 1:
+            for i in range(6):
+                ann[i][2] = ShardDim.get_shard_dim(sid)
+            combs[sid] = [_gather(2)] * 3
         return ann, combs
 
     @register_preset(ed.ce_fwd.default)

@@ -98,6 +98,35 @@ def _ring_forward(q, k, v, group, causal):
     return out.to(q.dtype), lse
 
 
+def ring_bwd(grad, q, k, v, out, lse, group, causal):
+    """Two-ring exact backward; the per-block math is exactly
+    flash_attention_bwd evaluated against the GLOBAL out/lse."""
+    w = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+    dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+    kb, vb = k, v                       # rotating K/V, source dtype
+    dkv = torch.zeros((2,) + k.shape, dtype=torch.float32,
+                      device=k.device)  # rotating dK/dV, fp32
+    grad = grad.contiguous()
+    with torch.no_grad():
+        for step in range(w):
+            if step > 0:
+                kb, vb = _exchange_pair(kb, vb, group)
+                dkv = _ring_exchange(dkv, group)
+            src = (rank - step) % w
+            mode = _mode(src, rank, causal)
+            if mode != "skip":
+                dq_b, dk_b, dv_b = torch.ops.easydist_amd.\
+                    flash_attention_bwd(grad, q, kb, vb, out, lse,
+                                        mode == "causal")
+                dq += dq_b.float()
+                dkv[0] += dk_b.float()
+                dkv[1] += dv_b.float()
+        # one final hop returns each block (with its grads) home
+        dkv = _ring_exchange(dkv, group)
+    return dq.to(q.dtype), dkv[0].to(k.dtype), dkv[1].to(v.dtype)
+
+
 class _RingAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, group, causal):
@@ -112,36 +141,9 @@ class _RingAttention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad):
         q, k, v, out, lse = ctx.saved_tensors
-        group, causal = ctx.group, ctx.causal
-        w = dist.get_world_size(group)
-        rank = dist.get_rank(group)
-        # flash_attention_bwd's math (P from the GLOBAL lse, delta from
-        # the GLOBAL out) is exactly one ring block's contribution, so
-        # the per-block backward dispatches to the same op family as the
-        # forward (HIP kernel / batched hipBLASLt on GPU).
-        dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
-        kb, vb = k, v                       # rotating K/V, source dtype
-        dkv = torch.zeros((2,) + k.shape, dtype=torch.float32,
-                          device=k.device)  # rotating dK/dV, fp32
-        grad = grad.contiguous()
-        with torch.no_grad():
-            for step in range(w):
-                if step > 0:
-                    kb, vb = _exchange_pair(kb, vb, group)
-                    dkv = _ring_exchange(dkv, group)
-                src = (rank - step) % w
-                mode = _mode(src, rank, causal)
-                if mode != "skip":
-                    dq_b, dk_b, dv_b = torch.ops.easydist_amd.\
-                        flash_attention_bwd(grad, q, kb, vb, out, lse,
-                                            mode == "causal")
-                    dq += dq_b.float()
-                    dkv[0] += dk_b.float()
-                    dkv[1] += dv_b.float()
-            # one final hop returns each block (with its grads) home
-            dkv = _ring_exchange(dkv, group)
-        return (dq.to(q.dtype), dkv[0].to(k.dtype), dkv[1].to(v.dtype),
-                None, None)
+        dq, dk, dv = ring_bwd(grad, q, k, v, out, lse, ctx.group,
+                              ctx.causal)
+        return dq, dk, dv, None, None
 
 
 def ring_attention(q, k, v, group=None, causal: bool = True):

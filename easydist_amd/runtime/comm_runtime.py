@@ -52,3 +52,27 @@ def rt_partial_localize(t, mesh_dim: int):
 COMM_START_TARGETS = {rt_all_reduce_start, rt_all_gather_start,
                       rt_reduce_scatter_start, rt_all_to_all_start}
 COMM_LOCAL_TARGETS = {rt_local_chunk, rt_partial_localize}
+
+
+def rt_ring_attention(q, k, v, causal: bool, mesh_dim: int):
+    """Sequence-parallel exact attention over the xGMI ring (SP chosen by
+    the solver: flash_attention with S(seq) inputs is rewritten to this —
+    passes/sharding.py). Returns (out_local, lse_local)."""
+    from ..ops import ring_attention as ra
+    g = _group(mesh_dim)
+    import torch.distributed as dist
+    if g is None or dist.get_world_size(g) == 1:
+        return torch.ops.easydist_amd.flash_attention(q, k, v, causal)
+    out, lse = ra._ring_forward(q, k, v, g, causal)
+    return out, lse
+
+
+def rt_ring_attention_bwd(grad, q, k, v, out, lse, causal: bool,
+                          mesh_dim: int):
+    from ..ops import ring_attention as ra
+    g = _group(mesh_dim)
+    import torch.distributed as dist
+    if g is None or dist.get_world_size(g) == 1:
+        return torch.ops.easydist_amd.flash_attention_bwd(
+            grad, q, k, v, out, lse, causal)
+    return ra.ring_bwd(grad, q, k, v, out, lse, g, causal)
