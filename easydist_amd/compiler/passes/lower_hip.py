@@ -630,9 +630,10 @@ def lower_bias_grad_fuse(gm: fx.GraphModule) -> int:
                     out = graph.call_function(aten._to_copy.default, (out,),
                                               {"dtype": sv.dtype})
                 if keepdim:
-                    out = graph.call_function(
-                        aten.view.default,
-                        (out, [1, _val(dy).shape[1]]))
+                    # -1: the traced global length is wrong once the
+                    # transform shards the bias grad
+                    out = graph.call_function(aten.view.default,
+                                              (out, [1, -1]))
             out.meta["val"] = sv
             sn.replace_all_uses_with(out)
             graph.erase_node(sn)

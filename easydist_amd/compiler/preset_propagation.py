@@ -20,6 +20,12 @@ aten = torch.ops.aten
 # op overload -> fn(input_shapes, args, kwargs) -> (ShardAnnotation, {id: comb})
 _PRESET_REGISTRY: Dict[object, Callable] = {}
 
+# Rules whose sharded semantics are realized by a RUNTIME exchange (ring
+# attention, MoE EP) rather than by independent local execution: the
+# machine-check in tests/test_presets.py skips these (op -> "all" or a
+# set of input dims whose shard groups are runtime-realized).
+RUNTIME_REALIZED: Dict[object, object] = {}
+
 
 def register_preset(*ops):
     def deco(fn):
@@ -698,6 +704,7 @@ def _register_custom_ops():
         # sequence-parallel strategy: S(2) on q/k/v is realized by the
         # ring-attention runtime (the sharding transform rewrites the op
         # to rt_ring_attention) — long-context choice for the solver
+        RUNTIME_REALIZED[ed.flash_attention.default] = {2}
         if input_shapes[0][2] > 1:
             for i in range(3):
                 ann[i][2] = ShardDim.get_shard_dim(sid)
@@ -720,6 +727,7 @@ def _register_custom_ops():
             combs[sid] = [_gather(d)] * 3
             sid += 1
         # sequence-parallel: rewritten to rt_ring_attention_bwd
+        RUNTIME_REALIZED[ed.flash_attention_bwd.default] = {2}
         if input_shapes[1][2] > 1:
             for i in range(6):
                 ann[i][2] = ShardDim.get_shard_dim(sid)
@@ -739,6 +747,9 @@ def _register_custom_ops():
         return ann, {1: [_reduce_add(), _gather(0)]}
 
     # ---- MoE routing/combine (EP semantics -- see ops/moe_ops.py) ----
+    for _op in (ed.moe_bins.default, ed.moe_bins_bwd.default,
+                ed.moe_combine.default, ed.moe_combine_bwd.default):
+        RUNTIME_REALIZED[_op] = "all"
     @register_preset(ed.moe_bins.default)
     def _moe_bins_rule(input_shapes, args, kwargs):
         # (tokens [N,C], topi [N,K], topv [N,K]) -> bins/gates/src/valid
