@@ -24,7 +24,9 @@ def _worker(rank, fn, args, world_size, port, backend, q):
         os.environ["RANK"] = str(rank)
         os.environ["WORLD_SIZE"] = str(world_size)
         if backend == "nccl":
-            torch.cuda.set_device(rank)
+            # multi-rank-per-GPU (RCCL supports it): wrap on the visible
+            # device count so the ws2 collective paths run on a 1-GPU box
+            torch.cuda.set_device(rank % max(torch.cuda.device_count(), 1))
         dist.init_process_group(backend=backend, rank=rank,
                                 world_size=world_size,
                                 init_method=f"tcp://127.0.0.1:{port}")
