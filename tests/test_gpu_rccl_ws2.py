@@ -1,8 +1,10 @@
-"""RCCL world-2 on ONE MI355X: the collective code paths
-that CPU tests only exercise under gloo emulation — all_gather,
-reduce_scatter, all_to_all, and the full auto-SPMD golden — run here over
-real RCCL with two ranks sharing the device (RCCL supports
-multi-rank-per-GPU)."""
+"""RCCL world-2 collective paths over real RCCL.
+
+VERDICT item 6 suggested two ranks sharing ONE MI355X; RCCL 2.26 rejects
+that outright ("Duplicate GPU detected: rank 0 and rank 1 both on CUDA
+device" — ncclInvalidUsage), so these tests require >= 2 visible GPUs
+and run whenever the driver lands on a multi-GPU box; the gloo goldens
+cover the same graph-level paths every round on CPU."""
 import copy
 
 import pytest
@@ -12,8 +14,10 @@ from easydist_amd.utils.testing import spawn
 
 pytestmark = pytest.mark.gpu
 
-requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
-                                  reason="needs MI355X")
+requires_gpu = pytest.mark.skipif(
+    not torch.cuda.is_available() or torch.cuda.device_count() < 2,
+    reason="RCCL refuses 2 ranks on one device (Duplicate GPU detected); "
+           "needs >= 2 MI355X")
 
 
 def _collectives_body(world_size):
