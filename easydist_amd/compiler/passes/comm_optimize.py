@@ -13,6 +13,7 @@ odd-even heuristic (schedule/rcpsp.py, reference schedule/rcpsp.py:
 from __future__ import annotations
 
 import logging
+import operator
 from typing import Dict, List, Optional
 
 import torch.fx as fx
@@ -127,6 +128,7 @@ def comm_optimize(gm: fx.GraphModule, durations: Optional[Dict[str, float]]
     reorder independent compute between start/wait pairs with the RCPSP
     odd-even heuristic."""
     sink_waits_raise_starts(gm)
+    group_collectives(gm)
     if durations:
         from ...schedule.rcpsp import odd_even_schedule
         order = odd_even_schedule(gm, durations)
@@ -147,3 +149,100 @@ def _relink(gm: fx.GraphModule, order: List[fx.Node]):
         anchor = n
     graph.lint()
     gm.recompile()
+
+
+def group_collectives(gm: fx.GraphModule, min_group: int = 2,
+                      max_bucket_bytes: int = 64 << 20) -> int:
+    """Fuse compatible small collectives into ONE flat bucketed call
+    (reference comm_optimize.py:356-390 grouped_comm).
+
+    Run construction: walk candidate (start, wait) pairs in graph order,
+    accumulating while (a) no already-collected wait has a consumer that
+    precedes the next start (fusing would need that result before the
+    bucket executes — a cycle), and (b) the bucket stays under the size
+    cap. The grouped start+wait land at the LAST member start's position
+    (every member's input precedes its own start, hence the bucket).
+    all_gather members carry per-tensor gather dims (ZeRO re-gathers
+    updated param shards along the dim the solver picked).
+    """
+    from ...runtime.comm_runtime import (rt_all_gather_start,
+                                         rt_all_reduce_start,
+                                         rt_grouped_all_gather_start,
+                                         rt_grouped_all_reduce_start,
+                                         rt_grouped_wait)
+    graph = gm.graph
+    order = {n: i for i, n in enumerate(graph.nodes)}
+
+    def nbytes(n):
+        v = n.meta.get("val") if hasattr(n, "meta") else None
+        import torch as _t
+        return (v.numel() * v.element_size()
+                if isinstance(v, _t.Tensor) else 1 << 30)
+
+    groups = {}
+    for n in graph.nodes:
+        if n.op != "call_function":
+            continue
+        if n.target is rt_all_reduce_start:
+            key = ("ar", n.args[1], n.args[2])
+        elif n.target is rt_all_gather_start:
+            key = ("ag", n.args[2])
+        else:
+            continue
+        waits = [u for u in n.users if _is_wait(u)]
+        if len(waits) != 1 or len(n.users) != 1:
+            continue
+        groups.setdefault(key, []).append((n, waits[0]))
+
+    n_grouped = 0
+    INF = float("inf")
+    for key, pairs in groups.items():
+        if len(pairs) < min_group:
+            continue
+        pairs.sort(key=lambda p: order[p[0]])
+        runs = []
+        cur, cur_bytes, barrier = [], 0, INF
+        for s, w in pairs:
+            b = nbytes(s.args[0])
+            if cur and (order[s] > barrier
+                        or cur_bytes + b > max_bucket_bytes):
+                runs.append(cur)
+                cur, cur_bytes, barrier = [], 0, INF
+            cur.append((s, w))
+            cur_bytes += b
+            u_min = min((order[u] for u in w.users), default=INF)
+            barrier = min(barrier, u_min)
+        if cur:
+            runs.append(cur)
+        for run in runs:
+            if len(run) < min_group:
+                continue
+            starts = [s for s, _ in run]
+            waits = [w for _, w in run]
+            last_s = starts[-1]
+            with graph.inserting_before(last_s):
+                if key[0] == "ar":
+                    gs = graph.call_function(
+                        rt_grouped_all_reduce_start,
+                        ([s.args[0] for s in starts], key[1], key[2]))
+                else:
+                    gs = graph.call_function(
+                        rt_grouped_all_gather_start,
+                        ([s.args[0] for s in starts],
+                         [s.args[1] for s in starts], key[1]))
+                gw = graph.call_function(rt_grouped_wait, (gs,))
+                items = [graph.call_function(operator.getitem, (gw, i))
+                         for i in range(len(run))]
+            for (s, w), item in zip(run, items):
+                item.meta = dict(w.meta)
+                w.replace_all_uses_with(item)
+                graph.erase_node(w)
+                graph.erase_node(s)
+            n_grouped += len(run)
+
+    if n_grouped:
+        graph.lint()
+        gm.recompile()
+        logger.info("group_collectives: fused %d collectives into buckets",
+                    n_grouped)
+    return n_grouped

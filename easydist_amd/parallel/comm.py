@@ -280,3 +280,51 @@ def batch_p2p(p2p_ops: List[dist.P2POp]):
     if not p2p_ops:
         return []
     return dist.batch_isend_irecv(p2p_ops)
+
+
+# ------------------------------------------------------ grouped (bucketed) ---
+def all_reduce_bucket_start(tensors, op: str = "sum", group=None):
+    """ONE flat all-reduce for many small tensors (reference
+    comm_optimize.py:356-390 grouped_comm, re-done as a single fused
+    wire buffer: per-call RCCL latency is ~10us, so a ZeRO/ddp tail of
+    per-param collectives is latency-bound)."""
+    shapes = [tuple(t.shape) for t in tensors]
+    numels = [t.numel() for t in tensors]
+    flat = torch.cat([t.reshape(-1) for t in tensors])
+    w = all_reduce_start(flat, op, group)
+
+    def post(res):
+        outs = []
+        off = 0
+        for sh, n in zip(shapes, numels):
+            outs.append(res[off:off + n].reshape(sh))
+            off += n
+        return outs
+    base_post = w.post
+    w.post = (lambda x: post(base_post(x))) if base_post else post
+    return w
+
+
+def all_gather_bucket_start(tensors, gather_dims=None, group=None):
+    """ONE flat all-gather for many tensors (per-tensor gather dim).
+    Wire layout: concat of the local shards; each output tensor is
+    reassembled from the W rank segments along its own dim."""
+    n = _world(group)
+    gather_dims = gather_dims or [0] * len(tensors)
+    shapes = [tuple(t.shape) for t in tensors]
+    numels = [t.numel() for t in tensors]
+    total = sum(numels)
+    flat = torch.cat([t.contiguous().reshape(-1) for t in tensors])
+    out = torch.empty(n * total, dtype=flat.dtype, device=flat.device)
+    work = dist.all_gather_into_tensor(out, flat, group=group, async_op=True)
+
+    def post(res):
+        outs = []
+        off = 0
+        for sh, ne, d in zip(shapes, numels, gather_dims):
+            parts = [res[r * total + off: r * total + off + ne].reshape(sh)
+                     for r in range(n)]
+            outs.append(torch.cat(parts, dim=d))
+            off += ne
+        return outs
+    return _Work(out, work, post)

@@ -76,3 +76,20 @@ def rt_ring_attention_bwd(grad, q, k, v, out, lse, causal: bool,
         return torch.ops.easydist_amd.flash_attention_bwd(
             grad, q, k, v, out, lse, causal)
     return ra.ring_bwd(grad, q, k, v, out, lse, g, causal)
+
+
+def rt_grouped_all_reduce_start(tensors, op: str, mesh_dim: int):
+    return comm.all_reduce_bucket_start(list(tensors), op, _group(mesh_dim))
+
+
+def rt_grouped_all_gather_start(tensors, gather_dims, mesh_dim: int):
+    return comm.all_gather_bucket_start(list(tensors), list(gather_dims),
+                                        _group(mesh_dim))
+
+
+def rt_grouped_wait(w):
+    return comm.comm_wait(w)
+
+
+COMM_START_TARGETS.add(rt_grouped_all_reduce_start)
+COMM_START_TARGETS.add(rt_grouped_all_gather_start)
