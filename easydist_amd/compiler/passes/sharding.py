@@ -122,6 +122,33 @@ class ShardingTransform:
             return self._reshard_cache[key]
         node = arg
         cur = list(cur)
+        # P2P planner (reference sharding.py:336-612 rectangle planner):
+        # when MULTIPLE mesh dims change and everything is SHARD/REPLICATE,
+        # one batched xGMI exchange of rectangle intersections replaces
+        # the per-dim chain (which may all_to_all + all_gather the full
+        # tensor several times)
+        from ... import config as _cfg
+        changed = [d for d in range(self.ndim)
+                   if self.mesh_shape[d] > 1 and not _same(cur[d], want[d])]
+        if getattr(_cfg, "reshard_planner", "auto") in ("auto", "p2p")                 and len(changed) >= 2:
+            def ser(pl):
+                out = []
+                for p in pl:
+                    if p.is_shard():
+                        out.append(("S", p.dim))
+                    elif p.is_replicate():
+                        out.append(("R",))
+                    else:
+                        return None
+                return out
+            cs, ws = ser(cur), ser(want)
+            val = arg.meta.get("val") if hasattr(arg, "meta") else None
+            if cs is not None and ws is not None                     and isinstance(val, torch.Tensor):
+                new = graph.call_function(
+                    crt.rt_p2p_reshard,
+                    (arg, tuple(val.shape), cs, ws))
+                self._reshard_cache[key] = new
+                return new
         # pass 1 (reverse mesh-dim order): remove sharding / partials
         for d in reversed(range(self.ndim)):
             if self.mesh_shape[d] == 1 or _same(cur[d], want[d]):

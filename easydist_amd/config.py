@@ -103,6 +103,10 @@ comm_optimization = _env_flag("EASYDIST_COMM_OPT")
 rcpsp_method = os.environ.get("EASYDIST_RCPSP_METHOD", "odd_even")
 enable_tile_comm = _env_flag("EASYDIST_TILE_COMM")
 override_dtensor_rule = _env_flag("EASYDIST_OVERRIDE_DTENSOR_RULE")
+# reshard planner for multi-mesh-dim transitions: "auto"/"p2p" use the
+# rectangle-intersection P2P exchange (reference sharding.py:336-612);
+# "greedy" keeps the per-dim collective chain
+reshard_planner = os.environ.get("EASYDIST_RESHARD_PLANNER", "auto")
 
 # ----------------------------------------------------------------- kernels ---
 # Hand-written HIP/CDNA4 kernels for the hot ops. On a GPU box the extension

@@ -328,3 +328,101 @@ def all_gather_bucket_start(tensors, gather_dims=None, group=None):
             off += ne
         return outs
     return _Work(out, work, post)
+
+
+# -------------------------------------------------------- P2P reshard -------
+def _rect_of(shape, placements, coords, mesh_shape):
+    """Global index rectangle [(lo, hi), ...] owned by a rank at `coords`
+    under a per-mesh-dim placement vector (SHARD/REPLICATE only; outer
+    mesh dims chunk first, matching shard_tensor_local)."""
+    rect = [[0, s] for s in shape]
+    for d, p in enumerate(placements):
+        if p[0] != "S":
+            continue
+        td = p[1]
+        lo, hi = rect[td]
+        size = (hi - lo) // mesh_shape[d]
+        lo = lo + coords[d] * size
+        rect[td] = [lo, lo + size]
+    return rect
+
+
+def _intersect(a, b):
+    out = []
+    for (lo1, hi1), (lo2, hi2) in zip(a, b):
+        lo, hi = max(lo1, lo2), min(hi1, hi2)
+        if lo >= hi:
+            return None
+        out.append((lo, hi))
+    return out
+
+
+def p2p_reshard(t: torch.Tensor, global_shape, cur, want, mesh):
+    """Rectangle-intersection P2P reshard (reference sharding.py:336-612
+    re-designed for xGMI: every pairwise move is one direct link hop).
+
+    cur/want: per-mesh-dim placement tuples ("S", td) | ("R",). Valid when
+    both contain only SHARD/REPLICATE; a dim replicated in `cur` picks the
+    sender whose coords match the receiver on that mesh dim."""
+    md = mesh.mesh
+    mesh_shape = tuple(md.shape)
+    ranks = md.mesh.flatten().tolist()
+    me = dist.get_rank()
+    my_idx = ranks.index(me)
+
+    def coords_of(idx):
+        c = []
+        rem = idx
+        for s in reversed(mesh_shape):
+            c.append(rem % s)
+            rem //= s
+        return list(reversed(c))
+
+    my_c = coords_of(my_idx)
+    my_cur = _rect_of(global_shape, cur, my_c, mesh_shape)
+    my_want = _rect_of(global_shape, want, my_c, mesh_shape)
+    out = torch.empty([hi - lo for lo, hi in my_want], dtype=t.dtype,
+                      device=t.device)
+
+    def rel(rect, base):
+        return tuple(slice(lo - blo, hi - blo)
+                     for (lo, hi), (blo, bhi) in zip(rect, base))
+
+    def canonical_pair(sender_c, recv_c):
+        # on mesh dims where cur is replicated, the canonical sender has
+        # the receiver's coordinate (keeps traffic inside subgroups)
+        for d, p in enumerate(cur):
+            if p[0] != "S" and sender_c[d] != recv_c[d]:
+                return False
+        return True
+
+    ops = []
+    recv_bufs = []
+    t = t.contiguous()
+    for idx, r in enumerate(ranks):
+        c = coords_of(idx)
+        if r != me:
+            # what I send to r
+            r_want = _rect_of(global_shape, want, c, mesh_shape)
+            inter = _intersect(my_cur, r_want)
+            if inter is not None and canonical_pair(my_c, c):
+                piece = t[rel(inter, my_cur)].contiguous()
+                ops.append(dist.P2POp(dist.isend, piece, r))
+            # what I receive from r
+            r_cur = _rect_of(global_shape, cur, c, mesh_shape)
+            inter2 = _intersect(r_cur, my_want)
+            if inter2 is not None and canonical_pair(c, my_c):
+                buf = torch.empty([hi - lo for lo, hi in inter2],
+                                  dtype=t.dtype, device=t.device)
+                ops.append(dist.P2POp(dist.irecv, buf, r))
+                recv_bufs.append((inter2, buf))
+        else:
+            inter = _intersect(my_cur, my_want)
+            if inter is not None:
+                out[rel(inter, my_want)] = t[rel(inter, my_cur)]
+    if ops:
+        for wk in dist.batch_isend_irecv(ops):
+            wk.wait()
+    for rect, buf in recv_bufs:
+        out[rel(rect, my_want)] = buf
+    return out

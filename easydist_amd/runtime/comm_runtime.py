@@ -93,3 +93,11 @@ def rt_grouped_wait(w):
 
 COMM_START_TARGETS.add(rt_grouped_all_reduce_start)
 COMM_START_TARGETS.add(rt_grouped_all_gather_start)
+
+
+def rt_p2p_reshard(t, global_shape, cur, want):
+    """Rectangle-intersection P2P reshard over the whole mesh (emitted by
+    the sharding transform for multi-mesh-dim S->S transitions)."""
+    mesh = get_device_mesh()
+    return comm.p2p_reshard(t, list(global_shape), [tuple(p) for p in cur],
+                            [tuple(p) for p in want], mesh)
