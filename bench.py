@@ -68,6 +68,9 @@ def main():
     n_gpus = max(world_size, 1)
     use_cuda = torch.cuda.is_available()
     device = f"cuda:{local_rank}" if use_cuda else "cpu"
+    # the pluggable allocator (EASYDIST_MEM_OPT) must be swapped in
+    # BEFORE anything initializes the CUDA context
+    easydist_setup(backend="torch", device="cuda" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(local_rank)
 
@@ -78,8 +81,6 @@ def main():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29512")
         dist.init_process_group(backend=backend, rank=0, world_size=1)
-
-    easydist_setup(backend="torch", device="cuda" if use_cuda else "cpu")
     set_device_mesh(list(range(world_size)), ["spmd0"])
 
     global_batch = args.per_gpu_batch * n_gpus
