@@ -234,38 +234,48 @@ def fuse_optimizer(gm: fx.GraphModule, flat_outs: List, placeholders: List,
                 return src
         return None
 
+    # grad-cast folding: the autocast trace casts every bf16 grad to
+    # fp32 just for the optimizer. The HIP kernel reads bf16 grads
+    # directly but needs a UNIFORM grad dtype per call — partition the
+    # bank into a bf16-grad call and an fp32-grad call (two launches
+    # instead of ~hundred cast kernels).
     unwrapped = [_uncast_bf16(m[3]) for m in matched]
-    if all(u is not None for u in unwrapped):
-        matched = [(m[0], m[1], m[2], u, m[4], m[5], m[6])
-                   for m, u in zip(matched, unwrapped)]
+    bf16_bank = [(m[0], m[1], m[2], u, m[4], m[5], m[6])
+                 for m, u in zip(matched, unwrapped) if u is not None]
+    fp32_bank = [m for m, u in zip(matched, unwrapped) if u is None]
+    if bf16_bank:
         logger.info("fuse_optimizer: feeding %d bf16 grads directly "
-                    "(folded fp32 casts)", len(matched))
+                    "(folded fp32 casts; %d fp32-grad params in a second "
+                    "bank)", len(bf16_bank), len(fp32_bank))
+    banks = [b for b in (bf16_bank, fp32_bank) if b]
 
     with graph.inserting_before(out_node):
-        p_list = [m[2] for m in matched]
-        if decay_scale is not None:
-            # decoupled decay for the whole bank in one multi-tensor op
-            scaled = graph.call_function(
-                aten._foreach_mul.Scalar, (p_list, decay_scale))
-            p_list = [graph.call_function(operator.getitem, (scaled, i))
-                      for i in range(len(matched))]
-        fused = graph.call_function(
-            torch.ops.easydist_amd.fused_adam_step.default,
-            (p_list,                             # params (maybe pre-decayed)
-             [m[3] for m in matched],            # grads
-             [m[4] for m in matched],            # exp_avgs
-             [m[5] for m in matched],            # exp_avg_sqs
-             [m[6] for m in matched],            # steps (pre-increment)
-             lr, beta1, beta2, kernel_wd, eps))
-        lists = [graph.call_function(operator.getitem, (fused, k))
-                 for k in range(4)]
-        for i, (p_pos, outs, *_rest) in enumerate(matched):
-            items = [graph.call_function(operator.getitem, (lists[k], i))
+        for bank in banks:
+            p_list = [m[2] for m in bank]
+            if decay_scale is not None:
+                # decoupled decay for the bank in one multi-tensor op
+                scaled = graph.call_function(
+                    aten._foreach_mul.Scalar, (p_list, decay_scale))
+                p_list = [graph.call_function(operator.getitem, (scaled, i))
+                          for i in range(len(bank))]
+            fused = graph.call_function(
+                torch.ops.easydist_amd.fused_adam_step.default,
+                (p_list,                         # params (maybe pre-decayed)
+                 [m[3] for m in bank],           # grads
+                 [m[4] for m in bank],           # exp_avgs
+                 [m[5] for m in bank],           # exp_avg_sqs
+                 [m[6] for m in bank],           # steps (pre-increment)
+                 lr, beta1, beta2, kernel_wd, eps))
+            lists = [graph.call_function(operator.getitem, (fused, k))
                      for k in range(4)]
-            flat_outs[outs["param"]] = items[0]
-            flat_outs[outs["exp_avg"]] = items[1]
-            flat_outs[outs["exp_avg_sq"]] = items[2]
-            flat_outs[outs["step"]] = items[3]
+            for i, (p_pos, outs, *_rest) in enumerate(bank):
+                items = [graph.call_function(operator.getitem,
+                                             (lists[k], i))
+                         for k in range(4)]
+                flat_outs[outs["param"]] = items[0]
+                flat_outs[outs["exp_avg"]] = items[1]
+                flat_outs[outs["exp_avg_sq"]] = items[2]
+                flat_outs[outs["step"]] = items[3]
     logger.info("fuse_optimizer: fused %d/%d parameter Adam chains",
                 len(matched), len(param_positions))
     return len(matched)
