@@ -71,7 +71,8 @@ def _kernel_ok(q, k, v):
 def _fwd_cuda(q, k, v, causal):
     ext = load_extension()
     if ext is not None and _kernel_ok(q, k, v):
-        q, k, v = (t.contiguous() for t in (q, k, v))
+        # the kernel takes any last-dim-contiguous layout directly (the
+        # [B,S,H,D] qkv-split views) — no activation-sized copies
         return ext.flash_attn_fwd(q, k, v, causal)
     from . import require_hip_ops
     if _kernel_ok(q, k, v):
@@ -86,8 +87,7 @@ def _bwd_cpu(grad, q, k, v, out, lse, causal):
 def _bwd_cuda(grad, q, k, v, out, lse, causal):
     ext = load_extension()
     if ext is not None and _kernel_ok(q, k, v):
-        q, k, v, out = (t.contiguous() for t in (q, k, v, out))
-        return ext.flash_attn_bwd(grad.contiguous(), q, k, v, out,
+        return ext.flash_attn_bwd(grad, q, k, v, out.contiguous(),
                                   lse.contiguous(), causal)
     return _math_bwd(grad, q, k, v, out, lse, causal)
 

@@ -35,7 +35,9 @@ __global__ void __launch_bounds__(512, (D == 64 ? 4 : 1))  // VGPR cap at D=64 o
 flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                  const bf16* __restrict__ V, bf16* __restrict__ O,
                  float* __restrict__ LSE, int B, int H, int S, bool causal,
-                 float scale) {
+                 float scale,
+                 long qsb, long qsh, long qss, long ksb, long ksh, long kss,
+                 long vsb, long vsh, long vss) {
   // grid: (ceil(S/QBLK), B*H); QBLK = 128 q rows per workgroup, 8 waves
   // of ONE 16-row fragment each.  The round-1 4-wave/RF=2 form ran ONE
   // wave per SIMD (110 KB LDS, 256 threads) — every softmax/LDS stall
@@ -45,11 +47,14 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   constexpr int QBLK = 8 * 16 * RF;
   const int qb0 = blockIdx.x * QBLK;
   const int bh = blockIdx.y;
-  const long base = (long)bh * S * D;
-  const bf16* q = Q + base;
-  const bf16* k = K + base;
-  const bf16* v = V + base;
-  bf16* o = O + base;
+  const int b_ = bh / H, h_ = bh % H;
+  // inputs may be [B,S,H,D]-layout views (the qkv split) — per-tensor
+  // batch/head/row strides avoid the activation-sized .contiguous()
+  // copies the round-1 wrapper made every step
+  const bf16* q = Q + b_ * qsb + h_ * qsh;
+  const bf16* k = K + b_ * ksb + h_ * ksh;
+  const bf16* v = V + b_ * vsb + h_ * vsh;
+  bf16* o = O + (long)bh * S * D;
   float* lse = LSE + (long)bh * S;
 
   const int lane = threadIdx.x % WAVE;
@@ -83,7 +88,7 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     for (int ks = 0; ks < D / 32; ++ks) {
       if (row < S) {
         bf16x8v raw = *reinterpret_cast<const bf16x8v*>(
-            &q[(long)row * D + ks * 32 + fg * 8]);
+            &q[(long)row * qss + ks * 32 + fg * 8]);
         #pragma unroll
         for (int u = 0; u < 8; ++u)
           qf[rf][ks][u] = (__bf16)(bf2f((bf16)raw[u]) * qscale);
@@ -113,8 +118,9 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     #pragma unroll
     for (int pi = 0; pi < PF; ++pi) {
       const int e = threadIdx.x * 8 + pi * (NTHR * 8);
-      kreg[pi] = *reinterpret_cast<const bf16x8*>(&k[(long)t * TILE_K + e]);
-      vreg[pi] = *reinterpret_cast<const bf16x8*>(&v[(long)t * TILE_K + e]);
+      const int row = t * KB + e / D, col = e % D;
+      kreg[pi] = *reinterpret_cast<const bf16x8*>(&k[(long)row * kss + col]);
+      vreg[pi] = *reinterpret_cast<const bf16x8*>(&v[(long)row * vss + col]);
     }
   };
   auto store_tile = [&](int t) {
@@ -275,33 +281,37 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   }
 }
 
-std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
-                                                  const at::Tensor& k,
-                                                  const at::Tensor& v,
+// accept any 4-D view with a contiguous last dim (e.g. the [B,S,H,D]
+// layout the qkv split produces) without materializing a copy
+static inline at::Tensor ed_attn_arg(const at::Tensor& t) {
+  return t.stride(3) == 1 ? t : t.contiguous();
+}
+
+std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q_,
+                                                  const at::Tensor& k_,
+                                                  const at::Tensor& v_,
                                                   bool causal) {
+  auto q = ed_attn_arg(q_), k = ed_attn_arg(k_), v = ed_attn_arg(v_);
   TORCH_CHECK(q.dtype() == at::kBFloat16 && q.dim() == 4);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   TORCH_CHECK(q.sizes() == k.sizes() && q.sizes() == v.sizes(),
               "flash_attn_fwd: q/k/v shapes must match");
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(D == 64 || D == 128, "flash_attn_fwd: D in {64,128}");
   TORCH_CHECK(S % QB == 0, "flash_attn_fwd: S multiple of 64");
-  auto out = at::empty_like(q);
+  auto out = at::empty({B, H, S, D}, q.options());
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid((S + 127) / 128, B * H), block(512);
   size_t lds = (2 * ((size_t)KB * D + (size_t)KB * D) + 8 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
-  if (D == 64)
-    hipLaunchKernelGGL(flash_fwd_kernel<64>, grid, block, lds, stream,
-        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-        (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
-        lse.data_ptr<float>(), B, H, S, causal, scale);
-  else
-    hipLaunchKernelGGL(flash_fwd_kernel<128>, grid, block, lds, stream,
-        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-        (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
-        lse.data_ptr<float>(), B, H, S, causal, scale);
+  auto kern = (D == 64) ? flash_fwd_kernel<64> : flash_fwd_kernel<128>;
+  hipLaunchKernelGGL(kern, grid, block, lds, stream,
+      (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+      (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
+      lse.data_ptr<float>(), B, H, S, causal, scale,
+      q.stride(0), q.stride(1), q.stride(2),
+      k.stride(0), k.stride(1), k.stride(2),
+      v.stride(0), v.stride(1), v.stride(2));
   return {out, lse};
 }
 
@@ -324,17 +334,20 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                     const bf16* __restrict__ K, const bf16* __restrict__ V,
                     const float* __restrict__ LSE,
                     const float* __restrict__ DELTA, bf16* __restrict__ DQ,
-                    int B, int H, int S, bool causal, float scale) {
+                    int B, int H, int S, bool causal, float scale,
+                    long dsb, long dsh, long dss, long qsb, long qsh,
+                    long qss, long ksb, long ksh, long kss, long vsb,
+                    long vsh, long vss) {
   // 8 waves x 16 q rows: two 64-row halves share each staged K/V tile
   // (2x arithmetic intensity vs the 4-wave form) at 6+ waves/SIMD.
   const int qb0 = blockIdx.x * (2 * QB);
   const int bh = blockIdx.y;
-  const long base = (long)bh * S * D;
-  const bf16* dO_ = dO + base;
-  const bf16* q = Q + base;
-  const bf16* k = K + base;
-  const bf16* v = V + base;
-  bf16* dq = DQ + base;
+  const int b_ = bh / H, h_ = bh % H;
+  const bf16* dO_ = dO + b_ * dsb + h_ * dsh;
+  const bf16* q = Q + b_ * qsb + h_ * qsh;
+  const bf16* k = K + b_ * ksb + h_ * ksh;
+  const bf16* v = V + b_ * vsb + h_ * vsh;
+  bf16* dq = DQ + (long)bh * S * D;
   const float* lse = LSE + (long)bh * S;
   const float* delta = DELTA + (long)bh * S;
 
@@ -355,9 +368,9 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   #pragma unroll
   for (int ks = 0; ks < D / 32; ++ks) {
     qf[ks] = *reinterpret_cast<const bf16x8v*>(
-        &q[(long)(qr0 + fr) * D + ks * 32 + fg * 8]);
+        &q[(long)(qr0 + fr) * qss + ks * 32 + fg * 8]);
     dof[ks] = *reinterpret_cast<const bf16x8v*>(
-        &dO_[(long)(qr0 + fr) * D + ks * 32 + fg * 8]);
+        &dO_[(long)(qr0 + fr) * dss + ks * 32 + fg * 8]);
   }
   // per C-row lse/delta (rows 4*fg+r of this wave)
   float lse_r[4], dlt_r[4];
@@ -493,17 +506,20 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                      const float* __restrict__ LSE,
                      const float* __restrict__ DELTA, bf16* __restrict__ DK,
                      bf16* __restrict__ DV, int B, int H, int S, bool causal,
-                     float scale) {
+                     float scale,
+                     long dsb, long dsh, long dss, long qsb, long qsh,
+                     long qss, long ksb, long ksh, long kss, long vsb,
+                     long vsh, long vss) {
   // 8 waves x 16 key rows share each staged Q/dO tile (see dq note)
   const int kb0 = blockIdx.x * (2 * KB);
   const int bh = blockIdx.y;
-  const long base = (long)bh * S * D;
-  const bf16* dO_ = dO + base;
-  const bf16* q = Q + base;
-  const bf16* k = K + base;
-  const bf16* v = V + base;
-  bf16* dk = DK + base;
-  bf16* dv = DV + base;
+  const int b_ = bh / H, h_ = bh % H;
+  const bf16* dO_ = dO + b_ * dsb + h_ * dsh;
+  const bf16* q = Q + b_ * qsb + h_ * qsh;
+  const bf16* k = K + b_ * ksb + h_ * ksh;
+  const bf16* v = V + b_ * vsb + h_ * vsh;
+  bf16* dk = DK + (long)bh * S * D;
+  bf16* dv = DV + (long)bh * S * D;
   const float* lse = LSE + (long)bh * S;
   const float* delta = DELTA + (long)bh * S;
 
@@ -524,9 +540,9 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   #pragma unroll
   for (int ks = 0; ks < D / 32; ++ks) {
     kf[ks] = *reinterpret_cast<const bf16x8v*>(
-        &k[(long)(kr0 + fr) * D + ks * 32 + fg * 8]);
+        &k[(long)(kr0 + fr) * kss + ks * 32 + fg * 8]);
     vf[ks] = *reinterpret_cast<const bf16x8v*>(
-        &v[(long)(kr0 + fr) * D + ks * 32 + fg * 8]);
+        &v[(long)(kr0 + fr) * vss + ks * 32 + fg * 8]);
   }
 
   f32x4 dk_acc[D / 16], dv_acc[D / 16];
@@ -688,7 +704,8 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
 template <int D>
 __global__ void __launch_bounds__(256)
 attn_delta_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ O,
-                  float* __restrict__ delta, long rows) {
+                  float* __restrict__ delta, long rows, int H, int S,
+                  long dsb, long dsh, long dss) {
   constexpr int LPR = D / 8;            // lanes per row (16B each)
   const int lane = threadIdx.x % WAVE;
   const int wave = threadIdx.x / WAVE;
@@ -696,7 +713,10 @@ attn_delta_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ O,
   long row = (long)blockIdx.x * (4 * RPW) + wave * RPW + lane / LPR;
   if (row >= rows) return;
   const int c0 = (lane % LPR) * 8;
-  const bf16x8 d8 = *reinterpret_cast<const bf16x8*>(&dO[row * D + c0]);
+  const long b_ = row / ((long)H * S), hs = row % ((long)H * S);
+  const long h_ = hs / S, s_ = hs % S;
+  const bf16x8 d8 = *reinterpret_cast<const bf16x8*>(
+      &dO[b_ * dsb + h_ * dsh + s_ * dss + c0]);
   const bf16x8 o8 = *reinterpret_cast<const bf16x8*>(&O[row * D + c0]);
   float acc = 0.f;
   #pragma unroll
@@ -708,16 +728,17 @@ attn_delta_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ O,
 }
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor>
-flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
-               const at::Tensor& k, const at::Tensor& v,
+flash_attn_bwd(const at::Tensor& grad_, const at::Tensor& q_,
+               const at::Tensor& k_, const at::Tensor& v_,
                const at::Tensor& out, const at::Tensor& lse, bool causal) {
+  auto grad = ed_attn_arg(grad_);
+  auto q = ed_attn_arg(q_), k = ed_attn_arg(k_), v = ed_attn_arg(v_);
   TORCH_CHECK(q.dtype() == at::kBFloat16 && q.dim() == 4);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   TORCH_CHECK(q.sizes() == k.sizes() && q.sizes() == v.sizes());
+  TORCH_CHECK(out.is_contiguous(), "flash_attn_bwd: out must be contiguous");
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(D == 64 || D == 128, "flash_attn_bwd: D in {64,128}");
   TORCH_CHECK(S % (2 * QB) == 0, "flash_attn_bwd: S multiple of 128");
-  auto gradc = grad.contiguous();
   // delta = rowsum(dO * O), fp32 — one fused bf16 pass
   auto delta = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   {
@@ -725,48 +746,41 @@ flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
     int rpw = (D == 64) ? 8 : 4;
     long nblk = (rows + 4 * rpw - 1) / (4 * rpw);
     auto stream0 = at::cuda::getCurrentCUDAStream();
-    if (D == 64)
-      hipLaunchKernelGGL(attn_delta_kernel<64>, dim3(nblk), dim3(256), 0,
-          stream0, (const bf16*)gradc.data_ptr(), (const bf16*)out.data_ptr(),
-          delta.data_ptr<float>(), rows);
-    else
-      hipLaunchKernelGGL(attn_delta_kernel<128>, dim3(nblk), dim3(256), 0,
-          stream0, (const bf16*)gradc.data_ptr(), (const bf16*)out.data_ptr(),
-          delta.data_ptr<float>(), rows);
+    auto dkern = (D == 64) ? attn_delta_kernel<64> : attn_delta_kernel<128>;
+    hipLaunchKernelGGL(dkern, dim3(nblk), dim3(256), 0,
+        stream0, (const bf16*)grad.data_ptr(), (const bf16*)out.data_ptr(),
+        delta.data_ptr<float>(), rows, H, S,
+        grad.stride(0), grad.stride(1), grad.stride(2));
   }
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
+  auto dq = at::empty({B, H, S, D}, q.options());
+  auto dk = at::empty({B, H, S, D}, q.options());
+  auto dv = at::empty({B, H, S, D}, q.options());
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid(S / (2 * QB), B * H), block(512);
   size_t lds = (2 * KB * (size_t)D + (size_t)KB * D + 8 * 16 * KB) * 2;
   size_t lds_kv = (2 * QB * (size_t)D + 2 * (size_t)QB * D
                    + 8 * 16 * QB) * 2;
   float scale = 1.f / sqrtf((float)D);
-  if (D == 64) {
-    hipLaunchKernelGGL(flash_bwd_dq_kernel<64>, grid, block, lds, stream,
-        (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
-        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
-        lse.data_ptr<float>(), delta.data_ptr<float>(),
-        (bf16*)dq.data_ptr(), B, H, S, causal, scale);
-    hipLaunchKernelGGL(flash_bwd_dkv_kernel<64>, grid, block, lds_kv,
-        stream,
-        (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
-        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
-        lse.data_ptr<float>(), delta.data_ptr<float>(),
-        (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(), B, H, S, causal, scale);
-  } else {
-    hipLaunchKernelGGL(flash_bwd_dq_kernel<128>, grid, block, lds, stream,
-        (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
-        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
-        lse.data_ptr<float>(), delta.data_ptr<float>(),
-        (bf16*)dq.data_ptr(), B, H, S, causal, scale);
-    hipLaunchKernelGGL(flash_bwd_dkv_kernel<128>, grid, block, lds_kv,
-        stream,
-        (const bf16*)gradc.data_ptr(), (const bf16*)q.data_ptr(),
-        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
-        lse.data_ptr<float>(), delta.data_ptr<float>(),
-        (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(), B, H, S, causal, scale);
-  }
+  auto qkern = (D == 64) ? flash_bwd_dq_kernel<64> : flash_bwd_dq_kernel<128>;
+  auto kkern = (D == 64) ? flash_bwd_dkv_kernel<64>
+                         : flash_bwd_dkv_kernel<128>;
+  hipLaunchKernelGGL(qkern, grid, block, lds, stream,
+      (const bf16*)grad.data_ptr(), (const bf16*)q.data_ptr(),
+      (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
+      lse.data_ptr<float>(), delta.data_ptr<float>(),
+      (bf16*)dq.data_ptr(), B, H, S, causal, scale,
+      grad.stride(0), grad.stride(1), grad.stride(2),
+      q.stride(0), q.stride(1), q.stride(2),
+      k.stride(0), k.stride(1), k.stride(2),
+      v.stride(0), v.stride(1), v.stride(2));
+  hipLaunchKernelGGL(kkern, grid, block, lds_kv, stream,
+      (const bf16*)grad.data_ptr(), (const bf16*)q.data_ptr(),
+      (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
+      lse.data_ptr<float>(), delta.data_ptr<float>(),
+      (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(), B, H, S, causal, scale,
+      grad.stride(0), grad.stride(1), grad.stride(2),
+      q.stride(0), q.stride(1), q.stride(2),
+      k.stride(0), k.stride(1), k.stride(2),
+      v.stride(0), v.stride(1), v.stride(2));
   return {dq, dk, dv};
 }
