@@ -184,20 +184,33 @@ __global__ void norm_param_grads_kernel(const T* __restrict__ grad,
                                         float* __restrict__ dw,
                                         float* __restrict__ db,
                                         int rows, int D, int rows_per_blk) {
-  const int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= D) return;
+  // 8 columns per thread with 16-B vector loads: the scalar-bf16 form
+  // streamed at half rate (guide common-mistake 2)
+  const int col0 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (col0 >= D) return;
   const int r0 = blockIdx.y * rows_per_blk;
   const int r1 = min(r0 + rows_per_blk, rows);
-  float sw = 0.f, sb = 0.f;
+  float sw[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float sb[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   for (int r = r0; r < r1; ++r) {
-    float g = bf2f(grad[(long)r * D + col]);
-    float m = RMS ? 0.f : mean[r];
-    float xh = (bf2f(x[(long)r * D + col]) - m) * rstd[r];
-    sw += g * xh;
-    sb += g;
+    const bf16x8 g8 = *reinterpret_cast<const bf16x8*>(
+        &grad[(long)r * D + col0]);
+    const bf16x8 x8 = *reinterpret_cast<const bf16x8*>(
+        &x[(long)r * D + col0]);
+    const float m = RMS ? 0.f : mean[r];
+    const float rs = rstd[r];
+    #pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      float g = bf2f(g8.v[u]);
+      sw[u] += g * (bf2f(x8.v[u]) - m) * rs;
+      sb[u] += g;
+    }
   }
-  atomicAdd(&dw[col], sw);
-  atomicAdd(&db[col], sb);
+  #pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    atomicAdd(&dw[col0 + u], sw[u]);
+    atomicAdd(&db[col0 + u], sb[u]);
+  }
 }
 
 // ---------------------------------------------------------- host wrappers ---
@@ -269,8 +282,13 @@ layer_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
       w ? (const bf16*)w->data_ptr() : nullptr, (bf16*)dx.data_ptr(),
       dwf.data_ptr<float>(), dbf.data_ptr<float>(), rows, D);
-  const int rows_per_blk = 256;
-  dim3 gblock(256), ggrid((D + 255) / 256, (rows + rows_per_blk - 1) / rows_per_blk);
+  // fill the 256 CUs: x covers D/8 columns per 256-thread block; y
+  // splits rows so x*y lands near 1024 blocks
+  const int gx = (D / 8 + 255) / 256;
+  int rows_per_blk = 64;
+  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 2048)
+    rows_per_blk *= 2;
+  dim3 gblock(256), ggrid(gx, (rows + rows_per_blk - 1) / rows_per_blk);
   hipLaunchKernelGGL((norm_param_grads_kernel<bf16, false>), ggrid, gblock, 0,
       stream, (const bf16*)grad.data_ptr(), (const bf16*)x.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(), dwf.data_ptr<float>(),
@@ -327,8 +345,13 @@ rms_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
       rstd.data_ptr<float>(), w ? (const bf16*)w->data_ptr() : nullptr,
       (bf16*)dx.data_ptr(), dwf.data_ptr<float>(), dbf.data_ptr<float>(),
       rows, D);
-  const int rows_per_blk = 256;
-  dim3 gblock(256), ggrid((D + 255) / 256, (rows + rows_per_blk - 1) / rows_per_blk);
+  // fill the 256 CUs: x covers D/8 columns per 256-thread block; y
+  // splits rows so x*y lands near 1024 blocks
+  const int gx = (D / 8 + 255) / 256;
+  int rows_per_blk = 64;
+  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 2048)
+    rows_per_blk *= 2;
+  dim3 gblock(256), ggrid(gx, (rows + rows_per_blk - 1) / rows_per_blk);
   hipLaunchKernelGGL((norm_param_grads_kernel<bf16, true>), ggrid, gblock, 0,
       stream, (const bf16*)grad.data_ptr(), (const bf16*)x.data_ptr(),
       nullptr, rstd.data_ptr<float>(), dwf.data_ptr<float>(),

@@ -197,9 +197,11 @@ def _nt_act_cuda(a, bt, bias, act, aux):
     bt = bt.contiguous()
     b_c = bias.contiguous() if bias is not None else None
     x_c = aux.contiguous() if aux is not None else None
-    key = ("nt", a.shape[0], a.shape[1], bt.shape[0], bias is not None)
-    if _choose(key, lambda: ext.gemm_nt(a, bt, b_c),
-               lambda: _nt_cpu(a, bt, bias)):
+    # the comparison must price the FUSION: choosing aten re-adds the
+    # separate activation kernel the epilogue absorbs
+    key = ("nt", a.shape[0], a.shape[1], bt.shape[0], bias is not None, act)
+    if _choose(key, lambda: ext.gemm_nt_act(a, bt, b_c, act, x_c),
+               lambda: _nt_act_cpu(a, bt, bias, act, aux)):
         return ext.gemm_nt_act(a, bt, b_c, act, x_c)
     return _nt_act_cpu(a, bt, bias, act, aux)
 
@@ -217,8 +219,9 @@ def _tn_asum_cuda(a, b):
         return _tn_asum_cpu(a, b)
     a = a.contiguous()
     b = b.contiguous()
-    key = ("tn", a.shape[0], a.shape[1], b.shape[1])
-    if _choose(key, lambda: ext.gemm_tn(a, b), lambda: _tn_cpu(a, b)):
+    key = ("tn_asum", a.shape[0], a.shape[1], b.shape[1])
+    if _choose(key, lambda: ext.gemm_tn_asum(a, b),
+               lambda: _tn_asum_cpu(a, b)):
         return ext.gemm_tn_asum(a, b)
     return _tn_asum_cpu(a, b)
 
