@@ -282,13 +282,13 @@ layer_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
       w ? (const bf16*)w->data_ptr() : nullptr, (bf16*)dx.data_ptr(),
       dwf.data_ptr<float>(), dbf.data_ptr<float>(), rows, D);
-  // fill the 256 CUs: x covers D/8 columns per 256-thread block; y
-  // splits rows so x*y lands near 1024 blocks
-  const int gx = (D / 8 + 255) / 256;
-  int rows_per_blk = 64;
-  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 2048)
+  // one-wave blocks, 512 columns each (64 lanes x 8 cols of 16-B
+  // loads); y splits rows so x*y fills the 256 CUs
+  const int gx = (D + 511) / 512;
+  int rows_per_blk = 16;
+  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 4096)
     rows_per_blk *= 2;
-  dim3 gblock(256), ggrid(gx, (rows + rows_per_blk - 1) / rows_per_blk);
+  dim3 gblock(64), ggrid(gx, (rows + rows_per_blk - 1) / rows_per_blk);
   hipLaunchKernelGGL((norm_param_grads_kernel<bf16, false>), ggrid, gblock, 0,
       stream, (const bf16*)grad.data_ptr(), (const bf16*)x.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(), dwf.data_ptr<float>(),
@@ -345,13 +345,13 @@ rms_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
       rstd.data_ptr<float>(), w ? (const bf16*)w->data_ptr() : nullptr,
       (bf16*)dx.data_ptr(), dwf.data_ptr<float>(), dbf.data_ptr<float>(),
       rows, D);
-  // fill the 256 CUs: x covers D/8 columns per 256-thread block; y
-  // splits rows so x*y lands near 1024 blocks
-  const int gx = (D / 8 + 255) / 256;
-  int rows_per_blk = 64;
-  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 2048)
+  // one-wave blocks, 512 columns each (64 lanes x 8 cols of 16-B
+  // loads); y splits rows so x*y fills the 256 CUs
+  const int gx = (D + 511) / 512;
+  int rows_per_blk = 16;
+  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 4096)
     rows_per_blk *= 2;
-  dim3 gblock(256), ggrid(gx, (rows + rows_per_blk - 1) / rows_per_blk);
+  dim3 gblock(64), ggrid(gx, (rows + rows_per_blk - 1) / rows_per_blk);
   hipLaunchKernelGGL((norm_param_grads_kernel<bf16, true>), ggrid, gblock, 0,
       stream, (const bf16*)grad.data_ptr(), (const bf16*)x.data_ptr(),
       nullptr, rstd.data_ptr<float>(), dwf.data_ptr<float>(),
