@@ -6,6 +6,8 @@
 // accumulate into fp32 global scratch with device-scope atomics.
 // Replaces aten.native_layer_norm(+backward) via the lower_hip pass.
 #include "common.h"
+#include <string>
+#include <cstdlib>
 
 // ---------------------------------------------------------------- fwd -------
 template <typename T, bool RMS>
@@ -213,6 +215,42 @@ __global__ void norm_param_grads_kernel(const T* __restrict__ grad,
   }
 }
 
+template <typename T, bool RMS>
+__global__ void norm_param_grads_scalar_kernel(const T* __restrict__ grad,
+                                               const T* __restrict__ x,
+                                               const float* __restrict__ mean,
+                                               const float* __restrict__ rstd,
+                                               float* __restrict__ dw,
+                                               float* __restrict__ db,
+                                               int rows, int D,
+                                               int rows_per_blk) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= D) return;
+  const int r0 = blockIdx.y * rows_per_blk;
+  const int r1 = min(r0 + rows_per_blk, rows);
+  float sw = 0.f, sb = 0.f;
+  for (int r = r0; r < r1; ++r) {
+    float g = bf2f(grad[(long)r * D + col]);
+    float m = RMS ? 0.f : mean[r];
+    float xh = (bf2f(x[(long)r * D + col]) - m) * rstd[r];
+    sw += g * xh;
+    sb += g;
+  }
+  atomicAdd(&dw[col], sw);
+  atomicAdd(&db[col], sb);
+}
+
+static bool lnpg_scalar() {
+  // measured on MI355X (65536x768): scalar-column form 0.166 ms/call vs
+  // 0.221 for the vectorized+fewer-atomics form — per-address atomic
+  // count and full-block occupancy beat wider loads here
+  static int v = [] {
+    const char* e = getenv("EASYDIST_LNPG");
+    return (e && std::string(e) == "vector") ? 0 : 1;
+  }();
+  return v;
+}
+
 // ---------------------------------------------------------- host wrappers ---
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
@@ -285,10 +323,23 @@ layer_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
   // one-wave blocks, 512 columns each (64 lanes x 8 cols of 16-B
   // loads); y splits rows so x*y fills the 256 CUs
   const int gx = (D + 511) / 512;
-  int rows_per_blk = 16;
-  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 4096)
+  // few row-chunks: every block atomicAdds the whole 2*D output, so
+  // block count IS the atomic contention per address
+  int rows_per_blk = 256;
+  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 1024)
     rows_per_blk *= 2;
   dim3 gblock(64), ggrid(gx, (rows + rows_per_blk - 1) / rows_per_blk);
+  if (lnpg_scalar()) {
+    const int rpb2 = 256;
+    dim3 sb2(256), sg2((D + 255) / 256, (rows + rpb2 - 1) / rpb2);
+    hipLaunchKernelGGL((norm_param_grads_scalar_kernel<bf16, false>), sg2,
+        sb2, 0, stream, (const bf16*)grad.data_ptr(),
+        (const bf16*)x.data_ptr(), mean.data_ptr<float>(),
+        rstd.data_ptr<float>(), dwf.data_ptr<float>(),
+        dbf.data_ptr<float>(), rows, D, rpb2);
+    auto dtype0 = at::kFloat;
+    return {dx, dwf.to(dtype0), dbf.to(dtype0)};
+  }
   hipLaunchKernelGGL((norm_param_grads_kernel<bf16, false>), ggrid, gblock, 0,
       stream, (const bf16*)grad.data_ptr(), (const bf16*)x.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(), dwf.data_ptr<float>(),
@@ -348,8 +399,10 @@ rms_norm_bwd(const at::Tensor& grad, const at::Tensor& x,
   // one-wave blocks, 512 columns each (64 lanes x 8 cols of 16-B
   // loads); y splits rows so x*y fills the 256 CUs
   const int gx = (D + 511) / 512;
-  int rows_per_blk = 16;
-  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 4096)
+  // few row-chunks: every block atomicAdds the whole 2*D output, so
+  // block count IS the atomic contention per address
+  int rows_per_blk = 256;
+  while ((long)gx * ((rows + rows_per_blk - 1) / rows_per_blk) > 1024)
     rows_per_blk *= 2;
   dim3 gblock(64), ggrid(gx, (rows + rows_per_blk - 1) / rows_per_blk);
   hipLaunchKernelGGL((norm_param_grads_kernel<bf16, true>), ggrid, gblock, 0,
