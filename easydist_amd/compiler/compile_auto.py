@@ -77,15 +77,18 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
     comm_optimize(gm)
 
     # ---- 5b2. lower hot aten ops to the gfx950 kernels -------------------
-    from .passes.lower_hip import (lower_bias_grad_fuse, lower_cross_entropy,
-                                   lower_gelu_bwd_fuse, lower_gemm,
+    from .passes.lower_hip import (lower_attn_pack, lower_bias_grad_fuse,
+                                   lower_cross_entropy, lower_gelu_bwd_fuse,
+                                   lower_gelu_fwd_fuse, lower_gemm,
                                    lower_layer_norm, lower_sdpa)
     lower_layer_norm(gm)
     lower_sdpa(gm)
+    lower_attn_pack(gm)
     lower_cross_entropy(gm)
     if mdconfig.hip_gemm:
         lower_gemm(gm)
         lower_gelu_bwd_fuse(gm)
+        lower_gelu_fwd_fuse(gm)
         lower_bias_grad_fuse(gm)
 
     # ---- 5c. re-fuse the decomposed Adam chains into ONE kernel ----------

@@ -645,3 +645,172 @@ def lower_bias_grad_fuse(gm: fx.GraphModule) -> int:
         logger.info("lower_hip: fused %d bias-grad sums into TN GEMMs",
                     n_fused)
     return n_fused
+
+
+def lower_attn_pack(gm: fx.GraphModule) -> int:
+    """Collapse the dq/dk/dv repack chain into a packed flash backward.
+
+    The qkv projection's autograd materializes
+
+        cat([unsafe_view(clone(transpose(flash_bwd[i], 1, 2)))
+             for i in (0, 1, 2)], dim=2)
+
+    — three strided [B,H,S,D]->[B,S,H*D] gathers plus a cat copy, ~1.2 GB
+    of pure layout traffic per GPT layer at batch 64 (reference keeps the
+    equivalent aten chain: easydist/torch/experimental/pp/split_utils.py
+    has no such fusion). flash_attention_bwd_pack writes dq/dk/dv through
+    output strides straight into one [B, S, 3*H*D] buffer, so the whole
+    chain becomes one node. Run AFTER lower_sdpa.
+    """
+    flash_bwd = torch.ops.easydist_amd.flash_attention_bwd.default
+    flash_pack = torch.ops.easydist_amd.flash_attention_bwd_pack.default
+    graph = gm.graph
+    n_fused = 0
+
+    def _single_user(n):
+        return len(n.users) == 1
+
+    for n in list(graph.nodes):
+        if n.op != "call_function" or n.target is not aten.cat.default:
+            continue
+        parts = n.args[0]
+        dim = n.args[1] if len(n.args) > 1 else 0
+        if len(parts) != 3 or dim not in (2, -1):
+            continue
+        fab = None
+        chains = []
+        ok = True
+        for want_idx, view in enumerate(parts):
+            # view(...chain of (1,2)-transposes and clones...) — the model's
+            # [B,H,S,D]->[B,S,H*D] repack, possibly with lower_sdpa's
+            # stride-relayout (transpose+clone+transpose) in between. A NET
+            # odd transpose count is the [B,S,...] layout the cat needs.
+            chain = [view]
+            m = view
+            if m.op != "call_function" or m.target not in (
+                    aten._unsafe_view.default, aten.view.default):
+                ok = False
+                break
+            m = m.args[0]
+            n_t = 0
+            gi = None
+            while True:
+                if not isinstance(m, fx.Node) or not _single_user(m):
+                    break
+                if (m.target is aten.transpose.int
+                        and set(m.args[1:]) == {1, 2}):
+                    n_t += 1
+                elif m.target is aten.clone.default:
+                    pass
+                elif m.target is operator.getitem:
+                    gi = m
+                    break
+                else:
+                    break
+                chain.append(m)
+                if len(chain) > 8:
+                    break
+                m = m.args[0]
+            if gi is None or n_t % 2 == 0 or gi.args[1] != want_idx:
+                ok = False
+                break
+            chain.append(gi)
+            src = gi.args[0]
+            if src.op != "call_function" or src.target is not flash_bwd:
+                ok = False
+                break
+            if fab is None:
+                fab = src
+            elif fab is not src:
+                ok = False
+                break
+            chains.append(chain)
+        if not ok or fab is None:
+            continue
+        # every flash_bwd consumer must be one of the three getitems
+        gitems = {c[-1] for c in chains}
+        if set(fab.users) != gitems:
+            continue
+        # cat must be each view's only consumer (views feed nothing else)
+        if any(len(v.users) != 1 for v in parts):
+            continue
+        with graph.inserting_before(fab):
+            pack = graph.call_function(flash_pack, tuple(fab.args))
+        pack.meta = dict(n.meta)
+        n.replace_all_uses_with(pack)
+        graph.erase_node(n)
+        for chain in chains:
+            for m in chain:
+                graph.erase_node(m)
+        graph.erase_node(fab)
+        n_fused += 1
+
+    if n_fused:
+        graph.lint()
+        gm.recompile()
+        logger.info("lower_hip: packed %d flash-bwd repack chains", n_fused)
+    return n_fused
+
+
+def lower_gelu_fwd_fuse(gm: fx.GraphModule) -> int:
+    """Fuse the forward gelu into the MFMA GEMM that produces its input.
+
+    Pattern (after lower_gemm):
+
+        pre = gemm_nt(x2, w, bias)      # [M, N]
+        v   = view(pre, [b, s, N])
+        y   = gelu(v, approximate=...)
+
+    becomes ``act2, pre = gemm_nt_gelu(x2, w, bias, tanh)`` with ``y``
+    re-created as a view of ``act2``. The epilogue applies gelu during
+    the (already coalesced) store pass and writes the pre-activation as a
+    second output for gelu_backward, so the standalone activation kernel
+    and its HBM re-read of the GEMM result disappear.
+    """
+    gemm_nt = torch.ops.easydist_amd.gemm_nt.default
+    gemm_nt_gelu = torch.ops.easydist_amd.gemm_nt_gelu.default
+    graph = gm.graph
+    n_fused = 0
+    for n in list(graph.nodes):
+        if n.op != "call_function" or n.target is not aten.gelu.default:
+            continue
+        approx = "none"
+        if len(n.args) > 1:
+            approx = n.args[1]
+        approx = n.kwargs.get("approximate", approx)
+        v = n.args[0]
+        views = (aten.view.default, aten._unsafe_view.default)
+        if v.op == "call_function" and v.target in views:
+            gnt = v.args[0]
+            view_args = v.args[1]
+        else:
+            gnt = v
+            view_args = None
+        if gnt.op != "call_function" or gnt.target is not gemm_nt:
+            continue
+        with graph.inserting_before(gnt):
+            fused = graph.call_function(
+                gemm_nt_gelu, tuple(gnt.args) + (approx == "tanh",))
+            act2 = graph.call_function(operator.getitem, (fused, 0))
+            pre2 = graph.call_function(operator.getitem, (fused, 1))
+        act2.meta = dict(gnt.meta)
+        pre2.meta = dict(gnt.meta)
+        gnt.replace_all_uses_with(pre2)
+        if view_args is not None:
+            with graph.inserting_before(n):
+                act_v = graph.call_function(aten.view.default,
+                                            (act2, view_args))
+            act_v.meta = dict(n.meta)
+            n.replace_all_uses_with(act_v)
+        else:
+            n.replace_all_uses_with(act2)
+        graph.erase_node(n)
+        graph.erase_node(gnt)
+        n_fused += 1
+    if n_fused:
+        graph.eliminate_dead_code()
+        graph.lint()
+        gm.recompile()
+        logger.info("lower_hip: fused %d forward gelu into GEMM epilogues",
+                    n_fused)
+    return n_fused

@@ -22,6 +22,11 @@ lib.define("flash_attention(Tensor q, Tensor k, Tensor v, bool causal) "
            "-> (Tensor, Tensor)")
 lib.define("flash_attention_bwd(Tensor grad, Tensor q, Tensor k, Tensor v, "
            "Tensor out, Tensor lse, bool causal) -> (Tensor, Tensor, Tensor)")
+# packed variant: dq/dk/dv written straight into one [B, S, 3*H*D] buffer
+# (the qkv-projection-backward layout) — the lowering pass substitutes it
+# for the flash_bwd -> transpose -> clone -> cat chain (lower_hip.py)
+lib.define("flash_attention_bwd_pack(Tensor grad, Tensor q, Tensor k, "
+           "Tensor v, Tensor out, Tensor lse, bool causal) -> Tensor")
 
 
 def _math_fwd(q, k, v, causal):
@@ -92,10 +97,30 @@ def _bwd_cuda(grad, q, k, v, out, lse, causal):
     return _math_bwd(grad, q, k, v, out, lse, causal)
 
 
+def _pack3(dq, dk, dv):
+    B, H, S, D = dq.shape
+    return torch.cat([d.transpose(1, 2).reshape(B, S, H * D)
+                      for d in (dq, dk, dv)], dim=-1)
+
+
+def _bwd_pack_cpu(grad, q, k, v, out, lse, causal):
+    return _pack3(*_math_bwd(grad, q, k, v, out, lse, causal))
+
+
+def _bwd_pack_cuda(grad, q, k, v, out, lse, causal):
+    ext = load_extension()
+    if ext is not None and _kernel_ok(q, k, v):
+        return ext.flash_attn_bwd_pack(grad, q, k, v, out.contiguous(),
+                                       lse.contiguous(), causal)
+    return _pack3(*_math_bwd(grad, q, k, v, out, lse, causal))
+
+
 lib.impl("flash_attention", _fwd_cpu, "CPU")
 lib.impl("flash_attention", _fwd_cuda, "CUDA")
 lib.impl("flash_attention_bwd", _bwd_cpu, "CPU")
 lib.impl("flash_attention_bwd", _bwd_cuda, "CUDA")
+lib.impl("flash_attention_bwd_pack", _bwd_pack_cpu, "CPU")
+lib.impl("flash_attention_bwd_pack", _bwd_pack_cuda, "CUDA")
 
 
 @torch.library.register_fake("easydist_amd::flash_attention")
@@ -111,6 +136,12 @@ def _fab_fake(grad, q, k, v, out, lse, causal):
     # contiguous outputs (matmul results), independent of input strides
     return (q.new_empty(tuple(q.shape)), k.new_empty(tuple(k.shape)),
             v.new_empty(tuple(v.shape)))
+
+
+@torch.library.register_fake("easydist_amd::flash_attention_bwd_pack")
+def _fabp_fake(grad, q, k, v, out, lse, causal):
+    B, H, S, D = q.shape
+    return q.new_empty((B, S, 3 * H * D))
 
 
 def _fa_backward(ctx, grad_out, grad_lse):

@@ -337,7 +337,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                     int B, int H, int S, bool causal, float scale,
                     long dsb, long dsh, long dss, long qsb, long qsh,
                     long qss, long ksb, long ksh, long kss, long vsb,
-                    long vsh, long vss) {
+                    long vsh, long vss, long osb, long osh, long oss) {
   // 8 waves x 16 q rows: two 64-row halves share each staged K/V tile
   // (2x arithmetic intensity vs the 4-wave form) at 6+ waves/SIMD.
   const int qb0 = blockIdx.x * (2 * QB);
@@ -347,7 +347,9 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   const bf16* q = Q + b_ * qsb + h_ * qsh;
   const bf16* k = K + b_ * ksb + h_ * ksh;
   const bf16* v = V + b_ * vsb + h_ * vsh;
-  bf16* dq = DQ + (long)bh * S * D;
+  // output strides (elements): the packed-dqkv variant writes straight
+  // into a [B,S,3,H,D] buffer, eliminating the transpose-clone-cat chain
+  bf16* dq = DQ + b_ * osb + h_ * osh;
   const float* lse = LSE + (long)bh * S;
   const float* delta = DELTA + (long)bh * S;
 
@@ -495,7 +497,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     if (qrow >= S) continue;
     #pragma unroll
     for (int j = 0; j < D / 16; ++j)
-      dq[(long)qrow * D + j * 16 + fr] = f2bf(dq_acc[j][r]);
+      dq[(long)qrow * oss + j * 16 + fr] = f2bf(dq_acc[j][r]);
   }
 }
 
@@ -509,7 +511,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                      float scale,
                      long dsb, long dsh, long dss, long qsb, long qsh,
                      long qss, long ksb, long ksh, long kss, long vsb,
-                     long vsh, long vss) {
+                     long vsh, long vss, long osb, long osh, long oss) {
   // 8 waves x 16 key rows share each staged Q/dO tile (see dq note)
   const int kb0 = blockIdx.x * (2 * KB);
   const int bh = blockIdx.y;
@@ -518,8 +520,8 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   const bf16* q = Q + b_ * qsb + h_ * qsh;
   const bf16* k = K + b_ * ksb + h_ * ksh;
   const bf16* v = V + b_ * vsb + h_ * vsh;
-  bf16* dk = DK + (long)bh * S * D;
-  bf16* dv = DV + (long)bh * S * D;
+  bf16* dk = DK + b_ * osb + h_ * osh;
+  bf16* dv = DV + b_ * osb + h_ * osh;
   const float* lse = LSE + (long)bh * S;
   const float* delta = DELTA + (long)bh * S;
 
@@ -692,8 +694,8 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     if (krow >= S) continue;
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
-      dk[(long)krow * D + j * 16 + fr] = f2bf(dk_acc[j][r]);
-      dv[(long)krow * D + j * 16 + fr] = f2bf(dv_acc[j][r]);
+      dk[(long)krow * oss + j * 16 + fr] = f2bf(dk_acc[j][r]);
+      dv[(long)krow * oss + j * 16 + fr] = f2bf(dv_acc[j][r]);
     }
   }
 }
@@ -727,10 +729,12 @@ attn_delta_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ O,
   if (lane % LPR == 0) delta[row] = acc;
 }
 
-std::tuple<at::Tensor, at::Tensor, at::Tensor>
-flash_attn_bwd(const at::Tensor& grad_, const at::Tensor& q_,
-               const at::Tensor& k_, const at::Tensor& v_,
-               const at::Tensor& out, const at::Tensor& lse, bool causal) {
+static void
+flash_attn_bwd_launch(const at::Tensor& grad_, const at::Tensor& q_,
+                      const at::Tensor& k_, const at::Tensor& v_,
+                      const at::Tensor& out, const at::Tensor& lse,
+                      bool causal, bf16* dq_p, bf16* dk_p, bf16* dv_p,
+                      long osb, long osh, long oss) {
   auto grad = ed_attn_arg(grad_);
   auto q = ed_attn_arg(q_), k = ed_attn_arg(k_), v = ed_attn_arg(v_);
   TORCH_CHECK(q.dtype() == at::kBFloat16 && q.dim() == 4);
@@ -752,9 +756,6 @@ flash_attn_bwd(const at::Tensor& grad_, const at::Tensor& q_,
         delta.data_ptr<float>(), rows, H, S,
         grad.stride(0), grad.stride(1), grad.stride(2));
   }
-  auto dq = at::empty({B, H, S, D}, q.options());
-  auto dk = at::empty({B, H, S, D}, q.options());
-  auto dv = at::empty({B, H, S, D}, q.options());
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid(S / (2 * QB), B * H), block(512);
   size_t lds = (2 * KB * (size_t)D + (size_t)KB * D + 8 * 16 * KB) * 2;
@@ -768,19 +769,54 @@ flash_attn_bwd(const at::Tensor& grad_, const at::Tensor& q_,
       (const bf16*)grad.data_ptr(), (const bf16*)q.data_ptr(),
       (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
       lse.data_ptr<float>(), delta.data_ptr<float>(),
-      (bf16*)dq.data_ptr(), B, H, S, causal, scale,
+      dq_p, B, H, S, causal, scale,
       grad.stride(0), grad.stride(1), grad.stride(2),
       q.stride(0), q.stride(1), q.stride(2),
       k.stride(0), k.stride(1), k.stride(2),
-      v.stride(0), v.stride(1), v.stride(2));
+      v.stride(0), v.stride(1), v.stride(2), osb, osh, oss);
   hipLaunchKernelGGL(kkern, grid, block, lds_kv, stream,
       (const bf16*)grad.data_ptr(), (const bf16*)q.data_ptr(),
       (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
       lse.data_ptr<float>(), delta.data_ptr<float>(),
-      (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(), B, H, S, causal, scale,
+      dk_p, dv_p, B, H, S, causal, scale,
       grad.stride(0), grad.stride(1), grad.stride(2),
       q.stride(0), q.stride(1), q.stride(2),
       k.stride(0), k.stride(1), k.stride(2),
-      v.stride(0), v.stride(1), v.stride(2));
+      v.stride(0), v.stride(1), v.stride(2), osb, osh, oss);
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor>
+flash_attn_bwd(const at::Tensor& grad_, const at::Tensor& q_,
+               const at::Tensor& k_, const at::Tensor& v_,
+               const at::Tensor& out, const at::Tensor& lse, bool causal) {
+  const int B = q_.size(0), H = q_.size(1), S = q_.size(2), D = q_.size(3);
+  auto opts = q_.options();
+  auto dq = at::empty({B, H, S, D}, opts);
+  auto dk = at::empty({B, H, S, D}, opts);
+  auto dv = at::empty({B, H, S, D}, opts);
+  flash_attn_bwd_launch(grad_, q_, k_, v_, out, lse, causal,
+                        (bf16*)dq.data_ptr(), (bf16*)dk.data_ptr(),
+                        (bf16*)dv.data_ptr(),
+                        (long)H * S * D, (long)S * D, D);
   return {dq, dk, dv};
+}
+
+// dq/dk/dv written straight into one [B, S, 3*H*D] buffer (the layout the
+// qkv-projection backward consumes) — replaces the reference-model
+// transpose+clone+cat chain (three strided gathers plus a cat copy,
+// ~1.2 GB of pure layout traffic per GPT layer at batch 64).
+at::Tensor
+flash_attn_bwd_pack(const at::Tensor& grad_, const at::Tensor& q_,
+                    const at::Tensor& k_, const at::Tensor& v_,
+                    const at::Tensor& out, const at::Tensor& lse,
+                    bool causal) {
+  const long B = q_.size(0), H = q_.size(1), S = q_.size(2), D = q_.size(3);
+  auto dqkv = at::empty({B, S, 3 * H * D}, q_.options());
+  bf16* base = (bf16*)dqkv.data_ptr();
+  // dq/dk/dv sections at column offsets 0, H*D, 2*H*D of the packed rows;
+  // per-(b,h,s,d) element strides: osb=S*3HD, osh=D, oss=3HD
+  flash_attn_bwd_launch(grad_, q_, k_, v_, out, lse, causal,
+                        base, base + H * D, base + 2 * H * D,
+                        S * 3 * H * D, D, 3 * H * D);
+  return dqkv;
 }

@@ -39,6 +39,9 @@ at::Tensor gemm_nt_act(const at::Tensor& a, const at::Tensor& bt,
                        const std::optional<at::Tensor>& aux);
 std::tuple<at::Tensor, at::Tensor> gemm_tn_asum(const at::Tensor& a,
                                                 const at::Tensor& b);
+std::tuple<at::Tensor, at::Tensor>
+gemm_nt_gelu(const at::Tensor& a, const at::Tensor& bt,
+             const std::optional<at::Tensor>& bias, bool tanh_approx);
 std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
                                                   const at::Tensor& k,
                                                   const at::Tensor& v,
@@ -47,6 +50,11 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor>
 flash_attn_bwd(const at::Tensor& grad, const at::Tensor& q,
                const at::Tensor& k, const at::Tensor& v,
                const at::Tensor& out, const at::Tensor& lse, bool causal);
+at::Tensor
+flash_attn_bwd_pack(const at::Tensor& grad, const at::Tensor& q,
+                    const at::Tensor& k, const at::Tensor& v,
+                    const at::Tensor& out, const at::Tensor& lse,
+                    bool causal);
 
 // wrappers adapting std::vector<bool> / TensorList signatures
 static std::tuple<at::Tensor, at::Tensor, at::Tensor>
@@ -100,10 +108,14 @@ TORCH_LIBRARY(easydist_amd_hip, m) {
   m.def("gemm_nt_act(Tensor a, Tensor bt, Tensor? bias, int act, "
         "Tensor? aux) -> Tensor");
   m.def("gemm_tn_asum(Tensor a, Tensor b) -> (Tensor, Tensor)");
+  m.def("gemm_nt_gelu(Tensor a, Tensor bt, Tensor? bias, bool tanh_approx) "
+        "-> (Tensor, Tensor)");
   m.def("flash_attn_fwd(Tensor q, Tensor k, Tensor v, bool causal) "
         "-> (Tensor, Tensor)");
   m.def("flash_attn_bwd(Tensor grad, Tensor q, Tensor k, Tensor v, "
         "Tensor out, Tensor lse, bool causal) -> (Tensor, Tensor, Tensor)");
+  m.def("flash_attn_bwd_pack(Tensor grad, Tensor q, Tensor k, Tensor v, "
+        "Tensor out, Tensor lse, bool causal) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
@@ -120,8 +132,10 @@ TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
   m.impl("gemm_tn", gemm_tn);
   m.impl("gemm_nt_act", gemm_nt_act);
   m.impl("gemm_tn_asum", gemm_tn_asum);
+  m.impl("gemm_nt_gelu", gemm_nt_gelu);
   m.impl("flash_attn_fwd", flash_attn_fwd);
   m.impl("flash_attn_bwd", flash_attn_bwd);
+  m.impl("flash_attn_bwd_pack", flash_attn_bwd_pack);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {}

@@ -241,3 +241,46 @@ def _nt_act_fake(a, bt, bias, act, aux):
 def _tn_asum_fake(a, b):
     return (a.new_empty((a.shape[1], b.shape[1])),
             a.new_empty((a.shape[1],), dtype=torch.float32))
+
+
+# fused forward linear+gelu: one kernel writes BOTH gelu(out) and the
+# pre-activation (kept for gelu_backward), saving the standalone
+# activation kernel's HBM round-trip (lower_hip.py:lower_gelu_fwd_fuse)
+lib.define("gemm_nt_gelu(Tensor a, Tensor bt, Tensor? bias, bool tanh_approx)"
+           " -> (Tensor, Tensor)")
+
+
+def _nt_gelu_cpu(a, bt, bias, tanh_approx):
+    pre = _nt_cpu(a, bt, bias)
+    out = torch.nn.functional.gelu(
+        pre, approximate="tanh" if tanh_approx else "none")
+    return out, pre
+
+
+def _nt_gelu_cuda(a, bt, bias, tanh_approx):
+    ext = load_extension()
+    if ext is None or not _nt_supported(a, bt) \
+            or not mdconfig.use_hip_kernels:
+        if ext is None and mdconfig.use_hip_kernels:
+            from . import require_hip_ops
+            require_hip_ops()
+        return _nt_gelu_cpu(a, bt, bias, tanh_approx)
+    a = a.contiguous()
+    bt = bt.contiguous()
+    b_c = bias.contiguous() if bias is not None else None
+    key = ("nt_gelu", a.shape[0], a.shape[1], bt.shape[0], bias is not None,
+           tanh_approx)
+    if _choose(key, lambda: ext.gemm_nt_gelu(a, bt, b_c, tanh_approx),
+               lambda: _nt_gelu_cpu(a, bt, bias, tanh_approx)):
+        return ext.gemm_nt_gelu(a, bt, b_c, tanh_approx)
+    return _nt_gelu_cpu(a, bt, bias, tanh_approx)
+
+
+lib.impl("gemm_nt_gelu", _nt_gelu_cpu, "CPU")
+lib.impl("gemm_nt_gelu", _nt_gelu_cuda, "CUDA")
+
+
+@torch.library.register_fake("easydist_amd::gemm_nt_gelu")
+def _nt_gelu_fake(a, bt, bias, tanh_approx):
+    return (a.new_empty((a.shape[0], bt.shape[0])),
+            a.new_empty((a.shape[0], bt.shape[0])))

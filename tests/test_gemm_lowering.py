@@ -62,7 +62,8 @@ def test_lower_gemm_node_mix_and_golden():
     names = [getattr(n.target, "__name__", "") for n in gm.graph.nodes
              if n.op == "call_function"]
     n_nt = sum(1 for s in names
-               if s in ("gemm_nt.default", "gemm_nt_act.default"))
+               if s in ("gemm_nt.default", "gemm_nt_act.default",
+                        "gemm_nt_gelu.default"))
     n_tn = sum(1 for s in names
                if s in ("gemm_tn.default", "gemm_tn_asum.default"))
     # 2 fwd NT + 2 dX NT (weight-transposed) and 2 dW TN
@@ -71,6 +72,9 @@ def test_lower_gemm_node_mix_and_golden():
     # the gelu backward is fused into the dX GEMM epilogue
     assert "gemm_nt_act.default" in names, names
     assert "gelu_backward.default" not in names, names
+    # ... and the forward gelu into the fwd GEMM epilogue
+    assert "gemm_nt_gelu.default" in names, names
+    assert "gelu.default" not in names, names
     # bias grads fused into the dW TN GEMMs
     assert "gemm_tn_asum.default" in names, names
     # no stray aten mm of the linear shapes left
