@@ -81,14 +81,17 @@ def _compile_auto(func, tracing_mode, args, kwargs, module, opt):
                                    lower_cross_entropy, lower_gelu_bwd_fuse,
                                    lower_gelu_fwd_fuse, lower_gemm,
                                    lower_layer_norm, lower_sdpa)
+    import os as _os
     lower_layer_norm(gm)
     lower_sdpa(gm)
-    lower_attn_pack(gm)
+    if _os.environ.get("EASYDIST_ATTN_PACK", "1") != "0":
+        lower_attn_pack(gm)
     lower_cross_entropy(gm)
     if mdconfig.hip_gemm:
         lower_gemm(gm)
         lower_gelu_bwd_fuse(gm)
-        lower_gelu_fwd_fuse(gm)
+        if _os.environ.get("EASYDIST_GELU_FWD_FUSE", "1") != "0":
+            lower_gelu_fwd_fuse(gm)
         lower_bias_grad_fuse(gm)
 
     # ---- 5c. re-fuse the decomposed Adam chains into ONE kernel ----------

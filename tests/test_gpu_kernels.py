@@ -232,3 +232,41 @@ def test_gemm_tn_asymmetric(ext):
     c = ext.gemm_tn(a, b)
     assert float(c[37, 85]) == pytest.approx(6.0, abs=1e-2)
     assert float(c.float().abs().sum()) == pytest.approx(6.0, abs=1e-2)
+
+
+@requires_gpu
+def test_flash_attn_bwd_pack(ext):
+    # packed dqkv must equal the separate outputs repacked to [B,S,3HD]
+    torch.manual_seed(7)
+    for (B, H, S, D) in [(2, 3, 128, 64), (1, 2, 256, 128)]:
+        q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        g = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        out, lse = ext.flash_attn_fwd(q, k, v, True)
+        dq, dk, dv = ext.flash_attn_bwd(g, q, k, v, out, lse, True)
+        packed = ext.flash_attn_bwd_pack(g, q, k, v, out, lse, True)
+        ref = torch.cat([d.transpose(1, 2).reshape(B, S, H * D)
+                         for d in (dq, dk, dv)], -1)
+        assert packed.shape == (B, S, 3 * H * D)
+        assert torch.equal(packed, ref), \
+            float((packed.float() - ref.float()).abs().max())
+
+
+@requires_gpu
+def test_gemm_nt_gelu(ext):
+    # fused fwd gelu: act output AND pre-activation in one kernel
+    torch.manual_seed(8)
+    for (M, N, K), tanh in [((512, 384, 128), True), ((4096, 256, 64), False),
+                            ((256, 128, 96), True)]:
+        a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        bt = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+        bias = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+        act, pre = ext.gemm_nt_gelu(a, bt, bias, tanh)
+        ref_pre = (a.float() @ bt.float().t() + bias.float())
+        err_pre = (pre.float() - ref_pre).abs().max()
+        assert err_pre < 0.5, float(err_pre)
+        ref_act = torch.nn.functional.gelu(
+            pre.float(), approximate="tanh" if tanh else "none")
+        err = (act.float() - ref_act).abs().max()
+        assert err < 3e-2, (float(err), M, N, K, tanh)
