@@ -45,10 +45,11 @@ def _policy() -> str:
 
 # ---------------------------------------------------------------- CPU -------
 def _nt_cpu(a, bt, bias):
-    out = a @ bt.t()
+    # addmm (not mm+add): hipBLASLt fuses the bias into its epilogue —
+    # the separate broadcast add costs an output-sized HBM round-trip
     if bias is not None:
-        out = out + bias
-    return out
+        return torch.addmm(bias, a, bt.t())
+    return a @ bt.t()
 
 
 def _tn_cpu(a, b):
