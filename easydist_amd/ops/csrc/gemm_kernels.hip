@@ -618,6 +618,139 @@ gemm_tn_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
 }
 
 // ===========================================================================
+// gemm_tn_256: 256x256 output tile, 512 threads (8 waves of 64x128).
+// Same tr-read image scheme as gemm_tn_kernel but with DOUBLE the tile
+// edge: operand re-read traffic scales with 1/tile_edge (A is re-read
+// Q/tile times, B P/tile times), and the 128x128 kernel measures exactly
+// at that memory floor (3.6 GB -> 0.57 ms on the 65536-row dW shapes).
+// The square 256 tile halves it. Occupancy is VGPR-bound at 2 waves/SIMD
+// (acc[4][8] = 128 VGPRs) -- acceptable: the loop is HBM-bound, the
+// double-buffered glds prefetch covers what latency can be covered.
+// ===========================================================================
+#define TBP2 256
+
+// one 16 KiB tr-image stage with 512 threads: tile [32 r][256 p]
+DEVINL void tn_stage2(const bf16* src, long ld, int r0, int p0, int pmax,
+                      char* dst_base, int t, int wave) {
+  #pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    unsigned p = (i * 512 + t) * 16;          // linear byte pos (16 KiB)
+    int chunk16 = p >> 10;                    // 1 KiB per 16-p chunk
+    unsigned w = p & 1023u;
+    int bpos = w >> 7;                        // 128 B per [4r][16p] block
+    int blk = tn_unswz(bpos);
+    int rr = blk * 4 + ((w & 127) >> 5);
+    int pp = chunk16 * 16 + ((w & 31) >> 1);
+    int gp = min(p0 + pp, pmax - 1);
+    glds16(src + (long)(r0 + rr) * ld + gp,
+           dst_base + (i * 512 + wave * 64) * 16);
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_tn_256_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
+                   float* __restrict__ Cw, float* __restrict__ Asum,
+                   int R, int P, int Q, int splitr) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const unsigned IMG = TBR * TBP2 * 2;         // 16 KiB per operand image
+  const unsigned BUF = 2 * IMG;                // A+B per buffer
+
+  unsigned nwg_p = P / TBP2, nwg_q = Q / TBP2;
+  unsigned ntile = nwg_p * nwg_q;
+  unsigned wg = xcd_swizzle(blockIdx.x, ntile * splitr);
+  unsigned tile = wg / splitr, slice = wg % splitr;
+  const unsigned p0 = (tile / nwg_q) * TBP2;
+  const unsigned q0 = (tile % nwg_q) * TBP2;
+
+  int nrt_all = R / TBR;
+  int per = (nrt_all + splitr - 1) / splitr;
+  int rt0 = slice * per;
+  int rt1 = min(nrt_all, rt0 + per);
+  if (rt0 >= rt1) return;
+
+  const int t = threadIdx.x;
+  const int lane = t % WAVE;
+  const int wave = t / WAVE;
+  const int wp = (wave >> 1) * 64;             // 4 p-rows of waves
+  const int wq = (wave & 1) * 128;             // 2 q-cols of waves
+
+  auto stage = [&](int buf, int rt) {
+    tn_stage2(Ag, P, rt * TBR, p0, P, smem + buf * BUF, t, wave);
+    tn_stage2(Bg, Q, rt * TBR, q0, Q, smem + buf * BUF + IMG, t, wave);
+  };
+
+  f32x4 acc[4][8];
+  float asum_acc[4] = {0.f, 0.f, 0.f, 0.f};
+  #pragma unroll
+  for (int i = 0; i < 4; ++i)
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage(0, rt0);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+  for (int rt = rt0; rt < rt1; ++rt) {
+    int buf = (rt - rt0) & 1;
+    if (rt + 1 < rt1) stage(buf ^ 1, rt + 1);
+    bf16x8v a_frag[4], b_frag[8];
+    #pragma unroll
+    for (int i = 0; i < 4; ++i)
+      a_frag[i] = tn_frag(smem + buf * BUF, (wp >> 4) + i, lane);
+    if (Asum && q0 == 0 && (wave & 1) == 0) {
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        float acc_s = 0.f;
+        #pragma unroll
+        for (int u = 0; u < 8; ++u) acc_s += bf2f((bf16)a_frag[i][u]);
+        asum_acc[i] += acc_s;
+      }
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      b_frag[j] = tn_frag(smem + buf * BUF + IMG, (wq >> 4) + j, lane);
+    #pragma unroll
+    for (int i = 0; i < 4; ++i)
+      #pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+  }
+
+  if (Asum && q0 == 0 && (wave & 1) == 0) {
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      float v = asum_acc[i];
+      v += __shfl_down(v, 32, WAVE);
+      v += __shfl_down(v, 16, WAVE);
+      if (lane < 16) {
+        int gp = p0 + wp + i * 16 + lane;
+        if (gp < P) atomicAdd(&Asum[gp], v);
+      }
+    }
+  }
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+  #pragma unroll
+  for (int i = 0; i < 4; ++i)
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int gp = p0 + wp + i * 16 + c_row0 + reg;
+        int gq = q0 + wq + j * 16 + c_col;
+        if (gp < P && gq < Q) {
+          if (splitr == 1)
+            Cw[(long)gp * Q + gq] = acc[i][j][reg];
+          else
+            atomicAdd(&Cw[(long)gp * Q + gq], acc[i][j][reg]);
+        }
+      }
+}
+
+// ===========================================================================
 // host wrappers
 // ===========================================================================
 at::Tensor gemm_nt_act(const at::Tensor& a, const at::Tensor& bt,
@@ -691,8 +824,11 @@ gemm_nt_gelu(const at::Tensor& a, const at::Tensor& bt,
 std::tuple<at::Tensor, at::Tensor> gemm_tn_asum(const at::Tensor& a,
                                                 const at::Tensor& b);
 
-at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
-  // C[P,Q] = a^T @ b with a:[R,P], b:[R,Q]
+// shared TN launch: picks the 256x256 tile (half the operand re-read
+// traffic) when P and Q allow it AND the grid still covers the chip;
+// falls back to the 128x128 kernel otherwise (edge + small shapes).
+static at::Tensor tn_launch(const at::Tensor& a, const at::Tensor& b,
+                            float* asum_p) {
   TORCH_CHECK(a.dtype() == at::kBFloat16 && b.dtype() == at::kBFloat16);
   TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
   const long R = a.size(0), P = a.size(1), Q = b.size(0) == R ? b.size(1) : -1;
@@ -700,46 +836,56 @@ at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
   TORCH_CHECK(R % TBR == 0, "gemm_tn: R must be a multiple of 32");
   TORCH_CHECK(P % TBP == 0 && Q % TBP == 0,
               "gemm_tn: P, Q must be multiples of 128");
-  unsigned ntile = (P / TBP) * (Q / TBP);
-  // split the reduction so the grid covers the chip's 256 CUs
+  static int use256 = []() {
+    const char* e = getenv("EASYDIST_TN256");
+    return e ? atoi(e) : 1;
+  }();
+  bool big = use256 && P % TBP2 == 0 && Q % TBP2 == 0 && R / TBR >= 64;
+  unsigned ntile;
   int splitr = 1;
-  while (ntile * splitr < 512 && splitr < 16 &&
-         (R / TBR) % (splitr * 2) == 0 && (R / TBR) / (splitr * 2) >= 1)
-    splitr *= 2;
+  if (big) {
+    ntile = (P / TBP2) * (Q / TBP2);
+    // 8-wave WGs run ~1/CU: target >=256 WGs; a deep split is fine, the
+    // fp32 atomic pass is tiny against the staged reads
+    while (ntile * splitr < 256 && splitr < 64 &&
+           (R / TBR) % (splitr * 2) == 0 && (R / TBR) / (splitr * 2) >= 2)
+      splitr *= 2;
+  } else {
+    ntile = (P / TBP) * (Q / TBP);
+    // split the reduction so the grid covers the chip's 256 CUs
+    while (ntile * splitr < 512 && splitr < 16 &&
+           (R / TBR) % (splitr * 2) == 0 && (R / TBR) / (splitr * 2) >= 1)
+      splitr *= 2;
+  }
   auto cw = splitr == 1
       ? at::empty({P, Q}, a.options().dtype(at::kFloat))
       : at::zeros({P, Q}, a.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
-  size_t lds = 2 * 2 * TBR * TBP * 2;   // 32 KiB
-  hipLaunchKernelGGL(gemm_tn_kernel, dim3(ntile * splitr), dim3(256), lds,
-      stream,
-      (const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
-      cw.data_ptr<float>(), nullptr, (int)R, (int)P, (int)Q, splitr);
-  return cw.to(at::kBFloat16);
+  if (big) {
+    size_t lds = 2 * 2 * TBR * TBP2 * 2;   // 64 KiB
+    hipLaunchKernelGGL(gemm_tn_256_kernel, dim3(ntile * splitr), dim3(512),
+        lds, stream,
+        (const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+        cw.data_ptr<float>(), asum_p, (int)R, (int)P, (int)Q, splitr);
+  } else {
+    size_t lds = 2 * 2 * TBR * TBP * 2;   // 32 KiB
+    hipLaunchKernelGGL(gemm_tn_kernel, dim3(ntile * splitr), dim3(256), lds,
+        stream,
+        (const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+        cw.data_ptr<float>(), asum_p, (int)R, (int)P, (int)Q, splitr);
+  }
+  return cw;
+}
+
+at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
+  // C[P,Q] = a^T @ b with a:[R,P], b:[R,Q]
+  return tn_launch(a, b, nullptr).to(at::kBFloat16);
 }
 
 std::tuple<at::Tensor, at::Tensor> gemm_tn_asum(const at::Tensor& a,
                                                 const at::Tensor& b) {
   // like gemm_tn, additionally returning colsum(a) (dBias) in ONE pass
-  TORCH_CHECK(a.dtype() == at::kBFloat16 && b.dtype() == at::kBFloat16);
-  TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
-  const long R = a.size(0), P = a.size(1), Q = b.size(0) == R ? b.size(1) : -1;
-  TORCH_CHECK(Q > 0 && R % TBR == 0 && P % TBP == 0 && Q % TBP == 0);
-  unsigned ntile = (P / TBP) * (Q / TBP);
-  int splitr = 1;
-  while (ntile * splitr < 512 && splitr < 16 &&
-         (R / TBR) % (splitr * 2) == 0 && (R / TBR) / (splitr * 2) >= 1)
-    splitr *= 2;
-  auto cw = splitr == 1
-      ? at::empty({P, Q}, a.options().dtype(at::kFloat))
-      : at::zeros({P, Q}, a.options().dtype(at::kFloat));
-  auto asum = at::zeros({P}, a.options().dtype(at::kFloat));
-  auto stream = at::cuda::getCurrentCUDAStream();
-  size_t lds = 2 * 2 * TBR * TBP * 2;
-  hipLaunchKernelGGL(gemm_tn_kernel, dim3(ntile * splitr), dim3(256), lds,
-      stream,
-      (const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
-      cw.data_ptr<float>(), asum.data_ptr<float>(), (int)R, (int)P, (int)Q,
-      splitr);
+  auto asum = at::zeros({a.size(1)}, a.options().dtype(at::kFloat));
+  auto cw = tn_launch(a, b, asum.data_ptr<float>());
   return {cw.to(at::kBFloat16), asum};
 }
