@@ -845,11 +845,17 @@ static at::Tensor tn_launch(const at::Tensor& a, const at::Tensor& b,
   int splitr = 1;
   if (big) {
     ntile = (P / TBP2) * (Q / TBP2);
-    // 8-wave WGs run ~1/CU: target >=256 WGs; a deep split is fine, the
-    // fp32 atomic pass is tiny against the staged reads
-    while (ntile * splitr < 256 && splitr < 64 &&
-           (R / TBR) % (splitr * 2) == 0 && (R / TBR) / (splitr * 2) >= 2)
-      splitr *= 2;
+    // 8-wave WGs are VGPR-bound at ONE resident WG per CU, so the grid
+    // should fill the 256 CUs in (near-)whole rounds: a ragged 1.1-round
+    // grid serializes a mostly-idle second round. Flat division (not
+    // powers of two) lands just under the target; slices ceil-div R.
+    static int target = []() {
+      const char* e = getenv("EASYDIST_TN256_TARGET");
+      return e ? atoi(e) : 256;
+    }();
+    splitr = std::max(1, target / (int)ntile);
+    splitr = std::min((long)splitr, R / TBR / 8);   // >=8 k-steps per slice
+    splitr = std::max(1, std::min(splitr, 64));
   } else {
     ntile = (P / TBP) * (Q / TBP);
     // split the reduction so the grid covers the chip's 256 CUs
