@@ -10,6 +10,8 @@
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include <ATen/cuda/CUDAGraphsUtils.cuh>
+#include <map>
+#include <utility>
 #include <vector>
 
 struct AdamChunk {
@@ -37,6 +39,66 @@ DEVINL void load_g4(const void* g, int i, int g_bf16, float* out) {
 }
 
 #define CHUNK_ELEMS 65536
+
+// ---------------------------------------------------------------------------
+// capture-safe chunk-table staging, keyed by (table bytes, first param
+// pointer): the optimizer fuse pass emits SEPARATE fused steps per grad
+// dtype bank, so ONE static pair is not enough — during hipGraph capture
+// the first call retires its recorded buffers and the second call must
+// find its OWN warm pair (sized by the eager warmup, same key).
+// ---------------------------------------------------------------------------
+struct StagingPair {
+  at::Tensor pinned, dev;
+  size_t capacity = 0;
+  hipEvent_t h2d_done = nullptr;
+};
+
+static StagingPair& staging_for(std::map<std::pair<size_t, const void*>,
+                                         StagingPair>& pool,
+                                size_t bytes, const void* key_ptr,
+                                const at::Device& dev, bool capturing,
+                                const char* who) {
+  auto& st = pool[{bytes, key_ptr}];
+  if (bytes > st.capacity) {
+    TORCH_CHECK(!capturing, who,
+                ": staging buffers must be warmed up (one eager call per "
+                "parameter bank) before hipGraph capture");
+    st.pinned = at::empty({(long)bytes},
+                          at::TensorOptions().dtype(at::kByte)
+                              .pinned_memory(true));
+    st.dev = at::empty({(long)bytes},
+                       at::TensorOptions().dtype(at::kByte).device(dev));
+    st.capacity = bytes;
+  }
+  if (st.h2d_done == nullptr)
+    C10_CUDA_CHECK(hipEventCreateWithFlags(&st.h2d_done,
+                                           hipEventDisableTiming));
+  else if (!capturing)
+    // the PREVIOUS call's async H2D must have consumed the pinned buffer
+    // before the host overwrites it. hipEventSynchronize is REJECTED
+    // inside capture (the runtime device-synchronizes right before
+    // capture begins, which retires any in-flight eager H2D).
+    C10_CUDA_CHECK(hipEventSynchronize(st.h2d_done));
+  return st;
+}
+
+static void staging_finish(std::map<std::pair<size_t, const void*>,
+                                    StagingPair>& pool,
+                           size_t bytes, const void* key_ptr,
+                           bool capturing, hipStream_t stream) {
+  auto it = pool.find({bytes, key_ptr});
+  if (it == pool.end()) return;
+  if (capturing) {
+    // the recorded H2D re-reads THIS pinned buffer on every replay:
+    // retire the pair so no later call can overwrite it
+    static std::vector<at::Tensor> keepalive;
+    keepalive.push_back(it->second.pinned);
+    keepalive.push_back(it->second.dev);
+    pool.erase(it);
+  } else {
+    C10_CUDA_CHECK(hipEventRecord(it->second.h2d_done, stream));
+  }
+}
 
 __global__ void fused_adam_kernel(const AdamChunk* __restrict__ chunks,
                                   int n_chunks, int g_bf16,
@@ -144,60 +206,25 @@ fused_adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
       chunks.push_back(c);
     }
   }
-  // capture-safe table upload: PERSISTENT pinned staging + device table,
-  // (re)allocated only OUTSIDE capture (the eager warmup sizes them);
-  // inside capture we only host-write the pinned buffer (recorded values
-  // stay valid: the graph pool re-serves the same output pointers on
-  // every replay) and record one async H2D memcpy node.
-  static at::Tensor g_pinned, g_dev;
-  static size_t g_capacity = 0;
-  static hipEvent_t g_h2d_done = nullptr;
+  // capture-safe table upload through the keyed staging pool (one pair
+  // per parameter bank — the fuse pass emits a bf16-grad and an
+  // fp32-grad fused step per optimizer).
+  static std::map<std::pair<size_t, const void*>, StagingPair> g_pool;
   const size_t bytes = chunks.size() * sizeof(AdamChunk);
+  const void* key = params[0].data_ptr();
   bool capturing = at::cuda::currentStreamCaptureStatusMayInitCtx() !=
                    at::cuda::CaptureStatus::None;
-  if (bytes > g_capacity) {
-    TORCH_CHECK(!capturing,
-                "fused_adam_step: staging buffers must be warmed up (one "
-                "eager call) before hipGraph capture");
-    g_pinned = at::empty({(long)bytes},
-                         at::TensorOptions().dtype(at::kByte)
-                             .pinned_memory(true));
-    g_dev = at::empty({(long)bytes},
-                      at::TensorOptions().dtype(at::kByte)
-                          .device(params[0].device()));
-    g_capacity = bytes;
-  }
-  // the PREVIOUS call's async H2D must have consumed the pinned buffer
-  // before the host overwrites it (race -> garbage chunk table -> NaNs).
-  // hipEventSynchronize is REJECTED inside capture on ROCm
-  // (hipErrorStreamCaptureUnsupported) — the runtime device-synchronizes
-  // right before capture begins (compiled_func.run_graph), which retires
-  // any in-flight eager H2D, so the capture path may skip the wait.
-  if (g_h2d_done == nullptr)
-    C10_CUDA_CHECK(hipEventCreateWithFlags(&g_h2d_done,
-                                           hipEventDisableTiming));
-  else if (!capturing)
-    C10_CUDA_CHECK(hipEventSynchronize(g_h2d_done));
-  memcpy(g_pinned.data_ptr(), chunks.data(), bytes);
+  auto& st = staging_for(g_pool, bytes, key, params[0].device(), capturing,
+                         "fused_adam_step");
+  memcpy(st.pinned.data_ptr(), chunks.data(), bytes);
   auto stream = at::cuda::getCurrentCUDAStream();
-  C10_CUDA_CHECK(hipMemcpyAsync(g_dev.data_ptr(), g_pinned.data_ptr(),
+  C10_CUDA_CHECK(hipMemcpyAsync(st.dev.data_ptr(), st.pinned.data_ptr(),
                                 bytes, hipMemcpyHostToDevice, stream));
-  if (!capturing)
-    C10_CUDA_CHECK(hipEventRecord(g_h2d_done, stream));
   hipLaunchKernelGGL(fused_adam_kernel, dim3(chunks.size()), dim3(256), 0,
-      stream, (const AdamChunk*)g_dev.data_ptr(), (int)chunks.size(),
+      stream, (const AdamChunk*)st.dev.data_ptr(), (int)chunks.size(),
       (int)g_bf16, steps_flat.data_ptr<float>(), (float)lr, (float)beta1,
       (float)beta2, (float)weight_decay, (float)eps);
-  if (capturing) {
-    // the recorded H2D re-reads THIS pinned buffer on every replay:
-    // retire the pair so no later call can overwrite it
-    static std::vector<at::Tensor> keepalive;
-    keepalive.push_back(g_pinned);
-    keepalive.push_back(g_dev);
-    g_pinned = at::Tensor();
-    g_dev = at::Tensor();
-    g_capacity = 0;
-  }
+  staging_finish(g_pool, bytes, key, capturing, stream);
   // bump steps on host-free path: out_step = step + 1
   for (int t = 0; t < T; ++t)
     out_steps[t] = steps[t] + 1;
@@ -274,45 +301,21 @@ fused_sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
       chunks.push_back(c);
     }
   }
-  static at::Tensor g_pinned, g_dev;
-  static size_t g_capacity = 0;
-  static hipEvent_t g_h2d_done = nullptr;
+  static std::map<std::pair<size_t, const void*>, StagingPair> g_pool;
   const size_t bytes = chunks.size() * sizeof(SgdChunk);
+  const void* key = params[0].data_ptr();
   bool capturing = at::cuda::currentStreamCaptureStatusMayInitCtx() !=
                    at::cuda::CaptureStatus::None;
-  if (bytes > g_capacity) {
-    TORCH_CHECK(!capturing,
-                "fused_sgd_step: warm up once eagerly before capture");
-    g_pinned = at::empty({(long)bytes},
-                         at::TensorOptions().dtype(at::kByte)
-                             .pinned_memory(true));
-    g_dev = at::empty({(long)bytes},
-                      at::TensorOptions().dtype(at::kByte)
-                          .device(params[0].device()));
-    g_capacity = bytes;
-  }
-  if (g_h2d_done == nullptr)
-    C10_CUDA_CHECK(hipEventCreateWithFlags(&g_h2d_done,
-                                           hipEventDisableTiming));
-  else if (!capturing)
-    C10_CUDA_CHECK(hipEventSynchronize(g_h2d_done));
-  memcpy(g_pinned.data_ptr(), chunks.data(), bytes);
+  auto& st = staging_for(g_pool, bytes, key, params[0].device(), capturing,
+                         "fused_sgd_step");
+  memcpy(st.pinned.data_ptr(), chunks.data(), bytes);
   auto stream = at::cuda::getCurrentCUDAStream();
-  C10_CUDA_CHECK(hipMemcpyAsync(g_dev.data_ptr(), g_pinned.data_ptr(),
+  C10_CUDA_CHECK(hipMemcpyAsync(st.dev.data_ptr(), st.pinned.data_ptr(),
                                 bytes, hipMemcpyHostToDevice, stream));
-  if (!capturing)
-    C10_CUDA_CHECK(hipEventRecord(g_h2d_done, stream));
   hipLaunchKernelGGL(fused_sgd_kernel, dim3(chunks.size()), dim3(256), 0,
-      stream, (const SgdChunk*)g_dev.data_ptr(), (int)chunks.size(),
+      stream, (const SgdChunk*)st.dev.data_ptr(), (int)chunks.size(),
       (int)g_bf16, (float)lr, (float)momentum, (float)dampening,
       (float)weight_decay, nesterov ? 1 : 0);
-  if (capturing) {
-    static std::vector<at::Tensor> keepalive;
-    keepalive.push_back(g_pinned);
-    keepalive.push_back(g_dev);
-    g_pinned = at::Tensor();
-    g_dev = at::Tensor();
-    g_capacity = 0;
-  }
+  staging_finish(g_pool, bytes, key, capturing, stream);
   return {out_p, out_buf};
 }
