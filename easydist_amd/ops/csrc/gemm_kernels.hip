@@ -662,7 +662,9 @@ gemm_tn_256_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
   const unsigned IMG = TBR * TBP2 * 2;         // 16 KiB per operand image
   const unsigned BUF = 2 * IMG;                // A+B per buffer
 
-  unsigned nwg_p = P / TBP2, nwg_q = Q / TBP2;
+  // ceil-div: partial edge tiles are safe — staging clamps the source
+  // row to pmax-1 (duplicate loads) and the store loop guards gp/gq
+  unsigned nwg_p = (P + TBP2 - 1) / TBP2, nwg_q = (Q + TBP2 - 1) / TBP2;
   unsigned ntile = nwg_p * nwg_q;
   unsigned wg = xcd_swizzle(blockIdx.x, ntile * splitr);
   unsigned tile = wg / splitr, slice = wg % splitr;
@@ -847,11 +849,14 @@ static at::Tensor tn_launch(const at::Tensor& a, const at::Tensor& b,
     const char* e = getenv("EASYDIST_TN256");
     return e ? atoi(e) : 1;
   }();
-  bool big = use256 && P % TBP2 == 0 && Q % TBP2 == 0 && R / TBR >= 64;
+  // P/Q only need the 128 alignment the op already requires — the 256
+  // kernel ceil-divs its grid and guards edge tiles (lm_head dW has
+  // P = vocab = 50304 = 128*393)
+  bool big = use256 && R / TBR >= 64 && P >= TBP2 && Q >= TBP2;
   unsigned ntile;
   int splitr = 1;
   if (big) {
-    ntile = (P / TBP2) * (Q / TBP2);
+    ntile = ((P + TBP2 - 1) / TBP2) * ((Q + TBP2 - 1) / TBP2);
     // 8-wave WGs are VGPR-bound at ONE resident WG per CU, so the grid
     // should fill the 256 CUs in (near-)whole rounds: a ragged 1.1-round
     // grid serializes a mostly-idle second round. Flat division (not
