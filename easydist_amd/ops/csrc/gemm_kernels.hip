@@ -496,7 +496,9 @@ DEVINL void tn_stage(const bf16* src, long ld, int r0, int p0, int pmax,
     int blk = tn_unswz(bpos);
     int rr = blk * 4 + ((w & 127) >> 5);
     int pp = chunk16 * 16 + ((w & 31) >> 1);
-    int gp = min(p0 + pp, pmax - 1);
+    // clamp keeps the full 16-B read inside the operand (the last
+    // partial tile's rows would otherwise read past the buffer end)
+    int gp = min(p0 + pp, pmax - 8);
     glds16(src + (long)(r0 + rr) * ld + gp,
            dst_base + (i * 256 + wave * 64) * 16);
   }
@@ -648,7 +650,9 @@ DEVINL void tn_stage2(const bf16* src, long ld, int r0, int p0, int pmax,
     int blk = tn_unswz(bpos);
     int rr = blk * 4 + ((w & 127) >> 5);
     int pp = chunk16 * 16 + ((w & 31) >> 1);
-    int gp = min(p0 + pp, pmax - 1);
+    // clamp keeps the full 16-B read inside the operand (the last
+    // partial tile's rows would otherwise read past the buffer end)
+    int gp = min(p0 + pp, pmax - 8);
     glds16(src + (long)(r0 + rr) * ld + gp,
            dst_base + (i * 512 + wave * 64) * 16);
   }
