@@ -30,20 +30,22 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 // and is injective across the 16 rows of a fragment for both row widths.
 DEVINL int lsw(int e) { return e ^ (((e >> 7) & 15) << 3); }
 
-template <int D>
-__global__ void __launch_bounds__(512, (D == 64 ? 4 : 1))  // VGPR cap at D=64 only
+template <int D, int RF = 1>
+__global__ void
+__launch_bounds__(512, (D == 64 ? (RF == 1 ? 4 : 3) : 1))  // waves/SIMD floor
 flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                  const bf16* __restrict__ V, bf16* __restrict__ O,
                  float* __restrict__ LSE, int B, int H, int S, bool causal,
                  float scale,
                  long qsb, long qsh, long qss, long ksb, long ksh, long kss,
                  long vsb, long vsh, long vss) {
-  // grid: (ceil(S/QBLK), B*H); QBLK = 128 q rows per workgroup, 8 waves
-  // of ONE 16-row fragment each.  The round-1 4-wave/RF=2 form ran ONE
-  // wave per SIMD (110 KB LDS, 256 threads) — every softmax/LDS stall
+  // grid: (ceil(S/QBLK), B*H); QBLK = 128*RF q rows per workgroup, 8
+  // waves of RF 16-row fragments each.  The round-1 4-wave/256-thread
+  // form ran ONE wave per SIMD (110 KB LDS) — every softmax/LDS stall
   // fully exposed, ~100 TF effective.  8 waves + the smaller strip pool
-  // give 3 blocks/CU at D=64 (6 waves/SIMD).
-  constexpr int RF = 1;
+  // give 4 waves/SIMD at D=64/RF=1.  RF=2 doubles the q rows sharing
+  // each staged K/V tile (half the staging + barrier cost per flop) at
+  // one wave/SIMD lower occupancy — A/B'd on hardware.
   constexpr int QBLK = 8 * 16 * RF;
   const int qb0 = blockIdx.x * QBLK;
   const int bh = blockIdx.y;
@@ -148,13 +150,15 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     if (t + 1 < n_tiles)
       load_tile(t + 1);          // global loads overlap the MFMA loop
 
-    // a 128-row block spans 8 waves: tiles strictly above this wave's
+    // a 128*RF-row block spans 8 waves: tiles strictly above this wave's
     // causal diagonal are all-masked — skip the compute (NOT the
     // barriers: every wave still arrives at __syncthreads)
-    const bool wave_active = !(causal && kv0 > qr0[0] + 15);
+    const bool wave_active = !(causal && kv0 > qr0[RF - 1] + 15);
     if (wave_active) {
     #pragma unroll
     for (int rf = 0; rf < RF; ++rf) {
+      if (RF > 1 && causal && kv0 > qr0[rf] + 15)
+        continue;                  // this fragment's rows are all-masked
       // ---- S = Q K^T ----------------------------------------------------
       f32x4 s_acc[KB / 16];
       __builtin_amdgcn_s_setprio(1);
@@ -301,10 +305,17 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q_,
   auto out = at::empty({B, H, S, D}, q.options());
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
-  dim3 grid((S + 127) / 128, B * H), block(512);
+  static int rf = []() {
+    const char* e = getenv("EASYDIST_FWD_RF");
+    return e ? atoi(e) : 1;
+  }();
+  const int use_rf = (rf == 2 && S % 256 == 0) ? 2 : 1;
+  dim3 grid((S + 128 * use_rf - 1) / (128 * use_rf), B * H), block(512);
   size_t lds = (2 * ((size_t)KB * D + (size_t)KB * D) + 8 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
-  auto kern = (D == 64) ? flash_fwd_kernel<64> : flash_fwd_kernel<128>;
+  auto kern = (D == 64)
+      ? (use_rf == 2 ? flash_fwd_kernel<64, 2> : flash_fwd_kernel<64, 1>)
+      : (use_rf == 2 ? flash_fwd_kernel<128, 2> : flash_fwd_kernel<128, 1>);
   hipLaunchKernelGGL(kern, grid, block, lds, stream,
       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
       (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
