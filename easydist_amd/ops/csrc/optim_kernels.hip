@@ -113,6 +113,8 @@ __global__ void fused_adam_kernel(const AdamChunk* __restrict__ chunks,
   const float bc2 = 1.f - __powf(beta2, step);
   const float inv_bc1 = 1.f / bc1;
   const float rsqrt_bc2 = rsqrtf(bc2);
+  // unroll keeps several independent 16-B load groups in flight per lane
+  #pragma unroll 4
   for (int i = threadIdx.x * 4; i + 3 < ch.n; i += blockDim.x * 4) {
     float4 p = *reinterpret_cast<const float4*>(&ch.p[i]);
     float gg[4];
