@@ -42,6 +42,8 @@ std::tuple<at::Tensor, at::Tensor> gemm_tn_asum(const at::Tensor& a,
 std::tuple<at::Tensor, at::Tensor>
 gemm_nt_gelu(const at::Tensor& a, const at::Tensor& bt,
              const std::optional<at::Tensor>& bias, bool tanh_approx);
+at::Tensor gelu_fast(const at::Tensor& x);
+at::Tensor gelu_bwd_fast(const at::Tensor& g, const at::Tensor& x);
 std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
                                                   const at::Tensor& k,
                                                   const at::Tensor& v,
@@ -110,6 +112,8 @@ TORCH_LIBRARY(easydist_amd_hip, m) {
   m.def("gemm_tn_asum(Tensor a, Tensor b) -> (Tensor, Tensor)");
   m.def("gemm_nt_gelu(Tensor a, Tensor bt, Tensor? bias, bool tanh_approx) "
         "-> (Tensor, Tensor)");
+  m.def("gelu_fast(Tensor x) -> Tensor");
+  m.def("gelu_bwd_fast(Tensor g, Tensor x) -> Tensor");
   m.def("flash_attn_fwd(Tensor q, Tensor k, Tensor v, bool causal) "
         "-> (Tensor, Tensor)");
   m.def("flash_attn_bwd(Tensor grad, Tensor q, Tensor k, Tensor v, "
@@ -133,6 +137,8 @@ TORCH_LIBRARY_IMPL(easydist_amd_hip, CUDA, m) {
   m.impl("gemm_nt_act", gemm_nt_act);
   m.impl("gemm_tn_asum", gemm_tn_asum);
   m.impl("gemm_nt_gelu", gemm_nt_gelu);
+  m.impl("gelu_fast", gelu_fast);
+  m.impl("gelu_bwd_fast", gelu_bwd_fast);
   m.impl("flash_attn_fwd", flash_attn_fwd);
   m.impl("flash_attn_bwd", flash_attn_bwd);
   m.impl("flash_attn_bwd_pack", flash_attn_bwd_pack);

@@ -270,3 +270,17 @@ def test_gemm_nt_gelu(ext):
             pre.float(), approximate="tanh" if tanh else "none")
         err = (act.float() - ref_act).abs().max()
         assert err < 3e-2, (float(err), M, N, K, tanh)
+
+
+@requires_gpu
+def test_gelu_fast(ext):
+    torch.manual_seed(9)
+    x = torch.randn(512, 768, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(512, 768, device="cuda", dtype=torch.bfloat16)
+    y = ext.gelu_fast(x)
+    ref = torch.nn.functional.gelu(x.float(), approximate="tanh")
+    assert (y.float() - ref).abs().max() < 2e-2
+    dy = ext.gelu_bwd_fast(g, x)
+    refb = torch.ops.aten.gelu_backward(g.float(), x.float(),
+                                        approximate="tanh")
+    assert (dy.float() - refb).abs().max() < 2e-2
