@@ -31,7 +31,7 @@ def build_case(name: str, device):
         y = torch.randint(0, GPT_BENCH_1L.vocab_size, (4, 1024),
                           device=device)
 
-        def step(model, opt):
+        def step(model, opt, x, y):
             from easydist_amd.ops import ce
             logits = model(x)
             loss = ce.cross_entropy(logits.view(-1, logits.size(-1)),
@@ -40,20 +40,20 @@ def build_case(name: str, device):
             opt.step()
             opt.zero_grad(True)
             return loss
-        return model, step
+        return model, step, (x, y)
     if name == "wresnet":
         from easydist_amd.models.resnet import wresnet50
         model = wresnet50().to(device)
         x = torch.randn(128, 3, 224, 224, device=device)
         y = torch.randint(0, 1000, (128,), device=device)
 
-        def step(model, opt):
+        def step(model, opt, x, y):
             loss = torch.nn.functional.cross_entropy(model(x), y)
             loss.backward()
             opt.step()
             opt.zero_grad(True)
             return loss
-        return model, step
+        return model, step, (x, y)
     if name == "gat":
         from easydist_amd.models.gat import GAT
         model = GAT(in_dim=12288, hidden=512).to(device)
@@ -61,13 +61,13 @@ def build_case(name: str, device):
         adj = (torch.rand(4096, 4096, device=device) < 0.01).float()
         y = torch.randint(0, 64, (4096,), device=device)
 
-        def step(model, opt):
+        def step(model, opt, x, y, adj):
             loss = torch.nn.functional.cross_entropy(model(x, adj), y)
             loss.backward()
             opt.step()
             opt.zero_grad(True)
             return loss
-        return model, step
+        return model, step, (x, y, adj)
     raise ValueError(name)
 
 
@@ -95,7 +95,7 @@ def main():
         dist.init_process_group("nccl" if use_cuda else "gloo",
                                 rank=rank, world_size=world)
 
-    model, step = build_case(args.model, device)
+    model, step, inputs = build_case(args.model, device)
 
     if args.mode in ("auto", "zero2", "zero3"):
         from easydist_amd import easydist_compile, easydist_setup, \
@@ -105,22 +105,22 @@ def main():
         set_device_mesh(list(range(world)), ["spmd0"])
         opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
         mode = args.mode if args.mode != "auto" else "auto"
-        compiled = easydist_compile(
-            lambda m, o: step(m, o), parallel_mode=mode, cuda_graph=False)
-        run = lambda: compiled(model, opt)           # noqa: E731
+        compiled = easydist_compile(step, parallel_mode=mode,
+                                    cuda_graph=False)
+        run = lambda: compiled(model, opt, *inputs)  # noqa: E731
     elif args.mode == "ddp":
         m = torch.nn.parallel.DistributedDataParallel(
             model, device_ids=[local] if use_cuda else None)
         opt = torch.optim.Adam(m.parameters(), lr=1e-4, fused=use_cuda)
-        run = lambda: step(m, opt)                   # noqa: E731
+        run = lambda: step(m, opt, *inputs)          # noqa: E731
     elif args.mode == "fsdp":
         from torch.distributed.fsdp import FullyShardedDataParallel
         m = FullyShardedDataParallel(model)
         opt = torch.optim.Adam(m.parameters(), lr=1e-4)
-        run = lambda: step(m, opt)                   # noqa: E731
+        run = lambda: step(m, opt, *inputs)          # noqa: E731
     else:
         opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
-        run = lambda: step(model, opt)               # noqa: E731
+        run = lambda: step(model, opt, *inputs)      # noqa: E731
 
     for _ in range(args.warmup):
         run()
