@@ -130,8 +130,13 @@ def comm_optimize(gm: fx.GraphModule, durations: Optional[Dict[str, float]]
     sink_waits_raise_starts(gm)
     group_collectives(gm)
     if durations:
-        from ...schedule.rcpsp import odd_even_schedule
-        order = odd_even_schedule(gm, durations)
+        from ...schedule.rcpsp import milp_schedule, odd_even_schedule
+        order = None
+        if method == "milp":
+            # exact disjunctive MILP for small graphs; None on fallthrough
+            order = milp_schedule(gm, durations)
+        if order is None:
+            order = odd_even_schedule(gm, durations)
         if order is not None:
             _relink(gm, order)
     return gm

@@ -73,3 +73,52 @@ def test_comm_cse_dedupes_identical_collectives():
     # tanh now consumes the FIRST wait's result
     tanh = next(n for n in gm.graph.nodes if n.target is torch.tanh)
     assert tanh.args[0].args[0] is starts[0]
+
+
+def test_milp_schedule_overlaps_and_stays_topological():
+    """Exact RCPSP: with one transfer and independent compute, the MILP
+    must start the transfer at t=0 (full overlap) and return a valid
+    topological order with the dependent consumer after the wait."""
+    from easydist_amd.schedule.rcpsp import milp_schedule
+
+    gm = _toy_graph()
+    durations = {"add": 1.0, "relu": 2.0, "relu_1": 2.0, "relu_2": 2.0,
+                 "rt_all_reduce_start": 5.0}
+    order = milp_schedule(gm, durations)
+    assert order is not None
+    idx = {n.name: i for i, n in enumerate(order)}
+    # topological validity
+    for i, n in enumerate(order):
+        for p in n.all_input_nodes:
+            assert idx[p.name] < i, (p.name, n.name)
+    # the transfer overlaps compute: start comes before at least two of
+    # the independent relus (sequentially it was after relu/relu_1)
+    assert idx["rt_all_reduce_start"] < idx["relu_2"]
+    # output node last
+    assert order[-1].op == "output"
+
+
+def test_milp_schedule_serializes_same_resource():
+    """Two independent transfers share the comm resource: the MILP must
+    NOT start both at once — makespan >= sum of transfer durations when
+    compute is negligible."""
+    from easydist_amd.schedule.rcpsp import milp_schedule
+
+    g = fx.Graph()
+    a = g.placeholder("a")
+    b = g.placeholder("b")
+    s1 = g.call_function(crt.rt_all_reduce_start, (a, "sum", 0))
+    w1 = g.call_function(crt.rt_wait, (s1,))
+    s2 = g.call_function(crt.rt_all_reduce_start, (b, "sum", 0))
+    w2 = g.call_function(crt.rt_wait, (s2,))
+    out = g.call_function(torch.add, (w1, w2))
+    g.output((out,))
+    gm = fx.GraphModule(torch.nn.Module(), g)
+    order = milp_schedule(gm, {"rt_all_reduce_start": 3.0,
+                               "rt_all_reduce_start_1": 3.0,
+                               "add": 0.1})
+    assert order is not None
+    idx = {n.name: i for i, n in enumerate(order)}
+    for i, n in enumerate(order):
+        for p in n.all_input_nodes:
+            assert idx[p.name] < i
