@@ -87,6 +87,12 @@ def main():
     local = int(os.environ.get("LOCAL_RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     device = f"cuda:{local}" if use_cuda else "cpu"
+    if args.mode in ("auto", "zero2", "zero3"):
+        # BEFORE any CUDA context init: easydist_setup swaps in the
+        # profiling allocator (EASYDIST_MEM_OPT), which torch rejects
+        # once the default allocator has served an allocation
+        from easydist_amd import easydist_setup
+        easydist_setup(backend="torch", device="cuda" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(local)
     if not dist.is_initialized():
@@ -98,10 +104,7 @@ def main():
     model, step, inputs = build_case(args.model, device)
 
     if args.mode in ("auto", "zero2", "zero3"):
-        from easydist_amd import easydist_compile, easydist_setup, \
-            set_device_mesh
-        easydist_setup(backend="torch",
-                       device="cuda" if use_cuda else "cpu")
+        from easydist_amd import easydist_compile, set_device_mesh
         set_device_mesh(list(range(world)), ["spmd0"])
         opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=use_cuda)
         mode = args.mode if args.mode != "auto" else "auto"
