@@ -129,14 +129,24 @@ def main():
         run()
     if use_cuda:
         torch.cuda.synchronize()
-        torch.cuda.reset_peak_memory_stats()
+        try:
+            torch.cuda.reset_peak_memory_stats()
+        except RuntimeError:
+            pass          # pluggable allocator: no torch-side peak stats
     t0 = time.time()
     for _ in range(args.steps):
         run()
     if use_cuda:
         torch.cuda.synchronize()
     dt = (time.time() - t0) / args.steps
-    peak = torch.cuda.max_memory_allocated() / 2**30 if use_cuda else 0
+    peak = 0.0
+    if use_cuda:
+        try:
+            peak = torch.cuda.max_memory_allocated() / 2**30
+        except RuntimeError:
+            # static-plan playback: the packed arena IS the footprint
+            from easydist_amd.memory import _mem_alloc
+            peak = _mem_alloc.arena_size() / 2**30
     if rank == 0:
         print(json.dumps({"model": args.model, "mode": args.mode,
                           "ms_per_step": dt * 1000,
