@@ -284,3 +284,24 @@ def test_gelu_fast(ext):
     refb = torch.ops.aten.gelu_backward(g.float(), x.float(),
                                         approximate="tanh")
     assert (dy.float() - refb).abs().max() < 2e-2
+
+
+@requires_gpu
+def test_kernel_stream_tracer(ext):
+    # the profiler-based tracer attributes kernels to their HIP streams
+    import torch
+
+    from easydist_amd.utils.stream_tracer import (streams_used,
+                                                  trace_kernel_streams)
+
+    q = torch.randn(2, 2, 128, 64, device="cuda", dtype=torch.bfloat16)
+    s = torch.cuda.Stream()
+
+    def work():
+        out, _ = ext.flash_attn_fwd(q, q, q, True)
+        with torch.cuda.stream(s):
+            (q.float() * 2).sum()
+
+    m = trace_kernel_streams(work)
+    assert any("flash" in k for k in m), list(m)[:8]
+    assert len(streams_used(work)) >= 2, m
