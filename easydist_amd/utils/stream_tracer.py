@@ -38,8 +38,8 @@ def trace_kernel_streams(fn: Callable, *args,
     for ev in raw:
         try:
             if ev.device_type() == torch.autograd.DeviceType.CUDA:
-                # kineto reports the HIP stream as the "resource" row id
-                out[ev.name()].add(int(ev.resource_id()))
+                # kineto reports the HIP stream as the device resource row
+                out[ev.name()].add(int(ev.device_resource_id()))
         except Exception:
             continue
     return dict(out)
