@@ -30,7 +30,7 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 // and is injective across the 16 rows of a fragment for both row widths.
 DEVINL int lsw(int e) { return e ^ (((e >> 7) & 15) << 3); }
 
-template <int D, int RF = 1>
+template <int D, int RF = 1, int KD = 0>
 __global__ void
 __launch_bounds__(512, (D == 64 ? (RF == 1 ? 4 : 3) : 1))  // waves/SIMD floor
 flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
@@ -116,12 +116,17 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   constexpr int PF = TILE_K / (NTHR * 8);    // 16B vectors per thread
   bf16x8 kreg[PF], vreg[PF];
 
+  // KD=1: K fragments are read straight from global memory (the per-
+  // (b,h) K slab is L2-resident at these shapes) — no K staging, no
+  // K-tile LDS reads, and the tile barrier only covers the V transpose
   auto load_tile = [&](int t) {
     #pragma unroll
     for (int pi = 0; pi < PF; ++pi) {
       const int e = threadIdx.x * 8 + pi * (NTHR * 8);
       const int row = t * KB + e / D, col = e % D;
-      kreg[pi] = *reinterpret_cast<const bf16x8*>(&k[(long)row * kss + col]);
+      if (!KD)
+        kreg[pi] = *reinterpret_cast<const bf16x8*>(
+            &k[(long)row * kss + col]);
       vreg[pi] = *reinterpret_cast<const bf16x8*>(&v[(long)row * vss + col]);
     }
   };
@@ -131,7 +136,8 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     #pragma unroll
     for (int pi = 0; pi < PF; ++pi) {
       const int e = threadIdx.x * 8 + pi * (NTHR * 8);
-      *reinterpret_cast<bf16x8*>(&kb[lsw(e)]) = kreg[pi];
+      if (!KD)
+        *reinterpret_cast<bf16x8*>(&kb[lsw(e)]) = kreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i)
@@ -167,8 +173,13 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
         s_acc[j] = {0.f, 0.f, 0.f, 0.f};
         #pragma unroll
         for (int ks = 0; ks < D / 32; ++ks) {
-          bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
-              &k_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
+          bf16x8v kf;
+          if (KD)
+            kf = *reinterpret_cast<const bf16x8v*>(
+                &k[(long)(kv0 + j * 16 + fr) * kss + ks * 32 + fg * 8]);
+          else
+            kf = *reinterpret_cast<const bf16x8v*>(
+                &k_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
           s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[rf][ks], kf,
                                                              s_acc[j], 0, 0,
                                                              0);
@@ -309,13 +320,19 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q_,
     const char* e = getenv("EASYDIST_FWD_RF");
     return e ? atoi(e) : 1;
   }();
+  static int kd = []() {
+    const char* e = getenv("EASYDIST_FWD_KDIRECT");
+    return e ? atoi(e) : 0;
+  }();
   const int use_rf = (rf == 2 && S % 256 == 0) ? 2 : 1;
   dim3 grid((S + 128 * use_rf - 1) / (128 * use_rf), B * H), block(512);
   size_t lds = (2 * ((size_t)KB * D + (size_t)KB * D) + 8 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
   auto kern = (D == 64)
-      ? (use_rf == 2 ? flash_fwd_kernel<64, 2> : flash_fwd_kernel<64, 1>)
-      : (use_rf == 2 ? flash_fwd_kernel<128, 2> : flash_fwd_kernel<128, 1>);
+      ? (use_rf == 2 ? flash_fwd_kernel<64, 2>
+         : kd ? flash_fwd_kernel<64, 1, 1> : flash_fwd_kernel<64, 1>)
+      : (use_rf == 2 ? flash_fwd_kernel<128, 2>
+         : kd ? flash_fwd_kernel<128, 1, 1> : flash_fwd_kernel<128, 1>);
   hipLaunchKernelGGL(kern, grid, block, lds, stream,
       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
       (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
