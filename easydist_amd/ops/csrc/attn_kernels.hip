@@ -30,7 +30,19 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 // and is injective across the 16 rows of a fragment for both row widths.
 DEVINL int lsw(int e) { return e ^ (((e >> 7) & 15) << 3); }
 
-template <int D, int RF = 1, int KD = 0>
+// SW=1 (swapped-operand form): QK^T runs as K*Q^T so the C-layout puts
+// ONE q row per lane (col = l&15 = qrow, regs = 16 keys). The online
+// softmax then reduces over REGISTERS (15 VALU max/add) plus TWO
+// shuffles (xor 16/32 across the fg groups) instead of four dependent
+// 4-level shuffle trees, and P reaches the PV A-operand layout through
+// a 12-shuffle in-register butterfly (v_cvt_pk_bf16_f32 pairs exchanged
+// at xor distances 16/32/48) instead of an LDS strip round-trip.
+// Derivation: key K = 16j + 4g_src + r = 32ks + 8g_tgt + u maps source
+// group g_src = (K>>2)&3 to target group g_tgt = (K>>3)&3; each target
+// half-fragment (4 keys) is one source lane's consecutive regs, so per
+// ks three selected-send exchanges (xor16: g0<-g1/g3<-g2, xor32:
+// g1<-g3/g2<-g0, xor48: g1<-g2/g2<-g1) deliver every piece.
+template <int D, int RF = 1, int KD = 0, int SW = 0>
 __global__ void
 __launch_bounds__(512, (D == 64 ? (RF == 1 ? 4 : 3) : 1))  // waves/SIMD floor
 flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
@@ -165,6 +177,109 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
     for (int rf = 0; rf < RF; ++rf) {
       if (RF > 1 && causal && kv0 > qr0[rf] + 15)
         continue;                  // this fragment's rows are all-masked
+      if (SW) {
+        // ---- S^T = K Q^T: lane col = qrow, regs = 16 keys -------------
+        f32x4 s_acc[KB / 16];
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int j = 0; j < KB / 16; ++j) {
+          s_acc[j] = {0.f, 0.f, 0.f, 0.f};
+          #pragma unroll
+          for (int ks = 0; ks < D / 32; ++ks) {
+            bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
+                &k_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
+            s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                kf, qf[rf][ks], s_acc[j], 0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+        const int qrow = qr0[rf] + fr;
+        const bool need_mask = (causal && kv0 + KB - 1 > qrow)
+                               || (kv0 + KB > S);
+        float mx = -1e30f;
+        if (need_mask) {
+          #pragma unroll
+          for (int j = 0; j < KB / 16; ++j)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              int kcol = kv0 + j * 16 + 4 * fg + r;
+              float sv = s_acc[j][r];
+              if ((causal && kcol > qrow) || kcol >= S) sv = -1e30f;
+              s_acc[j][r] = sv;
+              mx = fmaxf(mx, sv);
+            }
+        } else {
+          #pragma unroll
+          for (int j = 0; j < KB / 16; ++j)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) mx = fmaxf(mx, s_acc[j][r]);
+        }
+        mx = fmaxf(mx, __shfl_xor(mx, 16, WAVE));
+        mx = fmaxf(mx, __shfl_xor(mx, 32, WAVE));
+        float m_new = fmaxf(m_run[rf][0], mx);
+        float psum = 0.f;
+        #pragma unroll
+        for (int j = 0; j < KB / 16; ++j)
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float pp = __builtin_amdgcn_exp2f(s_acc[j][r] - m_new);
+            s_acc[j][r] = pp;
+            psum += pp;
+          }
+        psum += __shfl_xor(psum, 16, WAVE);
+        psum += __shfl_xor(psum, 32, WAVE);
+        float alpha = __builtin_amdgcn_exp2f(m_run[rf][0] - m_new);
+        l_run[rf][0] = l_run[rf][0] * alpha + psum;
+        m_run[rf][0] = m_new;
+        // o_acc rows are qrows 4*fg+r: fetch those rows' alpha
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float ar = __shfl(alpha, 4 * fg + r, WAVE);
+          #pragma unroll
+          for (int j = 0; j < D / 16; ++j) o_acc[rf][j][r] *= ar;
+        }
+        // pack to bf16 pairs and run the butterfly (header comment)
+        unsigned pk[KB / 16][2];
+        #pragma unroll
+        for (int j = 0; j < KB / 16; ++j)
+          #pragma unroll
+          for (int t2 = 0; t2 < 2; ++t2)
+            asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+                : "=v"(pk[j][t2])
+                : "v"(s_acc[j][2 * t2]), "v"(s_acc[j][2 * t2 + 1]));
+        const bool g_odd = (fg & 1), g_hi = (fg >= 2);
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int ks = 0; ks < KB / 32; ++ks) {
+          unsigned o0 = pk[2 * ks][0], o1 = pk[2 * ks][1];
+          unsigned p0 = pk[2 * ks + 1][0], p1 = pk[2 * ks + 1][1];
+          unsigned r16a = __shfl_xor((int)(g_odd ? o0 : p0), 16, WAVE);
+          unsigned r16b = __shfl_xor((int)(g_odd ? o1 : p1), 16, WAVE);
+          unsigned r32a = __shfl_xor((int)(g_hi ? o0 : p0), 32, WAVE);
+          unsigned r32b = __shfl_xor((int)(g_hi ? o1 : p1), 32, WAVE);
+          unsigned r48a = __shfl_xor((int)(fg == 2 ? o0 : p0), 48, WAVE);
+          unsigned r48b = __shfl_xor((int)(fg == 2 ? o1 : p1), 48, WAVE);
+          union { unsigned u[4]; bf16x8v v; } cv;
+          if (fg == 0) {
+            cv.u[0] = o0;  cv.u[1] = o1;  cv.u[2] = r16a; cv.u[3] = r16b;
+          } else if (fg == 1) {
+            cv.u[0] = r48a; cv.u[1] = r48b; cv.u[2] = r32a; cv.u[3] = r32b;
+          } else if (fg == 2) {
+            cv.u[0] = r32a; cv.u[1] = r32b; cv.u[2] = r48a; cv.u[3] = r48b;
+          } else {
+            cv.u[0] = r16a; cv.u[1] = r16b; cv.u[2] = p0;  cv.u[3] = p1;
+          }
+          #pragma unroll
+          for (int j = 0; j < D / 16; ++j) {
+            bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
+                &vt_lds[lsw((j * 16 + fr) * KB + ks * 32 + fg * 8)]);
+            o_acc[rf][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                cv.v, vf, o_acc[rf][j], 0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+        continue;                  // SW path done for this fragment
+      }
       // ---- S = Q K^T ----------------------------------------------------
       f32x4 s_acc[KB / 16];
       __builtin_amdgcn_s_setprio(1);
@@ -281,6 +396,25 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
   // ---- epilogue --------------------------------------------------------
   #pragma unroll
   for (int rf = 0; rf < RF; ++rf) {
+    if (SW) {
+      // softmax state lives per lane (col = qrow); o_acc rows need the
+      // row-owners' 1/l via shuffle
+      float invl = 1.f / l_run[rf][0];
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int qrow = qr0[rf] + 4 * fg + r;
+        if (qrow >= S) continue;
+        float ir = __shfl(invl, 4 * fg + r, WAVE);
+        #pragma unroll
+        for (int j = 0; j < D / 16; ++j)
+          o[(long)qrow * D + j * 16 + fr] = f2bf(o_acc[rf][j][r] * ir);
+      }
+      int myrow = qr0[rf] + fr;
+      if (fg == 0 && myrow < S)
+        lse[myrow] = 0.69314718055994530942f
+                     * (m_run[rf][0] + __log2f(l_run[rf][0]));
+      continue;
+    }
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int qrow = qr0[rf] + 4 * fg + r;
@@ -324,14 +458,20 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q_,
     const char* e = getenv("EASYDIST_FWD_KDIRECT");
     return e ? atoi(e) : 0;
   }();
+  static int sw = []() {
+    const char* e = getenv("EASYDIST_FWD_SWAP");
+    return e ? atoi(e) : 0;
+  }();
   const int use_rf = (rf == 2 && S % 256 == 0) ? 2 : 1;
   dim3 grid((S + 128 * use_rf - 1) / (128 * use_rf), B * H), block(512);
   size_t lds = (2 * ((size_t)KB * D + (size_t)KB * D) + 8 * 16 * KB) * 2;
   float scale = 1.f / sqrtf((float)D);
   auto kern = (D == 64)
       ? (use_rf == 2 ? flash_fwd_kernel<64, 2>
+         : sw ? flash_fwd_kernel<64, 1, 0, 1>
          : kd ? flash_fwd_kernel<64, 1, 1> : flash_fwd_kernel<64, 1>)
       : (use_rf == 2 ? flash_fwd_kernel<128, 2>
+         : sw ? flash_fwd_kernel<128, 1, 0, 1>
          : kd ? flash_fwd_kernel<128, 1, 1> : flash_fwd_kernel<128, 1>);
   hipLaunchKernelGGL(kern, grid, block, lds, stream,
       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
