@@ -458,9 +458,9 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q_,
     const char* e = getenv("EASYDIST_FWD_KDIRECT");
     return e ? atoi(e) : 0;
   }();
-  static int sw = []() {
+  static int sw = []() {   // default ON: -17% (D=64) / -12% (D=128) A/B'd
     const char* e = getenv("EASYDIST_FWD_SWAP");
-    return e ? atoi(e) : 0;
+    return e ? atoi(e) : 1;
   }();
   const int use_rf = (rf == 2 && S % 256 == 0) ? 2 : 1;
   dim3 grid((S + 128 * use_rf - 1) / (128 * use_rf), B * H), block(512);
