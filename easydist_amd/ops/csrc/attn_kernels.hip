@@ -30,6 +30,46 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 // and is injective across the 16 rows of a fragment for both row widths.
 DEVINL int lsw(int e) { return e ^ (((e >> 7) & 15) << 3); }
 
+
+// ---- shared pieces of the swapped-operand (S^T) form ----------------------
+// pack a C-layout f32 fragment row-pair into bf16x2 words
+DEVINL void swp_pack(const f32x4* sacc, int nj, unsigned pk[][2]) {
+  for (int j = 0; j < nj; ++j)
+    #pragma unroll
+    for (int t2 = 0; t2 < 2; ++t2)
+      asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+          : "=v"(pk[j][t2])
+          : "v"(sacc[j][2 * t2]), "v"(sacc[j][2 * t2 + 1]));
+}
+
+// in-register butterfly: C-layout (lane col = row m, regs = 16 k's) ->
+// A-operand bf16x8 fragment for mfma #ks.  Key K = 16j + 4g_src + r =
+// 32ks + 8g_tgt + u maps g_src=(K>>2)&3 to g_tgt=(K>>3)&3; each target
+// half-fragment is one source lane's consecutive regs, so three
+// selected-send exchanges (xor 16/32/48) deliver every piece.
+DEVINL bf16x8v swp_butterfly(const unsigned pk[][2], int ks, int fg) {
+  unsigned o0 = pk[2 * ks][0], o1 = pk[2 * ks][1];
+  unsigned p0 = pk[2 * ks + 1][0], p1 = pk[2 * ks + 1][1];
+  const bool g_odd = (fg & 1), g_hi = (fg >= 2);
+  unsigned r16a = __shfl_xor((int)(g_odd ? o0 : p0), 16, WAVE);
+  unsigned r16b = __shfl_xor((int)(g_odd ? o1 : p1), 16, WAVE);
+  unsigned r32a = __shfl_xor((int)(g_hi ? o0 : p0), 32, WAVE);
+  unsigned r32b = __shfl_xor((int)(g_hi ? o1 : p1), 32, WAVE);
+  unsigned r48a = __shfl_xor((int)(fg == 2 ? o0 : p0), 48, WAVE);
+  unsigned r48b = __shfl_xor((int)(fg == 2 ? o1 : p1), 48, WAVE);
+  union { unsigned u[4]; bf16x8v v; } cv;
+  if (fg == 0) {
+    cv.u[0] = o0;  cv.u[1] = o1;  cv.u[2] = r16a; cv.u[3] = r16b;
+  } else if (fg == 1) {
+    cv.u[0] = r48a; cv.u[1] = r48b; cv.u[2] = r32a; cv.u[3] = r32b;
+  } else if (fg == 2) {
+    cv.u[0] = r32a; cv.u[1] = r32b; cv.u[2] = r48a; cv.u[3] = r48b;
+  } else {
+    cv.u[0] = r16a; cv.u[1] = r16b; cv.u[2] = p0;  cv.u[3] = p1;
+  }
+  return cv.v;
+}
+
 // SW=1 (swapped-operand form): QK^T runs as K*Q^T so the C-layout puts
 // ONE q row per lane (col = l&15 = qrow, regs = 16 keys). The online
 // softmax then reduces over REGISTERS (15 VALU max/add) plus TWO
@@ -238,43 +278,19 @@ flash_fwd_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ K,
           #pragma unroll
           for (int j = 0; j < D / 16; ++j) o_acc[rf][j][r] *= ar;
         }
-        // pack to bf16 pairs and run the butterfly (header comment)
+        // pack to bf16 pairs and run the butterfly (swp_* helpers)
         unsigned pk[KB / 16][2];
-        #pragma unroll
-        for (int j = 0; j < KB / 16; ++j)
-          #pragma unroll
-          for (int t2 = 0; t2 < 2; ++t2)
-            asm("v_cvt_pk_bf16_f32 %0, %1, %2"
-                : "=v"(pk[j][t2])
-                : "v"(s_acc[j][2 * t2]), "v"(s_acc[j][2 * t2 + 1]));
-        const bool g_odd = (fg & 1), g_hi = (fg >= 2);
+        swp_pack(s_acc, KB / 16, pk);
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int ks = 0; ks < KB / 32; ++ks) {
-          unsigned o0 = pk[2 * ks][0], o1 = pk[2 * ks][1];
-          unsigned p0 = pk[2 * ks + 1][0], p1 = pk[2 * ks + 1][1];
-          unsigned r16a = __shfl_xor((int)(g_odd ? o0 : p0), 16, WAVE);
-          unsigned r16b = __shfl_xor((int)(g_odd ? o1 : p1), 16, WAVE);
-          unsigned r32a = __shfl_xor((int)(g_hi ? o0 : p0), 32, WAVE);
-          unsigned r32b = __shfl_xor((int)(g_hi ? o1 : p1), 32, WAVE);
-          unsigned r48a = __shfl_xor((int)(fg == 2 ? o0 : p0), 48, WAVE);
-          unsigned r48b = __shfl_xor((int)(fg == 2 ? o1 : p1), 48, WAVE);
-          union { unsigned u[4]; bf16x8v v; } cv;
-          if (fg == 0) {
-            cv.u[0] = o0;  cv.u[1] = o1;  cv.u[2] = r16a; cv.u[3] = r16b;
-          } else if (fg == 1) {
-            cv.u[0] = r48a; cv.u[1] = r48b; cv.u[2] = r32a; cv.u[3] = r32b;
-          } else if (fg == 2) {
-            cv.u[0] = r32a; cv.u[1] = r32b; cv.u[2] = r48a; cv.u[3] = r48b;
-          } else {
-            cv.u[0] = r16a; cv.u[1] = r16b; cv.u[2] = p0;  cv.u[3] = p1;
-          }
+          bf16x8v pv = swp_butterfly(pk, ks, fg);
           #pragma unroll
           for (int j = 0; j < D / 16; ++j) {
             bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
                 &vt_lds[lsw((j * 16 + fr) * KB + ks * 32 + fg * 8)]);
             o_acc[rf][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                cv.v, vf, o_acc[rf][j], 0, 0, 0);
+                pv, vf, o_acc[rf][j], 0, 0, 0);
           }
         }
         __builtin_amdgcn_s_setprio(0);
@@ -496,7 +512,7 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q_,
 // hold row fr, k-slice fg; C/D lanes hold row 4*fg+r, col fr.
 // ---------------------------------------------------------------------------
 
-template <int D>
+template <int D, int SW = 0>
 __global__ void __launch_bounds__(512)
 flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                     const bf16* __restrict__ K, const bf16* __restrict__ V,
@@ -542,13 +558,18 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     dof[ks] = *reinterpret_cast<const bf16x8v*>(
         &dO_[(long)(qr0 + fr) * dss + ks * 32 + fg * 8]);
   }
-  // per C-row lse/delta (rows 4*fg+r of this wave)
+  // per C-row lse/delta (rows 4*fg+r; SW: one row per lane = fr)
   float lse_r[4], dlt_r[4];
-  #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int qrow = qr0 + 4 * fg + r;
-    lse_r[r] = lse[qrow];
-    dlt_r[r] = delta[qrow];
+  if (!SW) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int qrow = qr0 + 4 * fg + r;
+      lse_r[r] = lse[qrow];
+      dlt_r[r] = delta[qrow];
+    }
+  } else {
+    lse_r[0] = lse[qr0 + fr];
+    dlt_r[0] = delta[qr0 + fr];
   }
 
   f32x4 dq_acc[D / 16];
@@ -588,7 +609,8 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     if (kv0 + KB < kv_end) kv_load(kv0 + KB);
     if (causal && kv0 > qr0 + 15) { __syncthreads(); continue; }
 
-    // S = Q K^T and dP = dO V^T (both mfma-native: B = rows)
+    // S(^T) = Q K^T and dP(^T) = dO V^T: SW=1 swaps operands so the
+    // C-layout holds ONE q row per lane (col = fr) with 16 keys in regs
     f32x4 s_acc[KB / 16], dp_acc[KB / 16];
     #pragma unroll
     for (int j = 0; j < KB / 16; ++j) {
@@ -600,14 +622,41 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
             &k_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
             &v_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
-        s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], kf,
-                                                           s_acc[j], 0, 0, 0);
-        dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[ks], vf,
-                                                            dp_acc[j], 0, 0,
-                                                            0);
+        if (SW) {
+          s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              kf, qf[ks], s_acc[j], 0, 0, 0);
+          dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              vf, dof[ks], dp_acc[j], 0, 0, 0);
+        } else {
+          s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[ks], kf, s_acc[j], 0, 0, 0);
+          dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              dof[ks], vf, dp_acc[j], 0, 0, 0);
+        }
       }
     }
     // dS = P * (dP - delta) * scale, P = exp(S*scale - lse)
+    bf16x8v dsf[KB / 32];
+    if (SW) {
+      const int qrow = qr0 + fr;
+      const bool need_mask = (causal && kv0 + KB - 1 > qrow)
+                             || (kv0 + KB > S);
+      #pragma unroll
+      for (int j = 0; j < KB / 16; ++j)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float p = 0.f;
+          int kcol = kv0 + j * 16 + 4 * fg + r;
+          if (!need_mask || (!(causal && kcol > qrow) && kcol < S))
+            p = __expf(s_acc[j][r] * scale - lse_r[0]);
+          s_acc[j][r] = p * (dp_acc[j][r] - dlt_r[0]) * scale;
+        }
+      unsigned pk[KB / 16][2];
+      swp_pack(s_acc, KB / 16, pk);
+      #pragma unroll
+      for (int ks = 0; ks < KB / 32; ++ks)
+        dsf[ks] = swp_butterfly(pk, ks, fg);
+    } else {
     const bool need_mask = (causal && kv0 + KB - 1 > qr0 + 4 * fg)
                            || (kv0 + KB > S);
     if (need_mask) {
@@ -640,11 +689,11 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       for (int r = 0; r < 4; ++r)
         s_lds[lsw((4 * fg + r) * KB + j * 16 + fr)] = f2bf(s_acc[j][r]);
     lds_fence();
-    bf16x8v dsf[KB / 32];
     #pragma unroll
     for (int ks = 0; ks < KB / 32; ++ks)
       dsf[ks] = *reinterpret_cast<const bf16x8v*>(
           &s_lds[lsw(fr * KB + ks * 32 + fg * 8)]);
+    }
     // dq += dS @ K: B-operand from K^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
@@ -930,7 +979,13 @@ flash_attn_bwd_launch(const at::Tensor& grad_, const at::Tensor& q_,
   size_t lds_kv = (2 * QB * (size_t)D + 2 * (size_t)QB * D
                    + 8 * 16 * QB) * 2;
   float scale = 1.f / sqrtf((float)D);
-  auto qkern = (D == 64) ? flash_bwd_dq_kernel<64> : flash_bwd_dq_kernel<128>;
+  static int swb = []() {   // swapped-operand dq (butterfly dS transport)
+    const char* e = getenv("EASYDIST_BWD_SWAP");
+    return e ? atoi(e) : 1;
+  }();
+  auto qkern = (D == 64)
+      ? (swb ? flash_bwd_dq_kernel<64, 1> : flash_bwd_dq_kernel<64>)
+      : (swb ? flash_bwd_dq_kernel<128, 1> : flash_bwd_dq_kernel<128>);
   auto kkern = (D == 64) ? flash_bwd_dkv_kernel<64>
                          : flash_bwd_dkv_kernel<128>;
   hipLaunchKernelGGL(qkern, grid, block, lds, stream,
