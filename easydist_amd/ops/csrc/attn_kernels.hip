@@ -718,7 +718,7 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   }
 }
 
-template <int D>
+template <int D, int SW = 0>
 __global__ void __launch_bounds__(512)
 flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                      const bf16* __restrict__ K, const bf16* __restrict__ V,
@@ -804,7 +804,10 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     if (q0 + QB < S) q_load(q0 + QB);
     if (causal && q0 + QB - 1 < kr0) { __syncthreads(); continue; }
 
-    // S^T = K Q^T and dP^T = V dO^T (B operands: row reads from LDS)
+    // S^T = K Q^T and dP^T = V dO^T.  SW=1 swaps the operand order so
+    // the C-layout holds the wave's 16 KEYS on the lane column and 16
+    // q cols in regs — the swp_butterfly then builds the P^T/dS^T
+    // A-operands in-register (no strip, no fences).
     f32x4 st_acc[QB / 16], dpt_acc[QB / 16];
     #pragma unroll
     for (int j = 0; j < QB / 16; ++j) {
@@ -816,18 +819,42 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
             &q_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         bf16x8v dob = *reinterpret_cast<const bf16x8v*>(
             &do_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
-        st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[ks], qfb,
-                                                            st_acc[j], 0, 0,
-                                                            0);
-        dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[ks], dob,
-                                                             dpt_acc[j], 0,
-                                                             0, 0);
+        if (SW) {
+          st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qfb, kf[ks], st_acc[j], 0, 0, 0);
+          dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              dob, vf[ks], dpt_acc[j], 0, 0, 0);
+        } else {
+          st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              kf[ks], qfb, st_acc[j], 0, 0, 0);
+          dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              vf[ks], dob, dpt_acc[j], 0, 0, 0);
+        }
       }
     }
     // P^T = exp(S^T*scale - lse[qcol]); dS^T = P^T (dP^T - delta[qcol]) scale
+    bf16x8v ptf[QB / 32];
+    const bool need_mask = (causal && kr0 + 15 > q0) || (q0 + QB > S);
+    if (SW) {
+      const int krow = kr0 + fr;             // this lane's key row
+      #pragma unroll
+      for (int j = 0; j < QB / 16; ++j)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int qcol = q0 + j * 16 + 4 * fg + r;
+          float p = 0.f;
+          if (!need_mask || (!(causal && krow > qcol) && qcol < S))
+            p = __expf(st_acc[j][r] * scale - lse[qcol]);
+          st_acc[j][r] = p;
+        }
+      unsigned pk[QB / 16][2];
+      swp_pack(st_acc, QB / 16, pk);
+      #pragma unroll
+      for (int ks = 0; ks < QB / 32; ++ks)
+        ptf[ks] = swp_butterfly(pk, ks, fg);
+    } else {
     // first pass: P^T into the strip for the dV mfma
     // no-mask fast path: whole wave's keys are <= every q col in tile
-    const bool need_mask = (causal && kr0 + 15 > q0) || (q0 + QB > S);
     if (need_mask) {
       #pragma unroll
       for (int j = 0; j < QB / 16; ++j) {
@@ -856,11 +883,11 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       }
     }
     lds_fence();
-    bf16x8v ptf[QB / 32];
     #pragma unroll
     for (int ks = 0; ks < QB / 32; ++ks)
       ptf[ks] = *reinterpret_cast<const bf16x8v*>(
           &s_lds[lsw(fr * QB + ks * 32 + fg * 8)]);
+    }
     // dV += P^T @ dO: B-operand from dO^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
@@ -873,7 +900,24 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
                                                             0);
       }
     }
-    // dS^T into the strip for the dK mfma
+    // dS^T for the dK mfma: SW builds it in-register via the butterfly
+    bf16x8v dstf[QB / 32];
+    if (SW) {
+      #pragma unroll
+      for (int j = 0; j < QB / 16; ++j)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int qcol = q0 + j * 16 + 4 * fg + r;
+          float ds = st_acc[j][r]
+                     * (dpt_acc[j][r] - delta[min(qcol, S - 1)]) * scale;
+          st_acc[j][r] = ds;
+        }
+      unsigned pk2[QB / 16][2];
+      swp_pack(st_acc, QB / 16, pk2);
+      #pragma unroll
+      for (int ks = 0; ks < QB / 32; ++ks)
+        dstf[ks] = swp_butterfly(pk2, ks, fg);
+    } else {
     lds_fence();
     #pragma unroll
     for (int j = 0; j < QB / 16; ++j) {
@@ -886,11 +930,11 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       }
     }
     lds_fence();
-    bf16x8v dstf[QB / 32];
     #pragma unroll
     for (int ks = 0; ks < QB / 32; ++ks)
       dstf[ks] = *reinterpret_cast<const bf16x8v*>(
           &s_lds[lsw(fr * QB + ks * 32 + fg * 8)]);
+    }
     // dK += dS^T @ Q: B-operand from Q^T — contiguous vector loads
     #pragma unroll
     for (int j = 0; j < D / 16; ++j) {
@@ -986,8 +1030,9 @@ flash_attn_bwd_launch(const at::Tensor& grad_, const at::Tensor& q_,
   auto qkern = (D == 64)
       ? (swb ? flash_bwd_dq_kernel<64, 1> : flash_bwd_dq_kernel<64>)
       : (swb ? flash_bwd_dq_kernel<128, 1> : flash_bwd_dq_kernel<128>);
-  auto kkern = (D == 64) ? flash_bwd_dkv_kernel<64>
-                         : flash_bwd_dkv_kernel<128>;
+  auto kkern = (D == 64)
+      ? (swb ? flash_bwd_dkv_kernel<64, 1> : flash_bwd_dkv_kernel<64>)
+      : (swb ? flash_bwd_dkv_kernel<128, 1> : flash_bwd_dkv_kernel<128>);
   hipLaunchKernelGGL(qkern, grid, block, lds, stream,
       (const bf16*)grad.data_ptr(), (const bf16*)q.data_ptr(),
       (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
