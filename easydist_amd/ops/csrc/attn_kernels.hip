@@ -837,21 +837,23 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     const bool need_mask = (causal && kr0 + 15 > q0) || (q0 + QB > S);
     if (SW) {
       const int krow = kr0 + fr;             // this lane's key row
-      // ONE lse/delta load per lane covers the tile's 64 q cols; the
-      // per-reg values arrive by shuffle (16 scattered 4-B loads per
-      // pass measured FAR slower than 16 ds_bpermutes)
-      const float lse_lane = lse[q0 + lane];
+      // the 4 lse values a lane needs per j are CONSECUTIVE (4*fg+r):
+      // one aligned 16-B vector load per j (scattered scalar loads and
+      // ds_bpermute broadcasts both measured slower — the latter fight
+      // the tile staging for the LDS port)
       #pragma unroll
-      for (int j = 0; j < QB / 16; ++j)
+      for (int j = 0; j < QB / 16; ++j) {
+        f32x4 l4 = *reinterpret_cast<const f32x4*>(
+            &lse[q0 + j * 16 + 4 * fg]);
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int qcol = q0 + j * 16 + 4 * fg + r;
-          float l = __shfl(lse_lane, j * 16 + 4 * fg + r, WAVE);
           float p = 0.f;
           if (!need_mask || (!(causal && krow > qcol) && qcol < S))
-            p = __expf(st_acc[j][r] * scale - l);
+            p = __expf(st_acc[j][r] * scale - l4[r]);
           st_acc[j][r] = p;
         }
+      }
       unsigned pk[QB / 16][2];
       swp_pack(st_acc, QB / 16, pk);
       #pragma unroll
@@ -908,15 +910,16 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     // dS^T for the dK mfma: SW builds it in-register via the butterfly
     bf16x8v dstf[QB / 32];
     if (SW) {
-      const float dlt_lane = delta[q0 + lane];
       #pragma unroll
-      for (int j = 0; j < QB / 16; ++j)
+      for (int j = 0; j < QB / 16; ++j) {
+        f32x4 d4 = *reinterpret_cast<const f32x4*>(
+            &delta[q0 + j * 16 + 4 * fg]);
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          float dlt = __shfl(dlt_lane, j * 16 + 4 * fg + r, WAVE);
-          float ds = st_acc[j][r] * (dpt_acc[j][r] - dlt) * scale;
+          float ds = st_acc[j][r] * (dpt_acc[j][r] - d4[r]) * scale;
           st_acc[j][r] = ds;
         }
+      }
       unsigned pk2[QB / 16][2];
       swp_pack(st_acc, QB / 16, pk2);
       #pragma unroll
