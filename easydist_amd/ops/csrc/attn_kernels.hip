@@ -1038,9 +1038,13 @@ flash_attn_bwd_launch(const at::Tensor& grad_, const at::Tensor& q_,
   auto qkern = (D == 64)
       ? (swb ? flash_bwd_dq_kernel<64, 1> : flash_bwd_dq_kernel<64>)
       : (swb ? flash_bwd_dq_kernel<128, 1> : flash_bwd_dq_kernel<128>);
+  static int swkv = []() {  // dkv butterfly measured SLOWER than the
+    const char* e = getenv("EASYDIST_DKV_SWAP");   // strip (2.11 vs 1.74
+    return e ? atoi(e) : 0;                        // ms) — default off
+  }();
   auto kkern = (D == 64)
-      ? (swb ? flash_bwd_dkv_kernel<64, 1> : flash_bwd_dkv_kernel<64>)
-      : (swb ? flash_bwd_dkv_kernel<128, 1> : flash_bwd_dkv_kernel<128>);
+      ? (swkv ? flash_bwd_dkv_kernel<64, 1> : flash_bwd_dkv_kernel<64>)
+      : (swkv ? flash_bwd_dkv_kernel<128, 1> : flash_bwd_dkv_kernel<128>);
   hipLaunchKernelGGL(qkern, grid, block, lds, stream,
       (const bf16*)grad.data_ptr(), (const bf16*)q.data_ptr(),
       (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
