@@ -841,6 +841,10 @@ at::Tensor gemm_nt_act(const at::Tensor& a, const at::Tensor& bt,
       else if (act == 3) c = at::gelu(c);
       else if (act == 2) c = at::gelu_backward(c, aux.value(), "tanh");
       else if (act == 4) c = at::gelu_backward(c, aux.value(), "none");
+      else if (act >= 5) {           // fused fwd: keep pre-act in aux
+        aux.value().copy_(c);
+        c = at::gelu(c, "tanh");
+      }
     }
   }
   return c;
