@@ -543,11 +543,17 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   const int fr = lane & 15;
   const int fg = lane >> 4;
 
+  // DOUBLE-BUFFERED K/V/K^T tiles (3 tiles per buffer): the single-
+  // buffered store->barrier->compute form serialized the full staging
+  // latency into every tile (bwd ran at ~127 TF vs the double-buffered
+  // forward's 158+)
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* k_lds = reinterpret_cast<bf16*>(smem);
+  constexpr int DQBUF = 3 * KB * D;                // elements per buffer
+  bf16* k_lds = reinterpret_cast<bf16*>(smem);     // + buf * DQBUF
   bf16* v_lds = k_lds + KB * D;
   bf16* kt_lds = v_lds + KB * D;                   // K^T [D][KB] + lsw
-  bf16* s_lds = kt_lds + KB * D + wave * 16 * KB;   // wave-private strip
+  bf16* s_lds = reinterpret_cast<bf16*>(smem) + 2 * DQBUF
+                + wave * 16 * KB;                   // wave-private strip
 
   // A-operand fragments for this wave's 16 q rows
   bf16x8v qf[D / 32], dof[D / 32];
@@ -590,24 +596,31 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       vreg[pi] = *reinterpret_cast<const bf16x8*>(&v[(long)t0 * D + e]);
     }
   };
-  auto kv_store = [&]() {
+  auto kv_store = [&](int buf) {
     #pragma unroll
     for (int pi = 0; pi < NV; ++pi) {
       const int e = threadIdx.x * 8 + pi * 4096;
-      *reinterpret_cast<bf16x8*>(&k_lds[lsw(e)]) = kreg[pi];
-      *reinterpret_cast<bf16x8*>(&v_lds[lsw(e)]) = vreg[pi];
+      *reinterpret_cast<bf16x8*>(&k_lds[buf * DQBUF + lsw(e)]) = kreg[pi];
+      *reinterpret_cast<bf16x8*>(&v_lds[buf * DQBUF + lsw(e)]) = vreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i)
-        kt_lds[lsw((col + i) * KB + row)] = kreg[pi].v[i];
+        kt_lds[buf * DQBUF + lsw((col + i) * KB + row)] = kreg[pi].v[i];
     }
   };
   kv_load(0);
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
-    kv_store();
-    __syncthreads();
-    if (kv0 + KB < kv_end) kv_load(kv0 + KB);
-    if (causal && kv0 > qr0 + 15) { __syncthreads(); continue; }
+  kv_store(0);
+  __syncthreads();
+  const int n_kt = (kv_end + KB - 1) / KB;
+  for (int tt = 0; tt < n_kt; ++tt) {
+    const int kv0 = tt * KB;
+    const int bofs = (tt & 1) * DQBUF;
+    if (tt + 1 < n_kt) kv_load(kv0 + KB);
+    if (causal && kv0 > qr0 + 15) {
+      if (tt + 1 < n_kt) kv_store((tt + 1) & 1);
+      __syncthreads();
+      continue;
+    }
 
     // S(^T) = Q K^T and dP(^T) = dO V^T: SW=1 swaps operands so the
     // C-layout holds ONE q row per lane (col = fr) with 16 keys in regs
@@ -619,9 +632,9 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < D / 32; ++ks) {
         bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
-            &k_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
+            &k_lds[bofs + lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
-            &v_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
+            &v_lds[bofs + lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         if (SW) {
           s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               kf, qf[ks], s_acc[j], 0, 0, 0);
@@ -700,13 +713,14 @@ flash_bwd_dq_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < KB / 32; ++ks) {
         bf16x8v kcol = *reinterpret_cast<const bf16x8v*>(
-            &kt_lds[lsw((j * 16 + fr) * KB + ks * 32 + fg * 8)]);
+            &kt_lds[bofs + lsw((j * 16 + fr) * KB + ks * 32 + fg * 8)]);
         dq_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf[ks], kcol,
                                                             dq_acc[j], 0, 0,
                                                             0);
       }
     }
-    __syncthreads();   // all reads of this tile done before re-staging
+    if (tt + 1 < n_kt) kv_store((tt + 1) & 1);  // other buffer: no hazard
+    __syncthreads();
   }
   #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -1027,7 +1041,8 @@ flash_attn_bwd_launch(const at::Tensor& grad_, const at::Tensor& q_,
   }
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid(S / (2 * QB), B * H), block(512);
-  size_t lds = (2 * KB * (size_t)D + (size_t)KB * D + 8 * 16 * KB) * 2;
+  // dq: DOUBLE-buffered (K+V+K^T) tiles + wave strips
+  size_t lds = (2 * (3 * (size_t)KB * D) + 8 * 16 * KB) * 2;
   size_t lds_kv = (2 * QB * (size_t)D + 2 * (size_t)QB * D
                    + 8 * 16 * QB) * 2;
   float scale = 1.f / sqrtf((float)D);
