@@ -762,12 +762,14 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
   const int fr = lane & 15;
   const int fg = lane >> 4;
 
+  // DOUBLE-BUFFERED Q/dO/Q^T/dO^T tiles (see dq note)
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* q_lds = reinterpret_cast<bf16*>(smem);
+  constexpr int KVBUF = 4 * QB * D;                // elements per buffer
+  bf16* q_lds = reinterpret_cast<bf16*>(smem);     // + buf * KVBUF
   bf16* do_lds = q_lds + QB * D;
   bf16* qt_lds = do_lds + QB * D;                  // Q^T [D][QB] + lsw
   bf16* dot_lds = qt_lds + QB * D;                 // dO^T [D][QB] + lsw
-  bf16* s_lds = dot_lds + QB * D + wave * 16 * QB;
+  bf16* s_lds = reinterpret_cast<bf16*>(smem) + 2 * KVBUF + wave * 16 * QB;
 
   bf16x8v kf[D / 32], vf[D / 32];
   #pragma unroll
@@ -797,26 +799,33 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       dreg[pi] = *reinterpret_cast<const bf16x8*>(&dO_[(long)t0 * D + e]);
     }
   };
-  auto q_store = [&]() {
+  auto q_store = [&](int buf) {
     #pragma unroll
     for (int pi = 0; pi < NV; ++pi) {
       const int e = threadIdx.x * 8 + pi * 4096;
-      *reinterpret_cast<bf16x8*>(&q_lds[lsw(e)]) = qreg[pi];
-      *reinterpret_cast<bf16x8*>(&do_lds[lsw(e)]) = dreg[pi];
+      *reinterpret_cast<bf16x8*>(&q_lds[buf * KVBUF + lsw(e)]) = qreg[pi];
+      *reinterpret_cast<bf16x8*>(&do_lds[buf * KVBUF + lsw(e)]) = dreg[pi];
       const int row = e / D, col = e % D;
       #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        qt_lds[lsw((col + i) * QB + row)] = qreg[pi].v[i];
-        dot_lds[lsw((col + i) * QB + row)] = dreg[pi].v[i];
+        qt_lds[buf * KVBUF + lsw((col + i) * QB + row)] = qreg[pi].v[i];
+        dot_lds[buf * KVBUF + lsw((col + i) * QB + row)] = dreg[pi].v[i];
       }
     }
   };
   q_load(q_start);
-  for (int q0 = q_start; q0 < S; q0 += QB) {
-    q_store();
-    __syncthreads();
-    if (q0 + QB < S) q_load(q0 + QB);
-    if (causal && q0 + QB - 1 < kr0) { __syncthreads(); continue; }
+  q_store(0);
+  __syncthreads();
+  const int n_qt = (S - q_start + QB - 1) / QB;
+  for (int tt = 0; tt < n_qt; ++tt) {
+    const int q0 = q_start + tt * QB;
+    const int bofs = (tt & 1) * KVBUF;
+    if (tt + 1 < n_qt) q_load(q0 + QB);
+    if (causal && q0 + QB - 1 < kr0) {
+      if (tt + 1 < n_qt) q_store((tt + 1) & 1);
+      __syncthreads();
+      continue;
+    }
 
     // S^T = K Q^T and dP^T = V dO^T.  SW=1 swaps the operand order so
     // the C-layout holds the wave's 16 KEYS on the lane column and 16
@@ -830,9 +839,9 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < D / 32; ++ks) {
         bf16x8v qfb = *reinterpret_cast<const bf16x8v*>(
-            &q_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
+            &q_lds[bofs + lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         bf16x8v dob = *reinterpret_cast<const bf16x8v*>(
-            &do_lds[lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
+            &do_lds[bofs + lsw((j * 16 + fr) * D + ks * 32 + fg * 8)]);
         if (SW) {
           st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               qfb, kf[ks], st_acc[j], 0, 0, 0);
@@ -915,7 +924,7 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < QB / 32; ++ks) {
         bf16x8v docol = *reinterpret_cast<const bf16x8v*>(
-            &dot_lds[lsw((j * 16 + fr) * QB + ks * 32 + fg * 8)]);
+            &dot_lds[bofs + lsw((j * 16 + fr) * QB + ks * 32 + fg * 8)]);
         dv_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf[ks], docol,
                                                             dv_acc[j], 0, 0,
                                                             0);
@@ -963,13 +972,14 @@ flash_bwd_dkv_kernel(const bf16* __restrict__ dO, const bf16* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < QB / 32; ++ks) {
         bf16x8v qcolf = *reinterpret_cast<const bf16x8v*>(
-            &qt_lds[lsw((j * 16 + fr) * QB + ks * 32 + fg * 8)]);
+            &qt_lds[bofs + lsw((j * 16 + fr) * QB + ks * 32 + fg * 8)]);
         dk_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf[ks], qcolf,
                                                             dk_acc[j], 0, 0,
                                                             0);
       }
     }
-    __syncthreads();   // all reads of this tile done before re-staging
+    if (tt + 1 < n_qt) q_store((tt + 1) & 1);   // other buffer: no hazard
+    __syncthreads();
   }
   #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -1043,8 +1053,8 @@ flash_attn_bwd_launch(const at::Tensor& grad_, const at::Tensor& q_,
   dim3 grid(S / (2 * QB), B * H), block(512);
   // dq: DOUBLE-buffered (K+V+K^T) tiles + wave strips
   size_t lds = (2 * (3 * (size_t)KB * D) + 8 * 16 * KB) * 2;
-  size_t lds_kv = (2 * QB * (size_t)D + 2 * (size_t)QB * D
-                   + 8 * 16 * QB) * 2;
+  // dkv: DOUBLE-buffered (Q+dO+Q^T+dO^T) tiles + wave strips
+  size_t lds_kv = (2 * (4 * (size_t)QB * D) + 8 * 16 * QB) * 2;
   float scale = 1.f / sqrtf((float)D);
   static int swb = []() {   // swapped-operand dq (butterfly dS transport)
     const char* e = getenv("EASYDIST_BWD_SWAP");
