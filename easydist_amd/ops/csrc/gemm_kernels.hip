@@ -735,21 +735,12 @@ gemm_tn_256_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
     for (int j = 0; j < 8; ++j)
       acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // TRIPLE-buffered counted-vmcnt pipeline: each stage is 4 glds per
-  // thread (2 per operand); keeping TWO stages in flight and waiting to
-  // vmcnt(4) (= oldest stage landed) covers most of the HBM latency the
-  // old full-drain-per-step form exposed (measured 24% above the
-  // traffic floor). The per-step barrier still publishes cross-wave.
   stage(0, rt0);
-  if (rt0 + 1 < rt1) stage(1, rt0 + 1);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
   for (int rt = rt0; rt < rt1; ++rt) {
-    int buf = (rt - rt0) % 3;
-    if (rt + 1 < rt1)
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    raw_barrier();   // NOT __syncthreads: it would drain the in-flight glds
-    if (rt + 2 < rt1) stage((rt - rt0 + 2) % 3, rt + 2);
+    int buf = (rt - rt0) & 1;
+    if (rt + 1 < rt1) stage(buf ^ 1, rt + 1);
     bf16x8v a_frag[4], b_frag[8];
     #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -772,7 +763,8 @@ gemm_tn_256_kernel(const bf16* __restrict__ Ag, const bf16* __restrict__ Bg,
       for (int j = 0; j < 8; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
-    raw_barrier();     // all reads done before buf is re-staged (+3)
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
   }
 
   if (Asum && q0 == 0 && (wave & 1) == 0) {
@@ -959,7 +951,7 @@ static at::Tensor tn_launch(const at::Tensor& a, const at::Tensor& b,
       : at::zeros({P, Q}, a.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   if (big) {
-    size_t lds = 3 * 2 * TBR * TBP2 * 2;   // 96 KiB (triple buffer)
+    size_t lds = 2 * 2 * TBR * TBP2 * 2;   // 64 KiB
     hipLaunchKernelGGL(gemm_tn_256_kernel, dim3(ntile * splitr), dim3(512),
         lds, stream,
         (const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
