@@ -149,6 +149,7 @@ def lower_gemm(gm: fx.GraphModule) -> int:
     graph = gm.graph
     n_lowered = 0
     gemm_nt = torch.ops.easydist_amd.gemm_nt.default
+    gemm_nn = torch.ops.easydist_amd.gemm_nn.default
     gemm_tn = torch.ops.easydist_amd.gemm_tn.default
 
     def bf16_2d(n):
@@ -191,22 +192,16 @@ def lower_gemm(gm: fx.GraphModule) -> int:
         elif x is not None and _contig(a):
             y = _strip_t(x)
             if y is not None and _contig(y):
-                # NN via weight transpose: b = t(t(y)) = y [K, N]
+                # NN (dX = dY @ W): gemm_nn keeps the layout decision
+                # inside the op — the hand route pays one weight-sized
+                # transpose copy, the hipBLASLt route takes the strided
+                # operand natively (the old graph-level clone(t(W))
+                # charged BOTH routes)
                 y_val = _val(y)
                 if y_val is not None and y_val.shape[0] % 32 == 0 \
                         and y_val.shape[1] % 8 == 0 and _val(a).shape[0] >= 16:
                     with graph.inserting_before(n):
-                        yt = graph.call_function(aten.t.default, (y,))
-                        ytc = graph.call_function(
-                            aten.clone.default, (yt,),
-                            {"memory_format": torch.contiguous_format})
-                        new = graph.call_function(gemm_nt, (a, ytc, bias))
-                    try:
-                        yt.meta["val"] = y_val.t()
-                        ytc.meta["val"] = y_val.t().clone(
-                            memory_format=torch.contiguous_format)
-                    except Exception:
-                        pass
+                        new = graph.call_function(gemm_nn, (a, y, bias))
         if new is None and bias is None:
             xa = _strip_t(a)
             if xa is not None and _contig(xa) and _contig(b) \

@@ -309,3 +309,49 @@ lib.impl("gemm_nt_gelu", _nt_gelu_cuda, "CUDA")
 def _nt_gelu_fake(a, bt, bias, tanh_approx):
     return (a.new_empty((a.shape[0], bt.shape[0])),
             a.new_empty((a.shape[0], bt.shape[0])))
+
+
+# NN-layout GEMM: C = a @ b with b [K, N] row-major (the dX backward:
+# dX = dY @ W). The HAND kernel wants a K-contiguous B operand, so ITS
+# route pays one weight-sized transpose copy; the hipBLASLt route takes
+# the strided layout natively and skips it. Keeping the transpose INSIDE
+# the op lets the per-shape profiler price each route honestly — the
+# old lowering cloned t(W) in the graph unconditionally.
+lib.define("gemm_nn(Tensor a, Tensor b, Tensor? bias) -> Tensor")
+
+
+def _nn_cpu(a, b, bias):
+    if bias is not None:
+        return torch.addmm(bias, a, b)
+    return a @ b
+
+
+def _nn_cuda(a, b, bias):
+    ext = load_extension()
+    if ext is None or not mdconfig.use_hip_kernels or a.dim() != 2 \
+            or b.dim() != 2 or a.dtype != torch.bfloat16 \
+            or b.dtype != torch.bfloat16 \
+            or not _nt_supported(a, b.t()):   # shape check only (view)
+        return _nn_cpu(a, b, bias)
+    a = a.contiguous()
+
+    def hand():
+        # the hand kernel's route INCLUDES its weight transpose copy, so
+        # the profiler prices it honestly against the strided aten mm
+        bt = b.t().contiguous()
+        b_c = bias.contiguous() if bias is not None else None
+        return ext.gemm_nt(a, bt, b_c)
+
+    key = ("nn", a.shape[0], a.shape[1], b.shape[1], bias is not None)
+    if _choose(key, hand, lambda: _nn_cpu(a, b, bias)):
+        return hand()
+    return _nn_cpu(a, b, bias)
+
+
+lib.impl("gemm_nn", _nn_cpu, "CPU")
+lib.impl("gemm_nn", _nn_cuda, "CUDA")
+
+
+@torch.library.register_fake("easydist_amd::gemm_nn")
+def _nn_fake(a, b, bias):
+    return a.new_empty((a.shape[0], b.shape[1]))

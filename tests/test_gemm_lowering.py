@@ -63,7 +63,7 @@ def test_lower_gemm_node_mix_and_golden():
              if n.op == "call_function"]
     n_nt = sum(1 for s in names
                if s in ("gemm_nt.default", "gemm_nt_act.default",
-                        "gemm_nt_gelu.default"))
+                        "gemm_nt_gelu.default", "gemm_nn.default"))
     n_tn = sum(1 for s in names
                if s in ("gemm_tn.default", "gemm_tn_asum.default"))
     # 2 fwd NT + 2 dX NT (weight-transposed) and 2 dW TN
