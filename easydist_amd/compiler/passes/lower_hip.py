@@ -538,7 +538,9 @@ def lower_gelu_bwd_fuse(gm: fx.GraphModule) -> int:
     """
     graph = gm.graph
     nt_act = torch.ops.easydist_amd.gemm_nt_act.default
+    nn_act = torch.ops.easydist_amd.gemm_nn_act.default
     nt = torch.ops.easydist_amd.gemm_nt.default
+    nn = torch.ops.easydist_amd.gemm_nn.default
     n_fused = 0
     for n in list(graph.nodes):
         if n.op != "call_function" \
@@ -552,7 +554,7 @@ def lower_gelu_bwd_fuse(gm: fx.GraphModule) -> int:
         g2 = _strip_view(g) if _strip_view(g) is not None else g
         x2 = _strip_view(x) if _strip_view(x) is not None else x
         if not (isinstance(g2, fx.Node) and g2.op == "call_function"
-                and g2.target is nt):
+                and g2.target in (nt, nn)):
             continue
         gv, xv = _val(g2), _val(x2)
         if not (isinstance(xv, torch.Tensor) and xv.dim() == 2
@@ -565,8 +567,9 @@ def lower_gelu_bwd_fuse(gm: fx.GraphModule) -> int:
         if g_users - {n, g}:
             continue
         a, bt, bias = g2.args
+        fused_op = nt_act if g2.target is nt else nn_act
         with graph.inserting_before(g2):
-            new = graph.call_function(nt_act, (a, bt, bias, act, x2))
+            new = graph.call_function(fused_op, (a, bt, bias, act, x2))
         new.meta = dict(g2.meta)
         g2.replace_all_uses_with(new)
         graph.erase_node(g2)

@@ -355,3 +355,48 @@ lib.impl("gemm_nn", _nn_cuda, "CUDA")
 @torch.library.register_fake("easydist_amd::gemm_nn")
 def _nn_fake(a, b, bias):
     return a.new_empty((a.shape[0], b.shape[1]))
+
+
+# gemm_nn with a fused activation epilogue (the dX + gelu_backward
+# fusion when the incoming grad is an NN-layout GEMM): hand route =
+# weight transpose + gemm_nt_act; aten route = strided mm + the
+# bandwidth-rate gelu_bwd kernel.
+lib.define("gemm_nn_act(Tensor a, Tensor b, Tensor? bias, int act, "
+           "Tensor? aux) -> Tensor")
+
+
+def _nn_act_cpu(a, b, bias, act, aux):
+    return _apply_act(_nn_cpu(a, b, bias), act, aux)
+
+
+def _nn_act_cuda(a, b, bias, act, aux):
+    ext = load_extension()
+    if ext is None or not mdconfig.use_hip_kernels or a.dim() != 2 \
+            or b.dim() != 2 or a.dtype != torch.bfloat16 \
+            or b.dtype != torch.bfloat16 \
+            or not _nt_supported(a, b.t()):
+        return _nn_act_cpu(a, b, bias, act, aux)
+    a = a.contiguous()
+    x_c = aux.contiguous() if aux is not None else None
+
+    def hand():
+        bt = b.t().contiguous()
+        b_c = bias.contiguous() if bias is not None else None
+        return ext.gemm_nt_act(a, bt, b_c, act, x_c)
+
+    def aten():
+        return _act_fast(ext, _nn_cpu(a, b, bias), act, aux)
+
+    key = ("nn", a.shape[0], a.shape[1], b.shape[1], bias is not None, act)
+    if _choose(key, hand, aten):
+        return hand()
+    return aten()
+
+
+lib.impl("gemm_nn_act", _nn_act_cpu, "CPU")
+lib.impl("gemm_nn_act", _nn_act_cuda, "CUDA")
+
+
+@torch.library.register_fake("easydist_amd::gemm_nn_act")
+def _nn_act_fake(a, b, bias, act, aux):
+    return a.new_empty((a.shape[0], b.shape[1]))
