@@ -200,8 +200,25 @@ def lower_gemm(gm: fx.GraphModule) -> int:
                 y_val = _val(y)
                 if y_val is not None and y_val.shape[0] % 32 == 0 \
                         and y_val.shape[1] % 8 == 0 and _val(a).shape[0] >= 16:
-                    with graph.inserting_before(n):
-                        new = graph.call_function(gemm_nn, (a, y, bias))
+                    import os as _os
+                    if _os.environ.get("EASYDIST_NN_LOWER", "1") != "0":
+                        with graph.inserting_before(n):
+                            new = graph.call_function(gemm_nn, (a, y, bias))
+                    else:
+                        # legacy form: graph-level clone(t(W)) + gemm_nt
+                        with graph.inserting_before(n):
+                            yt = graph.call_function(aten.t.default, (y,))
+                            ytc = graph.call_function(
+                                aten.clone.default, (yt,),
+                                {"memory_format": torch.contiguous_format})
+                            new = graph.call_function(gemm_nt,
+                                                      (a, ytc, bias))
+                        try:
+                            yt.meta["val"] = y_val.t()
+                            ytc.meta["val"] = y_val.t().clone(
+                                memory_format=torch.contiguous_format)
+                        except Exception:
+                            pass
         if new is None and bias is None:
             xa = _strip_t(a)
             if xa is not None and _contig(xa) and _contig(b) \
