@@ -200,8 +200,18 @@ def lower_gemm(gm: fx.GraphModule) -> int:
                 y_val = _val(y)
                 if y_val is not None and y_val.shape[0] % 32 == 0 \
                         and y_val.shape[1] % 8 == 0 and _val(a).shape[0] >= 16:
+                    # weight-size heuristic (both directions measured
+                    # within one box): small weights -> the per-step
+                    # clone is cheap and Tensile's NT layout beats its
+                    # strided NN (legacy wins GPT-2 small by ~1%); big
+                    # weights -> the clone dominates (gemm_nn wins
+                    # GPT-2 1.3B by ~2%). Threshold 8M elements.
                     import os as _os
-                    if _os.environ.get("EASYDIST_NN_LOWER", "1") != "0":
+                    nn_env = _os.environ.get("EASYDIST_NN_LOWER", "auto")
+                    use_nn = (nn_env == "1"
+                              or (nn_env == "auto"
+                                  and y_val.numel() >= (1 << 23)))
+                    if use_nn:
                         with graph.inserting_before(n):
                             new = graph.call_function(gemm_nn, (a, y, bias))
                     else:
