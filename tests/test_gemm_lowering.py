@@ -89,3 +89,16 @@ def test_lower_gemm_node_mix_and_golden():
         # operand layouts, so grads differ at bf16 rounding level
         assert torch.allclose(p1, p2, rtol=1e-2, atol=1e-3), \
             (n1, float((p1 - p2).abs().max()))
+
+
+def test_gemm_nn_op_numerics():
+    # NN-layout op (dX backward): C = a @ b (+bias), CPU reference path
+    torch.manual_seed(3)
+    a = torch.randn(32, 64)
+    b = torch.randn(64, 48)
+    bias = torch.randn(48)
+    out = torch.ops.easydist_amd.gemm_nn(a, b, bias)
+    assert torch.allclose(out, torch.addmm(bias, a, b), atol=1e-5)
+    out2 = torch.ops.easydist_amd.gemm_nn_act(a, b, None, 1, None)
+    ref = torch.nn.functional.gelu(a @ b, approximate="tanh")
+    assert torch.allclose(out2, ref, atol=1e-5)
